@@ -1,0 +1,195 @@
+"""In-memory cluster — the envtest/fake-client analog.
+
+The reference tests against controller-runtime's fake client and envtest
+(SURVEY §4). This FakeCluster plays both roles for the new framework: a
+thread-safe object store with API-server semantics (deep-copy on read and
+write, resourceVersion bumps, watch event streams) that the Manager,
+reconcilers and engines run against in tests, the emulator, and bench.py.
+
+A REST-backed client implementing the same surface can be substituted for
+a real Kubernetes API server.
+"""
+from __future__ import annotations
+
+import copy
+import queue
+import threading
+from dataclasses import dataclass
+from typing import Any, Callable, Dict, List, Optional, Tuple
+
+from ..api.types import utcnow
+
+ADDED = "ADDED"
+MODIFIED = "MODIFIED"
+DELETED = "DELETED"
+
+
+@dataclass
+class WatchEvent:
+    type: str  # ADDED | MODIFIED | DELETED
+    kind: str
+    obj: Any
+
+
+class NotFoundError(KeyError):
+    def __init__(self, kind: str, namespace: str, name: str):
+        super().__init__(f"{kind} {namespace}/{name} not found")
+        self.kind = kind
+        self.namespace = namespace
+        self.name = name
+
+
+class ConflictError(RuntimeError):
+    pass
+
+
+def _kind_of(obj: Any) -> str:
+    return getattr(obj, "kind", obj.__class__.__name__)
+
+
+def _meta(obj: Any):
+    return obj.metadata
+
+
+class FakeCluster:
+    """Thread-safe in-memory object store with watches."""
+
+    def __init__(self) -> None:
+        self._lock = threading.RLock()
+        # (kind, namespace, name) -> object
+        self._objects: Dict[Tuple[str, str, str], Any] = {}
+        self._rv = 0
+        self._watchers: List[Tuple[Optional[set], "queue.Queue[WatchEvent]"]] = []
+
+    # --- internals ---
+
+    def _key(self, obj: Any) -> Tuple[str, str, str]:
+        m = _meta(obj)
+        return (_kind_of(obj), m.namespace, m.name)
+
+    def _notify(self, event_type: str, obj: Any) -> None:
+        kind = _kind_of(obj)
+        for kinds, q in list(self._watchers):
+            if kinds is None or kind in kinds:
+                q.put(WatchEvent(type=event_type, kind=kind, obj=copy.deepcopy(obj)))
+
+    # --- CRUD ---
+
+    def create(self, obj: Any) -> Any:
+        with self._lock:
+            key = self._key(obj)
+            if key in self._objects:
+                raise ConflictError(f"{key} already exists")
+            stored = copy.deepcopy(obj)
+            self._rv += 1
+            m = _meta(stored)
+            m.resource_version = self._rv
+            if m.creation_timestamp is None:
+                m.creation_timestamp = utcnow()
+            if not m.uid:
+                m.uid = f"uid-{self._rv}"
+            self._objects[key] = stored
+            self._notify(ADDED, stored)
+            return copy.deepcopy(stored)
+
+    def get(self, kind: str, namespace: str, name: str) -> Any:
+        with self._lock:
+            obj = self._objects.get((kind, namespace, name))
+            if obj is None:
+                raise NotFoundError(kind, namespace, name)
+            return copy.deepcopy(obj)
+
+    def try_get(self, kind: str, namespace: str, name: str) -> Optional[Any]:
+        try:
+            return self.get(kind, namespace, name)
+        except NotFoundError:
+            return None
+
+    def list(
+        self,
+        kind: str,
+        namespace: Optional[str] = None,
+        label_selector: Optional[Dict[str, str]] = None,
+        predicate: Optional[Callable[[Any], bool]] = None,
+    ) -> List[Any]:
+        with self._lock:
+            out = []
+            for (k, ns, _), obj in self._objects.items():
+                if k != kind:
+                    continue
+                if namespace is not None and ns != namespace:
+                    continue
+                if label_selector:
+                    labels = _meta(obj).labels
+                    if not all(labels.get(lk) == lv for lk, lv in label_selector.items()):
+                        continue
+                if predicate is not None and not predicate(obj):
+                    continue
+                out.append(copy.deepcopy(obj))
+            out.sort(key=lambda o: (_meta(o).namespace, _meta(o).name))
+            return out
+
+    def update(self, obj: Any, bump_generation: bool = False) -> Any:
+        with self._lock:
+            key = self._key(obj)
+            if key not in self._objects:
+                raise NotFoundError(*key)
+            stored = copy.deepcopy(obj)
+            self._rv += 1
+            m = _meta(stored)
+            m.resource_version = self._rv
+            if bump_generation:
+                m.generation += 1
+            self._objects[key] = stored
+            self._notify(MODIFIED, stored)
+            return copy.deepcopy(stored)
+
+    def update_status(self, obj: Any) -> Any:
+        """Status-subresource write: replaces only the status of the stored
+        object (spec/metadata changes in `obj` are ignored), mirroring the
+        reference's Status().Patch with the full-nested-object merge patch
+        (variantautoscaling_controller.go:226-252, issue #731)."""
+        with self._lock:
+            key = self._key(obj)
+            stored = self._objects.get(key)
+            if stored is None:
+                raise NotFoundError(*key)
+            self._rv += 1
+            stored.status = copy.deepcopy(obj.status)
+            _meta(stored).resource_version = self._rv
+            self._notify(MODIFIED, stored)
+            return copy.deepcopy(stored)
+
+    def delete(self, kind: str, namespace: str, name: str) -> None:
+        with self._lock:
+            obj = self._objects.pop((kind, namespace, name), None)
+            if obj is None:
+                raise NotFoundError(kind, namespace, name)
+            self._notify(DELETED, obj)
+
+    # --- scale subresource ---
+
+    def scale(self, kind: str, namespace: str, name: str, replicas: int) -> Any:
+        """Write spec.replicas through the scale subresource (the
+        DirectActuator path, reference internal/actuator/direct_actuator.go)."""
+        with self._lock:
+            obj = self._objects.get((kind, namespace, name))
+            if obj is None:
+                raise NotFoundError(kind, namespace, name)
+            obj.replicas = int(replicas)
+            self._rv += 1
+            _meta(obj).resource_version = self._rv
+            self._notify(MODIFIED, obj)
+            return copy.deepcopy(obj)
+
+    # --- watches ---
+
+    def watch(self, kinds: Optional[List[str]] = None) -> "queue.Queue[WatchEvent]":
+        q: "queue.Queue[WatchEvent]" = queue.Queue()
+        with self._lock:
+            self._watchers.append((set(kinds) if kinds else None, q))
+        return q
+
+    def stop_watch(self, q: "queue.Queue[WatchEvent]") -> None:
+        with self._lock:
+            self._watchers = [(k, w) for (k, w) in self._watchers if w is not q]
