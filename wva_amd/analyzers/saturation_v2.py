@@ -413,6 +413,12 @@ class SaturationAnalyzerV2:
         rec = self.capacity_store.get(namespace, model_id, variant_name)
         if rec is None or rec.vllm_params is None:
             return None
+        # Improvement over the reference: when no live metrics exist for the
+        # variant the accelerator from the caller is empty; fall back to the
+        # accelerator recorded when the deployment was parsed so the
+        # cross-variant match can still succeed.
+        if not accelerator:
+            accelerator = rec.accelerator_name
         return self.capacity_store.find_compatible(
             model_id, accelerator, gpu_count, rec.vllm_params
         )
