@@ -1,0 +1,179 @@
+"""Pod-scraping metrics source (EPP endpoint-picker pods).
+
+Parity: reference internal/collector/source/pod/pod_scraping_source.go
+:29-388 — discovers ready pods behind the EPP Service selector, scrapes
+each pod's metrics endpoint concurrently (max 10 in flight, 5s timeout,
+optional bearer token), parses Prometheus text exposition and serves all
+samples under the single query name `all_metrics` with `pod` and
+`__name__` labels attached.
+"""
+from __future__ import annotations
+
+import concurrent.futures
+import re
+import time
+from typing import Callable, Dict, List, Optional
+
+import requests
+
+from ..kube.fake import FakeCluster
+from ..kube.objects import EndpointPool, Pod
+from ..utils.logging import get_logger
+from .query_template import QUERY_TYPE_METRIC_NAME, QueryList, QueryTemplate
+from .source import MetricResult, MetricValue, RefreshSpec
+
+log = get_logger("collector.pod_scrape")
+
+ALL_METRICS_QUERY = "all_metrics"
+DEFAULT_SCRAPE_TIMEOUT_SECONDS = 5.0
+DEFAULT_MAX_CONCURRENT_SCRAPES = 10
+
+_SAMPLE_RE = re.compile(
+    r'^(?P<name>[a-zA-Z_:][a-zA-Z0-9_:]*)'
+    r'(?:\{(?P<labels>[^}]*)\})?'
+    r'\s+(?P<value>[^\s]+)'
+    r'(?:\s+(?P<ts>\d+))?$'
+)
+_LABEL_RE = re.compile(r'(\w+)="((?:[^"\\]|\\.)*)"')
+
+
+def parse_prometheus_text(text: str) -> List[MetricValue]:
+    """Parse Prometheus text exposition format into MetricValues with
+    __name__ labels."""
+    now = time.time()
+    values: List[MetricValue] = []
+    for line in text.splitlines():
+        line = line.strip()
+        if not line or line.startswith("#"):
+            continue
+        m = _SAMPLE_RE.match(line)
+        if not m:
+            continue
+        labels = {"__name__": m.group("name")}
+        raw_labels = m.group("labels")
+        if raw_labels:
+            for lm in _LABEL_RE.finditer(raw_labels):
+                labels[lm.group(1)] = (
+                    lm.group(2).replace('\\"', '"').replace("\\\\", "\\")
+                    .replace("\\n", "\n")
+                )
+        try:
+            value = float(m.group("value"))
+        except ValueError:
+            continue
+        ts = m.group("ts")
+        timestamp = float(ts) / 1000.0 if ts else now
+        values.append(MetricValue(value=value, timestamp=timestamp, labels=labels))
+    return values
+
+
+# (url, headers, timeout) → response text; pluggable for tests/emulation
+FetchFunc = Callable[[str, Dict[str, str], float], str]
+
+
+def _default_fetch(url: str, headers: Dict[str, str], timeout: float) -> str:
+    resp = requests.get(url, headers=headers, timeout=timeout)
+    resp.raise_for_status()
+    return resp.text
+
+
+class PodScrapingSource:
+    """One instance per EndpointPool (InferencePool)."""
+
+    def __init__(
+        self,
+        cluster: FakeCluster,
+        pool: EndpointPool,
+        bearer_token: str = "",
+        scrape_timeout_seconds: float = DEFAULT_SCRAPE_TIMEOUT_SECONDS,
+        max_concurrent_scrapes: int = DEFAULT_MAX_CONCURRENT_SCRAPES,
+        fetch: Optional[FetchFunc] = None,
+        metrics_path: str = "/metrics",
+    ):
+        self.cluster = cluster
+        self.pool = pool
+        self.bearer_token = bearer_token
+        self.scrape_timeout_seconds = scrape_timeout_seconds
+        self.max_concurrent_scrapes = max(1, max_concurrent_scrapes)
+        self.fetch = fetch or _default_fetch
+        self.metrics_path = metrics_path
+        self._query_list = QueryList()
+        self._query_list.must_register(QueryTemplate(
+            name=ALL_METRICS_QUERY,
+            type=QUERY_TYPE_METRIC_NAME,
+            template="",
+            params=[],
+            description="All metrics scraped from EPP pods behind the pool service",
+        ))
+        self._last_result: Optional[MetricResult] = None
+
+    def name(self) -> str:
+        return f"pool/{self.pool.namespace}/{self.pool.name}"
+
+    def query_list(self) -> QueryList:
+        return self._query_list
+
+    # --- pod discovery via EPP service selector ---
+
+    def _discover_pods(self) -> List[Pod]:
+        picker = self.pool.endpoint_picker
+        svc = self.cluster.try_get("Service", picker.namespace, picker.service_name)
+        if svc is None:
+            return []
+        pods = self.cluster.list("Pod", namespace=picker.namespace)
+        ready = []
+        for pod in pods:
+            if not pod.is_ready():
+                continue
+            if svc.selector and all(
+                pod.metadata.labels.get(k) == v for k, v in svc.selector.items()
+            ):
+                ready.append(pod)
+        return ready
+
+    def _scrape_pod(self, pod: Pod) -> List[MetricValue]:
+        picker = self.pool.endpoint_picker
+        host = pod.status.pod_ip or pod.name
+        url = f"http://{host}:{picker.metrics_port_number}{self.metrics_path}"
+        headers = {}
+        if self.bearer_token:
+            headers["Authorization"] = f"Bearer {self.bearer_token}"
+        text = self.fetch(url, headers, self.scrape_timeout_seconds)
+        values = parse_prometheus_text(text)
+        for v in values:
+            v.labels["pod"] = pod.name
+        return values
+
+    # --- MetricsSource ---
+
+    def refresh(self, spec: RefreshSpec) -> Dict[str, MetricResult]:
+        pods = self._discover_pods()
+        all_values: List[MetricValue] = []
+        errors: List[Exception] = []
+        if pods:
+            with concurrent.futures.ThreadPoolExecutor(
+                max_workers=min(self.max_concurrent_scrapes, len(pods))
+            ) as pool:
+                futures = {pool.submit(self._scrape_pod, p): p for p in pods}
+                for fut in concurrent.futures.as_completed(futures):
+                    try:
+                        all_values.extend(fut.result())
+                    except Exception as e:  # noqa: BLE001 — per-pod failure tolerated
+                        errors.append(e)
+                        log.debug(
+                            "scrape failed for pod %s: %s", futures[fut].name, e
+                        )
+        result = MetricResult(
+            query=ALL_METRICS_QUERY,
+            values=all_values,
+            fetched_at=time.time(),
+            # All pods failing (with pods present) is an error condition
+            error=errors[0] if errors and not all_values else None,
+        )
+        self._last_result = result
+        return {ALL_METRICS_QUERY: result}
+
+    def get(self, query: str, params: Dict[str, str]) -> Optional[MetricResult]:
+        if query != ALL_METRICS_QUERY:
+            return None
+        return self._last_result
