@@ -1,0 +1,306 @@
+"""End-to-end tests on the emulated cluster (the Kind + llm-d-inference-sim
+analog, BASELINE config #1): a VariantAutoscaling CR on a FakeCluster,
+simulated vLLM replicas, the saturation engine deciding replicas, the
+reconciler persisting status, and wva_* metrics emitted for HPA.
+"""
+import time
+
+import pytest
+from prometheus_client import CollectorRegistry
+
+from wva_amd.api.types import (
+    CrossVersionObjectReference,
+    ObjectMeta,
+    VariantAutoscaling,
+    VariantAutoscalingSpec,
+)
+from wva_amd.api import conditions as cond
+from wva_amd.app import build_app
+from wva_amd.config.config import Config
+from wva_amd.config.saturation import SaturationScalingConfig
+from wva_amd.emulator.cluster_sim import ClusterSim
+from wva_amd.emulator.sim_source import SimMetricsSource
+from wva_amd.emulator.vllm_sim import RequestSpec, ServiceProfile
+from wva_amd.emulator.workload import constant_qps
+from wva_amd.kube.fake import FakeCluster
+from wva_amd.kube.objects import (
+    Container,
+    Deployment,
+    EndpointPicker,
+    EndpointPool,
+    InferencePool,
+    Node,
+    PodTemplateSpec,
+    Service,
+    ServicePort,
+)
+
+MODEL = "meta-llama/Llama-3.1-8B"
+NS = "default"
+VARIANT = "vllm-llama"
+
+
+def mi355x_node(name="mi355x-0", gpus=8):
+    return Node(
+        metadata=ObjectMeta(
+            name=name,
+            labels={
+                "amd.com/gpu.product": "AMD-Instinct-MI355X-288GB",
+                "amd.com/gpu.memory": "294912",
+            },
+        ),
+        allocatable={"amd.com/gpu": str(gpus)},
+    )
+
+
+def make_stack(
+    replicas=1,
+    pod_ready_delay=0.0,
+    analyzer="",
+    profile=None,
+    qps_profile=None,
+):
+    cluster = FakeCluster()
+    cluster.create(mi355x_node())
+    deploy = Deployment(
+        metadata=ObjectMeta(name=VARIANT, namespace=NS),
+        replicas=replicas,
+        selector={"app": VARIANT},
+        template=PodTemplateSpec(
+            labels={"app": VARIANT},
+            containers=[
+                Container(
+                    args=["--max-num-seqs", "256", "--block-size", "16"],
+                    requests={"amd.com/gpu": "1"},
+                )
+            ],
+        ),
+    )
+    cluster.create(deploy)
+    va = VariantAutoscaling(
+        metadata=ObjectMeta(
+            name=VARIANT,
+            namespace=NS,
+            labels={"inference.optimization/acceleratorName": "MI355X"},
+        ),
+        spec=VariantAutoscalingSpec(
+            scale_target_ref=CrossVersionObjectReference(name=VARIANT),
+            model_id=MODEL,
+        ),
+    )
+    cluster.create(va)
+
+    sim = ClusterSim(cluster, pod_ready_delay_s=pod_ready_delay)
+    prof = profile or ServiceProfile(num_gpu_blocks=20_000)  # small for tests
+    sim.register_variant(MODEL, NS, VARIANT, prof)
+    sim.reconcile_deployments()
+    source = SimMetricsSource(sim)
+
+    config = Config()
+    config.update_saturation_config(
+        SaturationScalingConfig.from_dict(
+            {"analyzerName": analyzer} if analyzer else {}
+        )
+    )
+    config.mark_bootstrap_complete()
+    app = build_app(
+        cluster,
+        config,
+        source=source,
+        metrics_registry=CollectorRegistry(),
+        start_engines=False,
+    )
+    return cluster, sim, app
+
+
+def run_sim(sim, model, qps, seconds, dt=0.25, input_tokens=100, output_tokens=50):
+    profile = constant_qps(qps)
+    steps = int(seconds / dt)
+    for _ in range(steps):
+        sim.generate_arrivals(model, profile, dt, input_tokens, output_tokens)
+        sim.advance(dt)
+
+
+class TestEndToEnd:
+    def test_idle_cluster_no_scale_up(self):
+        cluster, sim, app = make_stack(replicas=2)
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=0.5, seconds=10)
+        app.saturation_engine.optimize()
+        va = cluster.get("VariantAutoscaling", NS, VARIANT)
+        # engine wrote status directly; decision cache populated
+        d = app.decision_cache.get(NS, VARIANT)
+        assert d is not None
+        assert d.target_replicas <= 2
+
+    def test_overload_scales_up(self):
+        prof = ServiceProfile(
+            alpha_ms=50.0, beta_ms=2.0, max_num_seqs=8, num_gpu_blocks=500
+        )
+        cluster, sim, app = make_stack(replicas=1, profile=prof)
+        model = sim.model(MODEL, NS)
+        # overwhelm the single tiny replica (kv capacity 8000 tokens)
+        run_sim(sim, model, qps=20, seconds=10)
+        app.saturation_engine.optimize()
+        d = app.decision_cache.get(NS, VARIANT)
+        assert d is not None
+        assert d.target_replicas >= 2
+        assert d.accelerator_name == "MI355X"
+
+    def test_reconciler_persists_status(self):
+        prof = ServiceProfile(
+            alpha_ms=50.0, beta_ms=2.0, max_num_seqs=8, num_gpu_blocks=500
+        )
+        cluster, sim, app = make_stack(replicas=1, profile=prof)
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=20, seconds=10)
+        app.saturation_engine.optimize()
+        app.va_reconciler.reconcile(NS, VARIANT)
+        va = cluster.get("VariantAutoscaling", NS, VARIANT)
+        assert va.status.desired_optimized_alloc.num_replicas >= 2
+        assert va.status.desired_optimized_alloc.accelerator == "MI355X"
+        assert cond.is_condition_true(va, "TargetResolved")
+        assert cond.is_condition_true(va, "MetricsAvailable")
+        assert cond.is_condition_true(va, "OptimizationReady")
+
+    def test_wva_metrics_emitted(self):
+        cluster, sim, app = make_stack(replicas=1)
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=1, seconds=5)
+        app.saturation_engine.optimize()
+        metric_names = {
+            m.name for m in app.emitter.registry.collect()
+        }
+        assert "wva_desired_replicas" in metric_names
+        assert "wva_current_replicas" in metric_names
+        assert "wva_desired_ratio" in metric_names
+        # desired gauge carries the accelerator label
+        for family in app.emitter.registry.collect():
+            if family.name == "wva_desired_replicas":
+                sample = family.samples[0]
+                assert sample.labels["accelerator_type"] == "MI355X"
+                assert sample.labels["variant_name"] == VARIANT
+                assert sample.labels["namespace"] == NS
+
+    def test_v2_analyzer_path(self):
+        prof = ServiceProfile(
+            alpha_ms=50.0, beta_ms=2.0, max_num_seqs=8, num_gpu_blocks=500
+        )
+        cluster, sim, app = make_stack(
+            replicas=1, profile=prof, analyzer="saturation"
+        )
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=20, seconds=10)
+        app.saturation_engine.optimize()
+        d = app.decision_cache.get(NS, VARIANT)
+        assert d is not None
+        assert d.target_replicas >= 2
+        # V2 learned a live capacity record
+        rec = app.capacity_store.get(NS, MODEL, VARIANT)
+        assert rec is not None and rec.learned_from == "live"
+
+    def test_manager_full_loop_with_watches(self):
+        """Manager runs engines + reconcilers; deployment event and
+        decision trigger both feed the VA reconciler."""
+        prof = ServiceProfile(
+            alpha_ms=50.0, beta_ms=2.0, max_num_seqs=8, num_gpu_blocks=500
+        )
+        cluster, sim, app = make_stack(replicas=1, profile=prof)
+        model = sim.model(MODEL, NS)
+        app.manager.start()
+        try:
+            run_sim(sim, model, qps=20, seconds=10)
+            app.saturation_engine.optimize()  # manual tick (engines not started)
+            deadline = time.time() + 3
+            while time.time() < deadline:
+                va = cluster.get("VariantAutoscaling", NS, VARIANT)
+                if va.status.desired_optimized_alloc.num_replicas >= 2:
+                    break
+                time.sleep(0.05)
+            va = cluster.get("VariantAutoscaling", NS, VARIANT)
+            assert va.status.desired_optimized_alloc.num_replicas >= 2
+        finally:
+            app.manager.stop()
+
+    def test_scale_to_zero(self):
+        cluster, sim, app = make_stack(replicas=1)
+        model = sim.model(MODEL, NS)
+        # no traffic at all; enable scale-to-zero with short retention
+        from wva_amd.config.scale_to_zero import ModelScaleToZeroConfig
+
+        app.config.update_scale_to_zero_config(
+            {
+                "default": ModelScaleToZeroConfig(
+                    enable_scale_to_zero=True, retention_period="1m"
+                )
+            }
+        )
+        run_sim(sim, model, qps=0, seconds=5)
+        app.saturation_engine.optimize()
+        d = app.decision_cache.get(NS, VARIANT)
+        assert d is not None and d.target_replicas == 0
+
+    def test_scale_from_zero(self):
+        cluster, sim, app = make_stack(replicas=0)
+        model = sim.model(MODEL, NS)
+        # EPP infrastructure: InferencePool + EPP service + pods served by
+        # the sim's metrics text
+        cluster.create(Service(
+            metadata=ObjectMeta(name="pool-epp", namespace=NS),
+            selector={"app": "epp"},
+            ports=[ServicePort(name="metrics", port=9090)],
+        ))
+        from wva_amd.kube.objects import Pod, PodStatus
+
+        cluster.create(Pod(
+            metadata=ObjectMeta(
+                name="epp-0", namespace=NS, labels={"app": "epp"}
+            ),
+            status=PodStatus(phase="Running", ready=True, pod_ip="10.1.0.1"),
+        ))
+        # datastore fetch hook returns the sim's EPP metric text
+        app.datastore.scrape_fetch = lambda url, headers, timeout: (
+            sim.epp_metrics_text(NS)
+        )
+        cluster.create(InferencePool(
+            metadata=ObjectMeta(name="pool", namespace=NS),
+            selector={"app": VARIANT},
+            epp_service_name="pool-epp",
+        ))
+        app.inferencepool_reconciler.reconcile(NS, "pool")
+
+        # requests arrive while no replicas exist → scheduler queue grows
+        run_sim(sim, model, qps=2, seconds=2)
+        assert len(model.scheduler_queue) > 0
+
+        app.scale_from_zero_engine.optimize()
+        deploy = cluster.get("Deployment", NS, VARIANT)
+        assert deploy.replicas == 1
+        va = cluster.get("VariantAutoscaling", NS, VARIANT)
+        assert va.status.desired_optimized_alloc.num_replicas == 1
+        assert cond.is_condition_true(va, "ScaleFromZeroMode")
+
+    def test_full_autoscaling_convergence(self):
+        """QPS ramp: the engine+cluster loop converges to more replicas and
+        the added replicas drain the load (the north-star behavior)."""
+        prof = ServiceProfile(
+            alpha_ms=30.0, beta_ms=1.0, max_num_seqs=16, num_gpu_blocks=2_000
+        )
+        cluster, sim, app = make_stack(replicas=1, profile=prof)
+        model = sim.model(MODEL, NS)
+
+        def actuate():
+            """HPA analog: apply desired replicas to the deployment."""
+            d = app.decision_cache.get(NS, VARIANT)
+            if d is not None and d.target_replicas > 0:
+                deploy = cluster.get("Deployment", NS, VARIANT)
+                if deploy.replicas != d.target_replicas:
+                    cluster.scale("Deployment", NS, VARIANT, d.target_replicas)
+
+        for tick in range(12):
+            run_sim(sim, model, qps=30, seconds=5)
+            app.saturation_engine.optimize()
+            actuate()
+            sim.reconcile_deployments()
+        deploy = cluster.get("Deployment", NS, VARIANT)
+        assert deploy.replicas >= 2

@@ -198,6 +198,14 @@ class VariantDecision:
     metrics_reason: str = ""
     metrics_message: str = ""
 
+    # OptimizationReady condition payload, carried through the DecisionCache
+    # so the reconciler (the single status writer) can persist the condition
+    # the engine computed. The reference sets this condition only on the
+    # engine's local copy (engine.go:915-950) and it is never persisted;
+    # carrying it here is a documented improvement.
+    optimization_ready_reason: str = ""
+    optimization_ready_message: str = ""
+
     def add_decision_step(self, name: str, reason: str, was_constrained: bool) -> None:
         self.decision_steps.append(
             DecisionStep(

@@ -1,0 +1,152 @@
+"""ConfigMap reconciler — live config reload.
+
+Parity: reference internal/controller/configmap_reconciler.go:49-194 and
+configmap_helpers.go — routes `wva-saturation-scaling-config` and
+`wva-model-scale-to-zero-config` by name; a ConfigMap in the controller
+namespace is the global config, one in a tracked/opted-in namespace is a
+namespace-local override; deletion removes the namespace-local override.
+
+Saturation ConfigMap format (docs/saturation-scaling-config.md): data key
+"default" holds the global thresholds; additional keys hold per-model
+overrides with `model_id` (+ optional `namespace`) fields.
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import yaml
+
+from ..config.config import Config
+from ..config.saturation import SaturationScalingConfig
+from ..config.scale_to_zero import parse_scale_to_zero_configmap
+from ..constants import (
+    NAMESPACE_CONFIG_ENABLED_LABEL_KEY,
+    SATURATION_CONFIG_MAP_NAME,
+    SCALE_TO_ZERO_CONFIG_MAP_NAME,
+)
+from ..datastore.datastore import Datastore
+from ..kube.fake import FakeCluster
+from ..kube.objects import ConfigMap
+from ..utils.logging import get_logger
+
+log = get_logger("controllers.configmap")
+
+
+def parse_saturation_configmap(data: Optional[dict]) -> SaturationScalingConfig:
+    """Parse the saturation ConfigMap's data section: key "default" is the
+    base config, other keys are per-model overrides."""
+    data = data or {}
+    base = SaturationScalingConfig()
+    overrides = []
+    for key in sorted(data.keys()):
+        raw = data[key]
+        try:
+            parsed = yaml.safe_load(raw) or {}
+            if not isinstance(parsed, dict):
+                raise ValueError("entry must be a mapping")
+        except Exception as e:  # noqa: BLE001
+            log.info("failed to parse saturation config entry %s: %s", key, e)
+            continue
+        if key == "default":
+            base = SaturationScalingConfig.from_dict(parsed)
+        else:
+            entry = SaturationScalingConfig.from_dict(parsed)
+            if entry.model_id:
+                overrides.append(entry)
+            else:
+                log.info("skipping saturation override without model_id: %s", key)
+    for entry in overrides:
+        base.overrides[f"{entry.model_id}|{entry.namespace}"] = entry
+    base.apply_defaults()
+    try:
+        base.validate()
+    except Exception as e:  # noqa: BLE001
+        log.error("invalid saturation config, keeping previous: %s", e)
+        raise
+    return base
+
+
+def controller_namespace() -> str:
+    return os.environ.get("POD_NAMESPACE", "wva-system")
+
+
+class ConfigMapReconciler:
+    def __init__(self, cluster: FakeCluster, config: Config, datastore: Datastore):
+        self.cluster = cluster
+        self.config = config
+        self.datastore = datastore
+
+    def _namespace_opted_in(self, namespace: str) -> bool:
+        if self.datastore.namespace_is_tracked(namespace):
+            return True
+        ns_obj = self.cluster.try_get("Namespace", "", namespace)
+        if ns_obj is not None and (
+            ns_obj.metadata.labels.get(NAMESPACE_CONFIG_ENABLED_LABEL_KEY) == "true"
+        ):
+            return True
+        return False
+
+    def reconcile(self, namespace: str, name: str) -> None:
+        if name not in (SATURATION_CONFIG_MAP_NAME, SCALE_TO_ZERO_CONFIG_MAP_NAME):
+            return
+        cm: Optional[ConfigMap] = self.cluster.try_get(
+            "ConfigMap", namespace, name
+        )
+        is_global = namespace == controller_namespace()
+
+        if cm is None:
+            # Deletion: remove namespace-local override
+            if not is_global:
+                if name == SATURATION_CONFIG_MAP_NAME:
+                    self.config.remove_saturation_config_for_namespace(namespace)
+                else:
+                    self.config.remove_scale_to_zero_config_for_namespace(namespace)
+                log.info("removed namespace-local config %s/%s", namespace, name)
+            return
+
+        if not is_global and not self._namespace_opted_in(namespace):
+            log.debug(
+                "ignoring ConfigMap %s/%s: namespace not tracked or opted in",
+                namespace,
+                name,
+            )
+            return
+
+        if name == SATURATION_CONFIG_MAP_NAME:
+            try:
+                cfg = parse_saturation_configmap(cm.data)
+            except Exception:  # noqa: BLE001 — keep previous config on error
+                return
+            if is_global:
+                self.config.update_saturation_config(cfg)
+            else:
+                self.config.update_saturation_config_for_namespace(namespace, cfg)
+            log.info(
+                "updated %s saturation config from %s/%s",
+                "global" if is_global else "namespace-local",
+                namespace,
+                name,
+            )
+        else:
+            data = parse_scale_to_zero_configmap(cm.data)
+            if is_global:
+                self.config.update_scale_to_zero_config(data)
+            else:
+                self.config.update_scale_to_zero_config_for_namespace(namespace, data)
+            log.info(
+                "updated %s scale-to-zero config from %s/%s",
+                "global" if is_global else "namespace-local",
+                namespace,
+                name,
+            )
+
+    def bootstrap_initial_configmaps(self) -> None:
+        """Read config from existing ConfigMaps before the engines start
+        (cmd/main.go:322-336); marks bootstrap complete for readyz."""
+        ns = controller_namespace()
+        for name in (SATURATION_CONFIG_MAP_NAME, SCALE_TO_ZERO_CONFIG_MAP_NAME):
+            cm = self.cluster.try_get("ConfigMap", ns, name)
+            if cm is not None:
+                self.reconcile(ns, name)
+        self.config.mark_bootstrap_complete()

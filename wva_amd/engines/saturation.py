@@ -1,0 +1,613 @@
+"""Saturation engine — the 30-second collector → analyzer → optimizer →
+actuator hot loop.
+
+Parity: reference internal/engines/saturation/engine.go:98-1095 and
+engine_v2.go:20-160. Per tick:
+  1. list active VAs (deployment replicas > 0), group by modelID|namespace
+  2. V1 path (default): prepareModelData → V1 analysis → targets →
+     enforcer → decisions → optional GPU limiter
+     V2 path (analyzerName == "saturation"): capacity-store pre-population
+     from deployments → V2 analyzer → CostAwareOptimizer (all models at
+     once) → enforcer bridge per model
+  3. applySaturationDecisions: per VA refetch, set
+     status.desiredOptimizedAlloc + OptimizationReady condition, emit HPA
+     metrics via the Actuator, write the DecisionCache and push the
+     DecisionTrigger — the reconciler persists status.
+Safety net: on analysis failure, emit previous-desired/current metrics so
+the HPA never starves (engine.go:1022-1095).
+
+Improvement over the reference (documented): the V2 analyzer input includes
+scheduler flow-control queue metrics (the reference leaves that as a TODO
+in engine_v2.go:54).
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+from ..actuator.actuator import Actuator
+from ..analyzers.capacity_store import CapacityKnowledgeStore
+from ..analyzers.interfaces import (
+    ACTION_SCALE_DOWN,
+    ACTION_SCALE_UP,
+    AnalyzerInput,
+    ModelSaturationAnalysis,
+    VariantDecision,
+    VariantReplicaState,
+    VariantSaturationAnalysis,
+)
+from ..analyzers.saturation_v1 import SaturationAnalyzerV1
+from ..analyzers.saturation_v2 import SaturationAnalyzerV2
+from ..api import conditions as cond
+from ..api.types import OptimizedAlloc, VariantAutoscaling, utcnow
+from ..collector.replica_metrics import ReplicaMetricsCollector
+from ..config.config import Config
+from ..constants import ACCELERATOR_LABEL_KEY, GPU_VENDORS, GPU_RESOURCE_SUFFIX
+from ..kube.fake import FakeCluster
+from ..kube.objects import Deployment
+from ..pipeline.enforcer import Enforcer
+from ..pipeline.limiter import DefaultLimiter, ModelScalingRequest
+from ..pipeline.optimizer import ACTION_NO_CHANGE, CostAwareOptimizer
+from ..runtime.executor import PollingExecutor
+from ..utils.logging import get_logger
+from ..utils.variant import (
+    active_variant_autoscalings,
+    group_variant_autoscaling_by_model,
+)
+from .common import DecisionCache, DecisionTrigger
+
+log = get_logger("engines.saturation")
+
+METRICS_REASON_AVAILABLE = "MetricsFound"
+METRICS_MESSAGE_AVAILABLE = "Saturation metrics collected successfully"
+METRICS_REASON_UNAVAILABLE = "MetricsMissing"
+METRICS_MESSAGE_UNAVAILABLE = (
+    "Saturation metrics unavailable (check ServiceMonitor and Prometheus)"
+)
+
+DEFAULT_ENGINE_INTERVAL_SECONDS = 30.0
+
+
+class _ModelData:
+    def __init__(self):
+        self.model_id = ""
+        self.namespace = ""
+        self.replica_metrics = []
+        self.deployments: Dict[str, Deployment] = {}
+        self.variant_autoscalings: Dict[str, VariantAutoscaling] = {}
+        self.variant_costs: Dict[str, float] = {}
+        self.variant_states: List[VariantReplicaState] = []
+
+
+def get_deployment_gpus_per_replica(deploy: Optional[Deployment]) -> int:
+    """Sum GPU requests over containers for amd/nvidia/intel vendors;
+    default 1 (engine.go:564-585)."""
+    if deploy is None:
+        return 1
+    total = 0
+    for container in deploy.template.containers:
+        for vendor in GPU_VENDORS:
+            v = container.requests.get(vendor + GPU_RESOURCE_SUFFIX)
+            if v:
+                try:
+                    total += int(v)
+                except ValueError:
+                    pass
+    return total if total > 0 else 1
+
+
+class SaturationEngine:
+    def __init__(
+        self,
+        cluster: FakeCluster,
+        config: Config,
+        collector: ReplicaMetricsCollector,
+        enforcer: Enforcer,
+        actuator: Actuator,
+        decision_cache: DecisionCache,
+        decision_trigger: DecisionTrigger,
+        limiter: Optional[DefaultLimiter] = None,
+        capacity_store: Optional[CapacityKnowledgeStore] = None,
+        interval_seconds: float = DEFAULT_ENGINE_INTERVAL_SECONDS,
+    ):
+        self.cluster = cluster
+        self.config = config
+        self.collector = collector
+        self.enforcer = enforcer
+        self.actuator = actuator
+        self.decision_cache = decision_cache
+        self.decision_trigger = decision_trigger
+        self.limiter = limiter
+        self.capacity_store = (
+            capacity_store if capacity_store is not None else CapacityKnowledgeStore()
+        )
+        self.v1_analyzer = SaturationAnalyzerV1()
+        self.v2_analyzer = SaturationAnalyzerV2(self.capacity_store)
+        self.optimizer = CostAwareOptimizer()
+        self.executor = PollingExecutor(
+            interval_seconds, self.optimize, name="saturation-engine"
+        )
+
+    # --- lifecycle ---
+
+    def start(self) -> None:
+        self.executor.start()
+
+    def stop(self) -> None:
+        self.executor.stop()
+
+    # --- per-tick optimization ---
+
+    def optimize(self) -> None:
+        active_vas = active_variant_autoscalings(self.cluster)
+        if not active_vas:
+            log.debug("no active VariantAutoscalings found, skipping optimization")
+            return
+
+        if self.config.limited_mode_enabled() and self.limiter is not None:
+            # Limited mode: refresh inventory for observability (engine.go:203-212)
+            try:
+                self.limiter.inventory.refresh_all()
+            except Exception as e:  # noqa: BLE001
+                log.error("failed to collect cluster inventory: %s", e)
+                return
+
+        model_groups = group_variant_autoscaling_by_model(active_vas)
+        va_map = {f"{va.namespace}/{va.name}": va for va in active_vas}
+
+        sat_cfg = self.config.saturation_config().apply_defaults()
+        use_v2 = sat_cfg.analyzer_name == "saturation"
+
+        if use_v2:
+            all_decisions = self._optimize_v2(model_groups)
+        else:
+            all_decisions = self._optimize_v1(model_groups)
+
+        self.apply_saturation_decisions(all_decisions, va_map)
+
+    # --- V1 path ---
+
+    def _optimize_v1(
+        self, model_groups: Dict[str, List[VariantAutoscaling]]
+    ) -> List[VariantDecision]:
+        all_decisions: List[VariantDecision] = []
+        for _, model_vas in model_groups.items():
+            model_id = model_vas[0].spec.model_id
+            namespace = model_vas[0].namespace
+            saturation_config = self.config.saturation_config_for_namespace(
+                namespace
+            ).for_model(model_id, namespace)
+
+            try:
+                targets, analysis, variant_states = self.run_saturation_analysis(
+                    model_id, model_vas, saturation_config
+                )
+            except Exception as e:  # noqa: BLE001
+                log.error("saturation analysis failed for %s: %s", model_id, e)
+                self.emit_safety_net_metrics(model_vas)
+                continue
+            if analysis is None:
+                continue
+
+            stz_config = self.config.scale_to_zero_config_for_namespace(namespace)
+            targets, scaled_to_zero = self.enforcer.enforce_policy(
+                model_id, namespace, targets, analysis.variant_analyses, stz_config
+            )
+            if scaled_to_zero:
+                log.info(
+                    "scale-to-zero enforcement applied: model=%s targets=%s",
+                    model_id,
+                    targets,
+                )
+            all_decisions.extend(
+                self._convert_targets_to_decisions(targets, analysis, variant_states)
+            )
+
+        sat_cfg = self.config.saturation_config()
+        if sat_cfg.enable_limiter and self.limiter is not None and all_decisions:
+            try:
+                self.limiter.limit(all_decisions)
+            except Exception as e:  # noqa: BLE001
+                log.error("GPU limiter failed, proceeding unlimited: %s", e)
+        return all_decisions
+
+    def run_saturation_analysis(
+        self,
+        model_id: str,
+        model_vas: List[VariantAutoscaling],
+        saturation_config,
+    ):
+        saturation_config.apply_defaults()
+        data = self._prepare_model_data(model_id, model_vas)
+        if data is None:
+            return None, None, None
+        analysis = self.v1_analyzer.analyze_model_saturation(
+            model_id, data.namespace, data.replica_metrics, saturation_config
+        )
+        targets = self.v1_analyzer.calculate_saturation_targets(
+            analysis, data.variant_states
+        )
+        return targets, analysis, data.variant_states
+
+    # --- V2 path ---
+
+    def _optimize_v2(
+        self, model_groups: Dict[str, List[VariantAutoscaling]]
+    ) -> List[VariantDecision]:
+        requests: List[ModelScalingRequest] = []
+        for _, model_vas in model_groups.items():
+            model_id = model_vas[0].spec.model_id
+            namespace = model_vas[0].namespace
+            saturation_config = self.config.saturation_config_for_namespace(
+                namespace
+            ).for_model(model_id, namespace)
+            saturation_config.apply_defaults()
+
+            try:
+                data = self._prepare_model_data(model_id, model_vas)
+            except Exception as e:  # noqa: BLE001
+                log.error("model data preparation failed for %s: %s", model_id, e)
+                self.emit_safety_net_metrics(model_vas)
+                continue
+            if data is None:
+                continue
+
+            try:
+                result = self._run_v2_analysis(
+                    model_id, namespace, data, saturation_config
+                )
+            except Exception as e:  # noqa: BLE001
+                log.error("V2 analysis failed for %s: %s", model_id, e)
+                self.emit_safety_net_metrics(model_vas)
+                continue
+
+            requests.append(
+                ModelScalingRequest(
+                    model_id=model_id,
+                    namespace=namespace,
+                    result=result,
+                    variant_states=data.variant_states,
+                )
+            )
+
+        if not requests:
+            return []
+
+        all_decisions = self.optimizer.optimize(requests, None)
+
+        # Enforcer bridge per model
+        for req in requests:
+            stz_config = self.config.scale_to_zero_config_for_namespace(req.namespace)
+            targets = {
+                d.variant_name: d.target_replicas
+                for d in all_decisions
+                if d.model_id == req.model_id and d.namespace == req.namespace
+            }
+            analyses = [
+                VariantSaturationAnalysis(
+                    variant_name=d.variant_name,
+                    accelerator_name=d.accelerator_name,
+                    cost=d.cost,
+                    replica_count=d.current_replicas,
+                )
+                for d in all_decisions
+                if d.model_id == req.model_id and d.namespace == req.namespace
+            ]
+            enforced, scaled_to_zero = self.enforcer.enforce_policy(
+                req.model_id, req.namespace, targets, analyses, stz_config
+            )
+            if scaled_to_zero:
+                log.info(
+                    "scale-to-zero enforcement applied (V2): model=%s targets=%s",
+                    req.model_id,
+                    enforced,
+                )
+            for d in all_decisions:
+                if d.model_id != req.model_id or d.namespace != req.namespace:
+                    continue
+                new_target = enforced.get(d.variant_name)
+                if new_target is not None and new_target != d.target_replicas:
+                    d.target_replicas = new_target
+                    if new_target > d.current_replicas:
+                        d.action = ACTION_SCALE_UP
+                    elif new_target < d.current_replicas:
+                        d.action = ACTION_SCALE_DOWN
+                    else:
+                        d.action = ACTION_NO_CHANGE
+                    d.reason = (
+                        f"V2 {d.action} (optimizer: {self.optimizer.name()}, enforced)"
+                    )
+        return all_decisions
+
+    def _run_v2_analysis(self, model_id, namespace, data: _ModelData, config):
+        # Pre-populate the capacity store from deployment args
+        for key, va in data.variant_autoscalings.items():
+            deploy = data.deployments.get(
+                f"{va.namespace}/{va.get_scale_target_name()}"
+            )
+            if deploy is None:
+                continue
+            accelerator = va.metadata.labels.get(ACCELERATOR_LABEL_KEY, "")
+            gpu_count = get_deployment_gpus_per_replica(deploy)
+            self.capacity_store.load_from_deployment(
+                namespace, model_id, va.name, accelerator, gpu_count, deploy
+            )
+        scheduler_queue = self.collector.collect_scheduler_queue_metrics(model_id)
+        return self.v2_analyzer.analyze(
+            AnalyzerInput(
+                model_id=model_id,
+                namespace=namespace,
+                replica_metrics=data.replica_metrics,
+                variant_states=data.variant_states,
+                config=config,
+                scheduler_queue=scheduler_queue,
+            )
+        )
+
+    # --- shared data prep ---
+
+    def _prepare_model_data(
+        self, model_id: str, model_vas: List[VariantAutoscaling]
+    ) -> Optional[_ModelData]:
+        if not model_vas:
+            raise ValueError(f"no VAs provided for model {model_id}")
+        data = _ModelData()
+        data.model_id = model_id
+        data.namespace = model_vas[0].namespace
+
+        for va in model_vas:
+            deploy = self.cluster.try_get(
+                "Deployment", va.namespace, va.get_scale_target_name()
+            )
+            if deploy is None:
+                continue
+            deploy_key = f"{va.namespace}/{va.get_scale_target_name()}"
+            variant_key = f"{va.namespace}/{va.name}"
+            data.deployments[deploy_key] = deploy
+            data.variant_autoscalings[variant_key] = va
+            data.variant_costs[variant_key] = va.spec.cost()
+
+        data.replica_metrics = self.collector.collect_replica_metrics(
+            model_id,
+            data.namespace,
+            data.deployments,
+            data.variant_autoscalings,
+            data.variant_costs,
+        )
+        if not data.replica_metrics:
+            log.info(
+                "no saturation metrics available for model %s/%s, skipping",
+                data.namespace,
+                model_id,
+            )
+            return None
+
+        data.variant_states = self.build_variant_states(model_vas, data.deployments)
+        return data
+
+    def build_variant_states(
+        self,
+        vas: List[VariantAutoscaling],
+        deployments: Dict[str, Deployment],
+    ) -> List[VariantReplicaState]:
+        states = []
+        for va in vas:
+            key = f"{va.namespace}/{va.get_scale_target_name()}"
+            deploy = deployments.get(key)
+            if deploy is None:
+                deploy = self.cluster.try_get(
+                    "Deployment", va.namespace, va.get_scale_target_name()
+                )
+                if deploy is None:
+                    continue
+            current = deploy.status.replicas
+            if current == 0 and deploy.replicas is not None:
+                current = deploy.replicas
+            pending = max(current - deploy.status.ready_replicas, 0)
+            states.append(
+                VariantReplicaState(
+                    variant_name=va.name,
+                    current_replicas=current,
+                    desired_replicas=va.status.desired_optimized_alloc.num_replicas,
+                    pending_replicas=pending,
+                    gpus_per_replica=get_deployment_gpus_per_replica(deploy),
+                )
+            )
+        return states
+
+    def _convert_targets_to_decisions(
+        self,
+        targets: Dict[str, int],
+        analysis: ModelSaturationAnalysis,
+        variant_states: List[VariantReplicaState],
+    ) -> List[VariantDecision]:
+        va_map = {va.variant_name: va for va in analysis.variant_analyses}
+        state_map = {s.variant_name: s for s in variant_states}
+        decisions = []
+        for variant_name, target in targets.items():
+            state = state_map.get(
+                variant_name, VariantReplicaState(variant_name=variant_name)
+            )
+            if target > state.current_replicas:
+                action = ACTION_SCALE_UP
+            elif target < state.current_replicas:
+                action = ACTION_SCALE_DOWN
+            else:
+                action = ACTION_NO_CHANGE
+            d = VariantDecision(
+                variant_name=variant_name,
+                namespace=analysis.namespace,
+                model_id=analysis.model_id,
+                current_replicas=state.current_replicas,
+                target_replicas=target,
+                original_target_replicas=target,
+                desired_replicas=state.desired_replicas,
+                action=action,
+                saturation_based=True,
+                saturation_only=True,
+                reason=f"saturation-only mode: {action}",
+                gpus_per_replica=max(state.gpus_per_replica, 1),
+            )
+            va = va_map.get(variant_name)
+            if va is not None:
+                d.accelerator_name = va.accelerator_name
+                d.cost = va.cost
+                d.spare_capacity = va.avg_spare_kv_capacity
+            decisions.append(d)
+        return decisions
+
+    # --- decision application ---
+
+    def apply_saturation_decisions(
+        self,
+        decisions: List[VariantDecision],
+        va_map: Dict[str, VariantAutoscaling],
+    ) -> None:
+        decision_map = {f"{d.namespace}/{d.variant_name}": d for d in decisions}
+
+        for va_key, va in va_map.items():
+            decision = decision_map.get(va_key)
+
+            update_va = self.cluster.try_get(
+                "VariantAutoscaling", va.namespace, va.name
+            )
+            if update_va is None:
+                log.error("failed to get latest VA %s", va_key)
+                continue
+
+            if decision is not None:
+                target_replicas = decision.target_replicas
+                accelerator_name = decision.accelerator_name
+                reason = decision.reason
+            else:
+                target_replicas = (
+                    update_va.status.desired_optimized_alloc.num_replicas
+                    if update_va.status.desired_optimized_alloc.num_replicas > 0
+                    else 0
+                )
+                accelerator_name = update_va.status.desired_optimized_alloc.accelerator
+                reason = "No scaling decision (optimization loop)"
+
+            if not accelerator_name:
+                # Fallback: VA accelerator label (keeps MI355X-labeled VAs
+                # emitting metrics even before first metrics arrive)
+                accelerator_name = va.metadata.labels.get(ACCELERATOR_LABEL_KEY, "")
+
+            if not accelerator_name:
+                log.info(
+                    "skipping status update for %s (no accelerator info); "
+                    "setting MetricsAvailable=False",
+                    va_key,
+                )
+                self.decision_cache.set(
+                    va.namespace,
+                    va.name,
+                    VariantDecision(
+                        variant_name=va_key,
+                        namespace=va.namespace,
+                        metrics_available=False,
+                        metrics_reason=METRICS_REASON_UNAVAILABLE,
+                        metrics_message=METRICS_MESSAGE_UNAVAILABLE,
+                    ),
+                )
+                self.decision_trigger.push(va.namespace, va.name)
+                continue
+
+            update_va.status.desired_optimized_alloc = OptimizedAlloc(
+                last_run_time=utcnow(),
+                accelerator=accelerator_name,
+                num_replicas=target_replicas,
+            )
+            update_va.status.actuation.applied = False
+
+            if decision is not None:
+                if decision.safety_override:
+                    opt_reason = "SaturationSafetyOverride"
+                    opt_message = f"saturation safety override: {reason}"
+                elif decision.saturation_only:
+                    opt_reason = "SaturationOnlyMode"
+                    opt_message = (
+                        f"saturation-only decision: {reason} "
+                        f"(target: {target_replicas} replicas)"
+                    )
+                else:
+                    opt_reason = "OptimizationSucceeded"
+                    opt_message = (
+                        f"Hybrid mode: {reason} (target: {target_replicas} replicas)"
+                    )
+            else:
+                opt_reason = "OptimizationSucceeded"
+                opt_message = "Optimization loop ran (no scaling change needed)"
+            cond.set_condition(
+                update_va, "OptimizationReady", "True", opt_reason, opt_message
+            )
+
+            # Emit metrics for external autoscalers (HPA must see a signal
+            # every tick)
+            try:
+                self.actuator.emit_metrics(update_va)
+                update_va.status.actuation.applied = True
+            except Exception as e:  # noqa: BLE001
+                log.error("failed to emit metrics for %s: %s", va_key, e)
+
+            metrics_available = decision is not None
+            self.decision_cache.set(
+                va.namespace,
+                va.name,
+                VariantDecision(
+                    variant_name=va_key,
+                    namespace=va.namespace,
+                    target_replicas=target_replicas,
+                    accelerator_name=accelerator_name,
+                    last_run_time=utcnow(),
+                    metrics_available=metrics_available,
+                    metrics_reason=(
+                        METRICS_REASON_AVAILABLE
+                        if metrics_available
+                        else METRICS_REASON_UNAVAILABLE
+                    ),
+                    metrics_message=(
+                        METRICS_MESSAGE_AVAILABLE
+                        if metrics_available
+                        else METRICS_MESSAGE_UNAVAILABLE
+                    ),
+                    optimization_ready_reason=opt_reason,
+                    optimization_ready_message=opt_message,
+                ),
+            )
+            self.decision_trigger.push(va.namespace, va.name)
+
+    # --- safety net ---
+
+    def emit_safety_net_metrics(self, model_vas: List[VariantAutoscaling]) -> None:
+        """On analysis failure, keep the HPA fed: previous desired (or
+        current) replicas, accelerator from status → labels."""
+        for va in model_vas:
+            try:
+                current = self.actuator.get_current_deployment_replicas(va)
+            except Exception:  # noqa: BLE001
+                current = 0
+            if va.status.desired_optimized_alloc.num_replicas > 0:
+                desired = va.status.desired_optimized_alloc.num_replicas
+            else:
+                desired = current
+            accelerator = va.status.desired_optimized_alloc.accelerator
+            if not accelerator:
+                accelerator = va.metadata.labels.get(ACCELERATOR_LABEL_KEY, "")
+            if not accelerator:
+                log.info(
+                    "safety net: skipping %s — no accelerator name available",
+                    va.full_name(),
+                )
+                continue
+            try:
+                self.actuator.emitter.emit_replica_metrics(
+                    va.name, va.namespace, current, desired, accelerator
+                )
+                log.info(
+                    "safety net activated for %s: current=%d desired=%d",
+                    va.full_name(),
+                    current,
+                    desired,
+                )
+            except Exception as e:  # noqa: BLE001
+                log.error("safety net emission failed for %s: %s", va.full_name(), e)
