@@ -1,0 +1,91 @@
+"""CPU tests of the fp32 reference ops (shape/semantics sanity: these are
+the ground truth that the GPU kernels are compared against)."""
+import math
+
+import torch
+
+from wva_amd import ops
+
+
+class TestReferenceOps:
+    def test_rmsnorm_unit_weight_norms(self):
+        x = torch.randn(4, 64)
+        w = torch.ones(64)
+        out, _ = ops.rmsnorm_ref(x, w, None, 0.0)
+        rms = out.pow(2).mean(dim=-1)
+        torch.testing.assert_close(rms, torch.ones(4), atol=1e-5, rtol=1e-5)
+
+    def test_rmsnorm_residual_fold(self):
+        x = torch.randn(2, 8)
+        r = torch.randn(2, 8)
+        _, folded = ops.rmsnorm_ref(x, torch.ones(8), r)
+        torch.testing.assert_close(folded, x + r)
+
+    def test_rope_preserves_norm(self):
+        q = torch.randn(3, 2, 128)
+        k = torch.randn(3, 1, 128)
+        pos = torch.tensor([0, 5, 100])
+        qr, kr = ops.rope_ref(q, k, pos)
+        # rotation preserves the norm of each (x1, x2) pair
+        torch.testing.assert_close(
+            qr.norm(dim=-1), q.float().norm(dim=-1), atol=1e-4, rtol=1e-4
+        )
+
+    def test_silu_mul(self):
+        g = torch.tensor([[0.0, 1.0]])
+        u = torch.tensor([[3.0, 2.0]])
+        out = ops.silu_mul_ref(g, u)
+        assert out[0, 0] == 0.0
+        expected = 1.0 / (1.0 + math.exp(-1.0)) * 2.0
+        assert abs(out[0, 1].item() - expected) < 1e-6
+
+    def test_attention_uniform_v(self):
+        # all V rows identical → output equals that row regardless of scores
+        B, S, Hk, Hq, D = 1, 16, 2, 4, 128
+        q = torch.randn(B, Hq, D)
+        k = torch.randn(B, S, Hk, D)
+        v = torch.ones(B, S, Hk, D) * 0.5
+        lens = torch.tensor([10], dtype=torch.int32)
+        out = ops.gqa_decode_attn_ref(q, k, v, lens, 1.0 / math.sqrt(D))
+        torch.testing.assert_close(
+            out, torch.full((B, Hq, D), 0.5), atol=1e-5, rtol=1e-5
+        )
+
+    def test_attention_respects_context_len(self):
+        B, S, Hk, Hq, D = 1, 8, 1, 1, 128
+        q = torch.zeros(B, Hq, D)
+        q[0, 0, 0] = 10.0
+        k = torch.zeros(B, S, Hk, D)
+        v = torch.zeros(B, S, Hk, D)
+        # position 5 (beyond ctx=4) has huge value — must not leak
+        k[0, 5, 0, 0] = 100.0
+        v[0, 5, 0, :] = 999.0
+        lens = torch.tensor([4], dtype=torch.int32)
+        out = ops.gqa_decode_attn_ref(q, k, v, lens, 1.0)
+        assert out.abs().max() < 1.0
+
+    def test_cpu_dispatch_uses_reference(self):
+        x = torch.randn(2, 64, dtype=torch.bfloat16)
+        w = torch.ones(64, dtype=torch.bfloat16)
+        out = ops.rmsnorm(x, w)
+        assert out.dtype == torch.bfloat16
+
+    def test_fit_itl_curve(self):
+        from wva_amd.calibration.itl_benchmark import fit_itl_curve
+
+        # perfect linear data
+        alpha, beta, r2 = fit_itl_curve([1, 2, 4, 8], [10.5, 11.0, 12.0, 14.0])
+        assert abs(alpha - 10.0) < 1e-9
+        assert abs(beta - 0.5) < 1e-9
+        assert r2 > 0.999
+
+    def test_kv_capacity_288gb(self):
+        from wva_amd.calibration.itl_benchmark import derive_kv_capacity
+        from wva_amd.calibration.model import LLAMA_3_8B
+
+        hbm = 288 * 1024**3
+        blocks, tokens = derive_kv_capacity(LLAMA_3_8B, hbm_bytes=hbm)
+        # 8B bf16 weights ≈ 16 GB + embeddings; KV/token = 2*32*1024*2 = 128 KiB
+        # → ≈ (0.9*288GB − ~18GB) / 128KiB ≈ 1.9M tokens
+        assert 1_500_000 < tokens < 2_200_000
+        assert blocks == tokens // 16

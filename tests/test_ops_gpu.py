@@ -1,0 +1,138 @@
+"""GPU numerics tests: each HIP kernel vs the plain fp32 torch reference.
+
+All marked @pytest.mark.gpu — run on an MI355X box via gpurun / the
+driver's round-end pass. The HIP extension must be loaded (fail-loud: no
+eager fallback on GPU).
+"""
+import math
+
+import pytest
+import torch
+
+from wva_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+B, H = 16, 4096
+
+
+def tol(ref):
+    # bf16 output tolerance relative to fp32 reference
+    return dict(atol=2e-2, rtol=2e-2)
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    # The GPU path must use the in-tree HIP extension — no fallback.
+    assert ops.extension_available(), "HIP extension _wva_ops not loaded"
+    return torch.device("cuda:0")
+
+
+class TestRMSNorm:
+    def test_plain(self, dev):
+        x = torch.randn(B, H, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(H, device=dev, dtype=torch.bfloat16)
+        out = ops.rmsnorm(x, w, None, 1e-5)
+        ref, _ = ops.rmsnorm_ref(x, w, None, 1e-5)
+        torch.testing.assert_close(out.float(), ref, **tol(ref))
+
+    def test_fused_residual(self, dev):
+        x = torch.randn(B, H, device=dev, dtype=torch.bfloat16)
+        res = torch.randn(B, H, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(H, device=dev, dtype=torch.bfloat16)
+        res_before = res.clone()
+        out = ops.rmsnorm(x, w, res, 1e-5)
+        ref, folded = ops.rmsnorm_ref(x, w, res_before, 1e-5)
+        torch.testing.assert_close(out.float(), ref, **tol(ref))
+        # residual updated in place to x + residual
+        torch.testing.assert_close(res.float(), folded, **tol(folded))
+
+    def test_odd_rows(self, dev):
+        x = torch.randn(3, 2048, device=dev, dtype=torch.bfloat16)
+        w = torch.ones(2048, device=dev, dtype=torch.bfloat16)
+        out = ops.rmsnorm(x, w)
+        ref, _ = ops.rmsnorm_ref(x, w)
+        torch.testing.assert_close(out.float(), ref, **tol(ref))
+
+
+class TestRope:
+    def test_matches_reference(self, dev):
+        T, Hq, Hk, D = 9, 32, 8, 128
+        q = torch.randn(T, Hq, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(T, Hk, D, device=dev, dtype=torch.bfloat16)
+        pos = torch.randint(0, 2000, (T,), device=dev, dtype=torch.int32)
+        q_ref, k_ref = ops.rope_ref(q, k, pos, 500000.0)
+        ops.rope(q, k, pos, 500000.0)
+        torch.testing.assert_close(q.float(), q_ref, atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(k.float(), k_ref, atol=5e-2, rtol=5e-2)
+
+    def test_position_zero_identity(self, dev):
+        T, Hq, Hk, D = 4, 8, 2, 128
+        q = torch.randn(T, Hq, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(T, Hk, D, device=dev, dtype=torch.bfloat16)
+        q0, k0 = q.clone(), k.clone()
+        pos = torch.zeros(T, device=dev, dtype=torch.int32)
+        ops.rope(q, k, pos)
+        torch.testing.assert_close(q.float(), q0.float(), atol=1e-3, rtol=1e-3)
+
+
+class TestSiluMul:
+    def test_matches_reference(self, dev):
+        g = torch.randn(B, 14336, device=dev, dtype=torch.bfloat16)
+        u = torch.randn(B, 14336, device=dev, dtype=torch.bfloat16)
+        out = ops.silu_mul(g, u)
+        ref = ops.silu_mul_ref(g, u)
+        torch.testing.assert_close(out.float(), ref, **tol(ref))
+
+
+class TestDecodeAttention:
+    @pytest.mark.parametrize("batch,hq,hk,ctx", [
+        (1, 32, 8, 17),
+        (4, 32, 8, 511),
+        (2, 8, 8, 64),     # MHA (G=1)
+        (2, 64, 8, 300),   # 70B shape (G=8)
+    ])
+    def test_matches_reference(self, dev, batch, hq, hk, ctx):
+        D, S = 128, 512
+        q = torch.randn(batch, hq, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(batch, S, hk, D, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(batch, S, hk, D, device=dev, dtype=torch.bfloat16)
+        lens = torch.full((batch,), ctx, device=dev, dtype=torch.int32)
+        scale = 1.0 / math.sqrt(D)
+        out = ops.gqa_decode_attn(q, k, v, lens, scale)
+        ref = ops.gqa_decode_attn_ref(q, k, v, lens, scale)
+        torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+
+    def test_varied_context_lens(self, dev):
+        D, S, batch = 128, 256, 3
+        q = torch.randn(batch, 32, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(batch, S, 8, D, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(batch, S, 8, D, device=dev, dtype=torch.bfloat16)
+        lens = torch.tensor([1, 100, 256], device=dev, dtype=torch.int32)
+        out = ops.gqa_decode_attn(q, k, v, lens)
+        ref = ops.gqa_decode_attn_ref(q, k, v, lens, 1.0 / math.sqrt(D))
+        torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+
+
+class TestDecodeModel:
+    def test_tiny_decode_step(self, dev):
+        from wva_amd.calibration.model import TINY, LlamaDecodeModel
+
+        model = LlamaDecodeModel(TINY, max_batch=4, max_seq=64)
+        model.reset(4, 16)
+        tokens = torch.randint(0, TINY.vocab_size, (4,), device=dev)
+        logits = model.decode_step(tokens)
+        assert logits.shape == (4, TINY.vocab_size)
+        assert torch.isfinite(logits.float()).all()
+        assert int(model.context_lens[0]) == 17
+
+    def test_calibration_fit(self, dev):
+        from wva_amd.calibration.itl_benchmark import calibrate_service_profile
+        from wva_amd.calibration.model import TINY
+
+        profile, result = calibrate_service_profile(
+            TINY, batch_sizes=[1, 4, 8], context_len=16, max_seq=64, iters=3
+        )
+        assert result.alpha_ms > 0
+        assert profile.num_gpu_blocks > 0

@@ -1,0 +1,151 @@
+"""ITL-vs-batch calibration on MI355X.
+
+Measures the decode iteration time of a LlamaDecodeModel at a sweep of
+batch sizes, fits ITL(batch) = α + β·batch by least squares (the Inferno
+ServiceParms linear model, reference docs/design/modeling-optimization.md
+:52), derives the KV-token capacity from the GPU's free HBM, and emits a
+ServiceProfile for the emulator plus a JSON record for profiles/.
+
+This is the reference's offline "parameter estimation" methodology
+(docs/tutorials/parameter-estimation.md) turned into an in-repo measured
+path on the actual target hardware.
+"""
+from __future__ import annotations
+
+import json
+import time
+from dataclasses import asdict, dataclass
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from ..emulator.vllm_sim import ServiceProfile
+from .model import LlamaConfig, LlamaDecodeModel
+
+
+@dataclass
+class CalibrationResult:
+    model: str
+    device: str
+    gpu_count: int
+    alpha_ms: float
+    beta_ms: float
+    batch_sizes: List[int]
+    itl_ms: List[float]
+    r_squared: float
+    kv_capacity_tokens: int
+    num_gpu_blocks: int
+    block_size: int
+    decode_tokens_per_s_peak: float
+
+    def to_json(self) -> str:
+        return json.dumps(asdict(self), indent=2)
+
+
+def fit_itl_curve(
+    batch_sizes: List[int], itl_ms: List[float]
+) -> Tuple[float, float, float]:
+    """Least-squares fit ITL = alpha + beta * batch; returns (a, b, R²)."""
+    import numpy as np
+
+    x = np.asarray(batch_sizes, dtype=np.float64)
+    y = np.asarray(itl_ms, dtype=np.float64)
+    A = np.stack([np.ones_like(x), x], axis=1)
+    coef, *_ = np.linalg.lstsq(A, y, rcond=None)
+    alpha, beta = float(coef[0]), float(coef[1])
+    pred = alpha + beta * x
+    ss_res = float(((y - pred) ** 2).sum())
+    ss_tot = float(((y - y.mean()) ** 2).sum())
+    r2 = 1.0 - ss_res / ss_tot if ss_tot > 0 else 1.0
+    return alpha, beta, r2
+
+
+def measure_itl(
+    model: LlamaDecodeModel,
+    batch: int,
+    context_len: int = 512,
+    iters: int = 8,
+    warmup: int = 3,
+) -> float:
+    """Median decode-iteration wall time (ms) at the given batch size."""
+    model.reset(batch, context_len)
+    tokens = torch.randint(
+        0, model.cfg.vocab_size, (batch,), device=model.device
+    )
+    for _ in range(warmup):
+        model.decode_step(tokens)
+    torch.cuda.synchronize()
+    times = []
+    for _ in range(iters):
+        t0 = time.perf_counter()
+        model.decode_step(tokens)
+        torch.cuda.synchronize()
+        times.append((time.perf_counter() - t0) * 1000.0)
+    times.sort()
+    return times[len(times) // 2]
+
+
+def derive_kv_capacity(
+    cfg: LlamaConfig,
+    gpu_memory_utilization: float = 0.9,
+    block_size: int = 16,
+    hbm_bytes: Optional[int] = None,
+) -> Tuple[int, int]:
+    """(num_gpu_blocks, kv_capacity_tokens) from free HBM after weights —
+    the vllm:cache_config_info analog, sized for 288 GB HBM3E."""
+    if hbm_bytes is None:
+        hbm_bytes = torch.cuda.get_device_properties(0).total_memory
+    budget = int(hbm_bytes * gpu_memory_utilization) - cfg.weight_bytes()
+    per_token = cfg.kv_bytes_per_token()
+    tokens = max(budget // per_token, 0)
+    blocks = tokens // block_size
+    return blocks, blocks * block_size
+
+
+def calibrate_service_profile(
+    cfg: LlamaConfig,
+    batch_sizes: Optional[List[int]] = None,
+    context_len: int = 512,
+    max_seq: int = 1024,
+    gpu_count: int = 1,
+    iters: int = 8,
+) -> Tuple[ServiceProfile, CalibrationResult]:
+    """Measure on the current GPU and return an emulator ServiceProfile
+    + the raw calibration record."""
+    if batch_sizes is None:
+        batch_sizes = [1, 2, 4, 8, 16, 32, 64]
+    max_batch = max(batch_sizes)
+    model = LlamaDecodeModel(cfg, max_batch=max_batch, max_seq=max_seq)
+    itl: List[float] = []
+    for b in batch_sizes:
+        itl.append(measure_itl(model, b, context_len, iters=iters))
+    alpha, beta, r2 = fit_itl_curve(batch_sizes, itl)
+    blocks, kv_tokens = derive_kv_capacity(cfg)
+    peak_tps = max(
+        b / (t / 1000.0) for b, t in zip(batch_sizes, itl)
+    )
+    result = CalibrationResult(
+        model=cfg.name,
+        device=torch.cuda.get_device_name(0),
+        gpu_count=gpu_count,
+        alpha_ms=alpha,
+        beta_ms=beta,
+        batch_sizes=batch_sizes,
+        itl_ms=itl,
+        r_squared=r2,
+        kv_capacity_tokens=kv_tokens,
+        num_gpu_blocks=blocks,
+        block_size=16,
+        decode_tokens_per_s_peak=peak_tps,
+    )
+    profile = ServiceProfile(
+        alpha_ms=max(alpha, 0.1),
+        beta_ms=max(beta, 0.0),
+        max_num_seqs=256,
+        num_gpu_blocks=blocks,
+        block_size=16,
+    )
+    # free the model before returning (bench reuses the GPU)
+    del model
+    torch.cuda.empty_cache()
+    return profile, result

@@ -1,0 +1,217 @@
+"""MI355X-native Llama-family decode engine for capacity calibration.
+
+The reference's capacity model runs on parameters fit OFFLINE from vLLM
+benchmarks on NVIDIA/MI300X hardware (docs/design/modeling-optimization.md
+:46-84). This module is the MI355X-native replacement: a real bf16 decode
+step of the target architecture (random-init weights — no network for
+checkpoints) built from this repo's HIP/CDNA4 kernels (wva_amd.ops) with
+GEMMs on hipBLASLt via torch.matmul, used to MEASURE ITL-vs-batch curves
+and KV capacity on the actual GPU.
+
+Single-GPU by design: TP over RCCL/xGMI shards these same GEMMs; the
+autoscaler consumes per-(model, gpuCount) profiles, measured per TP degree
+(capacity_store keys records by gpu_count).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+
+from .. import ops
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama"
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_q_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    vocab_size: int = 128256
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+
+    @property
+    def q_size(self) -> int:
+        return self.num_q_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.num_kv_heads * self.head_dim
+
+    def kv_bytes_per_token(self, dtype_bytes: int = 2) -> int:
+        return 2 * self.num_layers * self.kv_size * dtype_bytes
+
+    def weight_bytes(self, dtype_bytes: int = 2) -> int:
+        per_layer = (
+            self.hidden_size * (self.q_size + 2 * self.kv_size)  # qkv
+            + self.q_size * self.hidden_size  # o
+            + 3 * self.hidden_size * self.intermediate_size  # gate/up/down
+            + 2 * self.hidden_size  # norms
+        )
+        return dtype_bytes * (
+            self.num_layers * per_layer
+            + 2 * self.vocab_size * self.hidden_size  # embed + lm_head
+            + self.hidden_size
+        )
+
+
+LLAMA_3_8B = LlamaConfig(
+    name="meta-llama/Llama-3.1-8B",
+    hidden_size=4096,
+    intermediate_size=14336,
+    num_layers=32,
+    num_q_heads=32,
+    num_kv_heads=8,
+)
+
+LLAMA_3_70B = LlamaConfig(
+    name="meta-llama/Llama-3.1-70B",
+    hidden_size=8192,
+    intermediate_size=28672,
+    num_layers=80,
+    num_q_heads=64,
+    num_kv_heads=8,
+)
+
+# Tiny config for smoke tests (same topology, minutes → milliseconds)
+TINY = LlamaConfig(
+    name="tiny-llama",
+    hidden_size=1024,
+    intermediate_size=2816,
+    num_layers=2,
+    num_q_heads=8,
+    num_kv_heads=2,
+    vocab_size=32000,
+)
+
+
+class _DecoderLayer:
+    def __init__(self, cfg: LlamaConfig, device, dtype, gen):
+        h, std = cfg.hidden_size, 0.02
+        def w(rows, cols):
+            return torch.empty(rows, cols, device=device, dtype=dtype).normal_(
+                0.0, std, generator=gen
+            )
+
+        self.input_norm = torch.ones(h, device=device, dtype=dtype)
+        self.post_attn_norm = torch.ones(h, device=device, dtype=dtype)
+        self.wqkv = w(cfg.q_size + 2 * cfg.kv_size, h)
+        self.wo = w(h, cfg.q_size)
+        self.w_gate_up = w(2 * cfg.intermediate_size, h)
+        self.w_down = w(h, cfg.intermediate_size)
+
+
+class LlamaDecodeModel:
+    """Decode-only engine with a contiguous KV cache [B, S, Hk, D]."""
+
+    def __init__(
+        self,
+        cfg: LlamaConfig,
+        max_batch: int = 256,
+        max_seq: int = 2048,
+        device: str = "cuda",
+        seed: int = 0,
+    ):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        self.dtype = torch.bfloat16
+        self.max_batch = max_batch
+        self.max_seq = max_seq
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed)
+
+        self.embed = torch.empty(
+            cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=self.dtype
+        ).normal_(0.0, 0.02, generator=gen)
+        self.layers: List[_DecoderLayer] = [
+            _DecoderLayer(cfg, self.device, self.dtype, gen)
+            for _ in range(cfg.num_layers)
+        ]
+        self.final_norm = torch.ones(
+            cfg.hidden_size, device=self.device, dtype=self.dtype
+        )
+        self.lm_head = torch.empty(
+            cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=self.dtype
+        ).normal_(0.0, 0.02, generator=gen)
+
+        # Contiguous KV cache per layer: [B, S, Hk, D]
+        self.k_cache = [
+            torch.zeros(
+                max_batch, max_seq, cfg.num_kv_heads, cfg.head_dim,
+                device=self.device, dtype=self.dtype,
+            )
+            for _ in range(cfg.num_layers)
+        ]
+        self.v_cache = [torch.zeros_like(self.k_cache[0]) for _ in range(cfg.num_layers)]
+        self.context_lens = torch.zeros(
+            max_batch, dtype=torch.int32, device=self.device
+        )
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+
+    def reset(self, batch: int, context_len: int) -> None:
+        """Random-fill KV up to context_len for `batch` sequences."""
+        self.context_lens.zero_()
+        self.context_lens[:batch] = context_len
+        for layer in range(self.cfg.num_layers):
+            self.k_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
+            self.v_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
+
+    @torch.no_grad()
+    def decode_step(self, token_ids: torch.Tensor) -> torch.Tensor:
+        """One decode iteration for `B = len(token_ids)` sequences.
+        Appends each sequence's new KV at position context_lens[b] and
+        returns logits [B, vocab]."""
+        cfg = self.cfg
+        B = token_ids.shape[0]
+        positions = self.context_lens[:B].clone()
+
+        x = self.embed.index_select(0, token_ids)  # [B, H]
+        residual: Optional[torch.Tensor] = None
+
+        batch_idx = torch.arange(B, device=self.device)
+        for li, layer in enumerate(self.layers):
+            if residual is None:
+                residual = x.clone()
+                h = ops.rmsnorm(x, layer.input_norm, None, cfg.rms_eps)
+            else:
+                h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
+
+            qkv = h @ layer.wqkv.t()  # hipBLASLt GEMM
+            q = qkv[:, : cfg.q_size].reshape(B, cfg.num_q_heads, cfg.head_dim)
+            k = qkv[:, cfg.q_size : cfg.q_size + cfg.kv_size].reshape(
+                B, cfg.num_kv_heads, cfg.head_dim
+            )
+            v = qkv[:, cfg.q_size + cfg.kv_size :].reshape(
+                B, cfg.num_kv_heads, cfg.head_dim
+            )
+            q = q.contiguous()
+            k = k.contiguous()
+            ops.rope(q, k, positions, cfg.rope_theta)
+
+            # append KV at the current position
+            self.k_cache[li][batch_idx, positions.long()] = k
+            self.v_cache[li][batch_idx, positions.long()] = v
+
+            ctx = positions + 1  # includes the new token
+            attn = ops.gqa_decode_attn(
+                q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
+            )
+            x = attn.reshape(B, cfg.q_size) @ layer.wo.t()
+
+            h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
+            gate_up = h2 @ layer.w_gate_up.t()
+            gate = gate_up[:, : cfg.intermediate_size].contiguous()
+            up = gate_up[:, cfg.intermediate_size :].contiguous()
+            act = ops.silu_mul(gate, up)
+            x = act @ layer.w_down.t()
+
+        final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)
+        logits = final @ self.lm_head.t()
+        self.context_lens[:B] += 1
+        return logits

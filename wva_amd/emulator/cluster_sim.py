@@ -45,9 +45,13 @@ class ClusterSim:
         cluster: FakeCluster,
         pod_ready_delay_s: float = 30.0,
         seed: int = 0,
+        warm_start: bool = False,
     ):
         self.cluster = cluster
         self.pod_ready_delay_s = pod_ready_delay_s
+        # warm_start: pods created at sim time 0 are ready immediately
+        # (the serving stack pre-exists the benchmark window)
+        self.warm_start = warm_start
         self.now = 0.0
         self.models: Dict[str, ModelSim] = {}
         # pod name → (ReplicaSim, ready_at, deployment_name, namespace)
@@ -106,6 +110,8 @@ class ClusterSim:
                 pod_name = f"{deploy.name}-{next(self._pod_counter):05x}"
                 sim = ReplicaSim(pod_name, profile)
                 ready_at = self.now + self.pod_ready_delay_s
+                if self.warm_start and self.now == 0.0:
+                    ready_at = 0.0
                 self.replicas[pod_name] = (
                     sim, ready_at, deploy.name, deploy.namespace
                 )
@@ -170,13 +176,24 @@ class ClusterSim:
                 out.append(sim)
         return out
 
+    # EPP flow-control admission bound: a replica accepts new work only up
+    # to max_num_seqs + this many waiting requests; the rest stays in the
+    # model-level scheduler queue (inference_extension_flow_control_*).
+    # Mirrors llm-d's EPP, and keeps vllm:num_requests_waiting a truthful
+    # saturation signal instead of an unbounded dump.
+    QUEUE_ALLOWANCE = 10
+
+    def _admission_cap(self, replica: ReplicaSim) -> float:
+        return replica.profile.max_num_seqs + self.QUEUE_ALLOWANCE
+
     def submit_request(self, model: ModelSim, spec: RequestSpec) -> None:
         model.submitted += 1
         ready = self.ready_replicas_of_model(model)
-        if not ready:
+        candidates = [r for r in ready if r.load() < self._admission_cap(r)]
+        if not candidates:
             model.scheduler_queue.append(spec)
             return
-        target = min(ready, key=lambda r: r.load())
+        target = min(candidates, key=lambda r: r.load())
         target.submit(spec)
 
     def _drain_scheduler_queue(self, model: ModelSim) -> None:
@@ -184,8 +201,11 @@ class ClusterSim:
         if not ready:
             return
         while model.scheduler_queue:
+            candidates = [r for r in ready if r.load() < self._admission_cap(r)]
+            if not candidates:
+                return
             spec = model.scheduler_queue.pop(0)
-            target = min(ready, key=lambda r: r.load())
+            target = min(candidates, key=lambda r: r.load())
             target.submit(spec)
 
     def generate_arrivals(

@@ -1,0 +1,184 @@
+"""MI355X (gfx950) fused op library.
+
+Hand-written HIP/CDNA4 kernels for the calibration decode path's hot
+non-GEMM ops (GEMMs go through hipBLASLt via torch.matmul — guide rule:
+library GEMMs stay in the library). The extension is built in-tree
+(`wva_amd/ops/_wva_ops*.so`, see setup.py / __graft_entry__.build).
+
+Fail-loud contract: on a ROCm GPU the HIP extension MUST load — there is
+no silent eager fallback on the GPU path (the driver verifies the .so is
+actually loaded). Pure-CPU processes (unit tests, the control plane) use
+the fp32 torch reference implementations below, which are also the
+numerics references for the GPU tests.
+"""
+from __future__ import annotations
+
+import glob
+import importlib.util
+import math
+import os
+from typing import Optional
+
+import torch
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_ext = None
+_ext_error: Optional[str] = None
+
+
+def _try_load_extension():
+    global _ext, _ext_error
+    if _ext is not None:
+        return _ext
+    candidates = sorted(glob.glob(os.path.join(_HERE, "_wva_ops*.so")))
+    if not candidates:
+        _ext_error = (
+            f"HIP extension _wva_ops*.so not found in {_HERE}; build it with "
+            "`python -m wva_amd.ops.build` (or __graft_entry__.build())"
+        )
+        return None
+    spec = importlib.util.spec_from_file_location("_wva_ops", candidates[0])
+    mod = importlib.util.module_from_spec(spec)
+    try:
+        spec.loader.exec_module(mod)
+    except Exception as e:  # noqa: BLE001
+        _ext_error = f"failed to load {candidates[0]}: {e}"
+        return None
+    _ext = mod
+    return _ext
+
+
+def extension_available() -> bool:
+    return _try_load_extension() is not None
+
+
+def _require_ext():
+    ext = _try_load_extension()
+    if ext is None:
+        raise RuntimeError(
+            f"wva_amd.ops: GPU tensor passed but HIP extension unavailable: "
+            f"{_ext_error}"
+        )
+    return ext
+
+
+# --- fp32 torch reference implementations (CPU tests + GPU numerics refs) ---
+
+
+def rmsnorm_ref(
+    input: torch.Tensor,
+    weight: torch.Tensor,
+    residual: Optional[torch.Tensor] = None,
+    eps: float = 1e-5,
+):
+    x = input.float()
+    if residual is not None:
+        x = x + residual.float()
+    folded = x
+    rms = torch.rsqrt(x.pow(2).mean(dim=-1, keepdim=True) + eps)
+    out = x * rms * weight.float()
+    return out, folded
+
+
+def rope_ref(
+    q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
+    theta: float = 500000.0,
+):
+    """NeoX half-rotation RoPE, fp32."""
+
+    def rot(x):
+        t, h, d = x.shape
+        half = d // 2
+        xf = x.float()
+        inv_freq = theta ** (
+            -2.0 * torch.arange(half, dtype=torch.float32, device=x.device) / d
+        )
+        angle = positions.float()[:, None] * inv_freq[None, :]  # [T, half]
+        cos = angle.cos()[:, None, :]
+        sin = angle.sin()[:, None, :]
+        x1, x2 = xf[..., :half], xf[..., half:]
+        return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], dim=-1)
+
+    return rot(q), rot(k)
+
+
+def silu_mul_ref(gate: torch.Tensor, up: torch.Tensor):
+    g = gate.float()
+    return torch.nn.functional.silu(g) * up.float()
+
+
+def gqa_decode_attn_ref(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: float,
+):
+    """fp32 reference: per-sequence masked softmax attention."""
+    B, Hq, D = q.shape
+    _, S, Hk, _ = k_cache.shape
+    G = Hq // Hk
+    out = torch.zeros(B, Hq, D, dtype=torch.float32, device=q.device)
+    for b in range(B):
+        ctx = int(context_lens[b])
+        k = k_cache[b, :ctx].float()  # [ctx, Hk, D]
+        v = v_cache[b, :ctx].float()
+        for h in range(Hq):
+            kvh = h // G
+            scores = (k[:, kvh] @ (q[b, h].float() * scale))  # [ctx]
+            p = torch.softmax(scores, dim=0)
+            out[b, h] = p @ v[:, kvh]
+    return out
+
+
+# --- dispatching public ops (GPU → HIP kernel, CPU → reference) ---
+
+
+def rmsnorm(
+    input: torch.Tensor,
+    weight: torch.Tensor,
+    residual: Optional[torch.Tensor] = None,
+    eps: float = 1e-5,
+) -> torch.Tensor:
+    if input.is_cuda:
+        return _require_ext().rmsnorm(input, weight, residual, eps)
+    out, folded = rmsnorm_ref(input, weight, residual, eps)
+    if residual is not None:
+        residual.copy_(folded.to(residual.dtype))
+    return out.to(input.dtype)
+
+
+def rope(
+    q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor,
+    theta: float = 500000.0,
+) -> None:
+    if q.is_cuda:
+        _require_ext().rope(q, k, positions.to(torch.int32), theta)
+        return
+    qr, kr = rope_ref(q, k, positions, theta)
+    q.copy_(qr.to(q.dtype))
+    k.copy_(kr.to(k.dtype))
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if gate.is_cuda:
+        return _require_ext().silu_mul(gate, up)
+    return silu_mul_ref(gate, up).to(gate.dtype)
+
+
+def gqa_decode_attn(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        return _require_ext().gqa_decode_attn(
+            q, k_cache, v_cache, context_lens.to(torch.int32), scale
+        )
+    return gqa_decode_attn_ref(q, k_cache, v_cache, context_lens, scale).to(
+        q.dtype
+    )

@@ -1,0 +1,137 @@
+// Torch bindings for the wva_amd MI355X (gfx950) kernels.
+//
+// Built in-tree via torch.utils.cpp_extension (PYTORCH_ROCM_ARCH=gfx950);
+// the .so travels with the repo snapshot to GPU boxes.
+
+#include <torch/extension.h>
+#include <hip/hip_runtime.h>
+#include <c10/hip/HIPStream.h>
+
+extern "C" void launch_rmsnorm(void* out, void* residual_out,
+                               const void* input, const void* residual,
+                               const void* weight, float eps, int rows,
+                               int hidden, hipStream_t stream);
+extern "C" void launch_rope(void* q, void* k, const int* positions, int tokens,
+                            int num_q_heads, int num_k_heads, int head_dim,
+                            float theta, hipStream_t stream);
+extern "C" void launch_silu_mul(void* out, const void* gate, const void* up,
+                                long n, hipStream_t stream);
+extern "C" void launch_gqa_decode_attn(void* out, const void* q,
+                                       const void* k_cache,
+                                       const void* v_cache,
+                                       const int* context_lens, int batch,
+                                       int num_q_heads, int num_kv_heads,
+                                       int max_seq, float scale,
+                                       hipStream_t stream);
+
+namespace {
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16_contig(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// out = rmsnorm(input [+ residual]) * weight; when residual is given it is
+// updated in place to (input + residual).
+torch::Tensor rmsnorm(torch::Tensor input, torch::Tensor weight,
+                      c10::optional<torch::Tensor> residual, double eps) {
+  check_bf16_contig(input, "input");
+  check_bf16_contig(weight, "weight");
+  const int hidden = input.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  TORCH_CHECK(weight.numel() == hidden, "weight size mismatch");
+  const long rows = input.numel() / hidden;
+  auto out = torch::empty_like(input);
+  const void* res_ptr = nullptr;
+  void* res_out_ptr = nullptr;
+  if (residual.has_value()) {
+    check_bf16_contig(residual.value(), "residual");
+    TORCH_CHECK(residual->sizes() == input.sizes(), "residual shape mismatch");
+    res_ptr = residual->data_ptr();
+    res_out_ptr = residual->data_ptr();  // in-place fold
+  }
+  launch_rmsnorm(out.data_ptr(), res_out_ptr, input.data_ptr(), res_ptr,
+                 weight.data_ptr(), (float)eps, (int)rows, hidden,
+                 current_stream());
+  return out;
+}
+
+// In-place NeoX-style RoPE on q [T,Hq,D] and k [T,Hk,D] at positions [T].
+void rope(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+          double theta) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  TORCH_CHECK(positions.is_cuda() && positions.scalar_type() == torch::kInt32,
+              "positions must be int32 on GPU");
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3, "q/k must be [T, H, D]");
+  const int tokens = q.size(0);
+  const int num_q_heads = q.size(1);
+  const int num_k_heads = k.size(1);
+  const int head_dim = q.size(2);
+  TORCH_CHECK(k.size(0) == tokens && k.size(2) == head_dim, "q/k mismatch");
+  TORCH_CHECK(head_dim % 2 == 0, "head_dim must be even");
+  launch_rope(q.data_ptr(), k.data_ptr(), positions.data_ptr<int>(), tokens,
+              num_q_heads, num_k_heads, head_dim, (float)theta,
+              current_stream());
+}
+
+torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
+  check_bf16_contig(gate, "gate");
+  check_bf16_contig(up, "up");
+  TORCH_CHECK(gate.sizes() == up.sizes(), "gate/up shape mismatch");
+  TORCH_CHECK(gate.numel() % 2 == 0, "numel must be even");
+  auto out = torch::empty_like(gate);
+  launch_silu_mul(out.data_ptr(), gate.data_ptr(), up.data_ptr(),
+                  (long)gate.numel(), current_stream());
+  return out;
+}
+
+torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
+                              torch::Tensor v_cache,
+                              torch::Tensor context_lens, double scale) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(context_lens.is_cuda() &&
+                  context_lens.scalar_type() == torch::kInt32,
+              "context_lens must be int32 on GPU");
+  TORCH_CHECK(q.dim() == 3, "q must be [B, Hq, D]");
+  TORCH_CHECK(k_cache.dim() == 4, "k_cache must be [B, S, Hk, D]");
+  const int batch = q.size(0);
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int max_seq = k_cache.size(1);
+  const int num_kv_heads = k_cache.size(2);
+  TORCH_CHECK(head_dim == 128, "head_dim must be 128 (Llama-3 family)");
+  TORCH_CHECK(num_q_heads % num_kv_heads == 0, "Hq must divide by Hk");
+  TORCH_CHECK(num_q_heads / num_kv_heads <= 8, "GQA group size must be <= 8");
+  TORCH_CHECK(v_cache.sizes() == k_cache.sizes(), "k/v cache mismatch");
+  auto out = torch::empty_like(q);
+  launch_gqa_decode_attn(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                         v_cache.data_ptr(), context_lens.data_ptr<int>(),
+                         batch, num_q_heads, num_kv_heads, max_seq,
+                         (float)scale, current_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "Fused (add-)RMSNorm bf16",
+        py::arg("input"), py::arg("weight"), py::arg("residual") = py::none(),
+        py::arg("eps") = 1e-5);
+  m.def("rope", &rope, "Fused in-place NeoX RoPE on q,k",
+        py::arg("q"), py::arg("k"), py::arg("positions"),
+        py::arg("theta") = 500000.0);
+  m.def("silu_mul", &silu_mul, "Fused SwiGLU silu(gate)*up",
+        py::arg("gate"), py::arg("up"));
+  m.def("gqa_decode_attn", &gqa_decode_attn,
+        "GQA decode attention over contiguous KV cache",
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("context_lens"), py::arg("scale"));
+}
