@@ -115,8 +115,16 @@ def build_scenario(profile, seed: int):
     # V2 token-based analyzer + cost-aware optimizer: scales directly to
     # required capacity instead of ±1 per tick — the path that tracks a
     # steep QPS ramp (reference engine_v2.go).
+    # scheduler-queue drain factor: a queued request occupies capacity for
+    # its service time (OUTPUT_TOKENS × ITL at SLO batch), not for the
+    # whole optimization interval (improvement over reference parity 1.0)
+    service_time_s = OUTPUT_TOKENS * profile.itl_ms(profile.max_num_seqs) / 1000.0
+    drain = min(1.0, service_time_s / ENGINE_INTERVAL_S)
     config.update_saturation_config(
-        SaturationScalingConfig.from_dict({"analyzerName": "saturation"})
+        SaturationScalingConfig.from_dict({
+            "analyzerName": "saturation",
+            "schedulerQueueDrainFactor": drain,
+        })
     )
     config.mark_bootstrap_complete()
     app = build_app(

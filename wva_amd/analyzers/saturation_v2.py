@@ -100,7 +100,9 @@ def estimate_capacity_from_params(
 
 
 def estimate_scheduler_queue_demand(
-    sq: Optional[SchedulerQueueMetrics], replica_metrics: List[ReplicaMetrics]
+    sq: Optional[SchedulerQueueMetrics],
+    replica_metrics: List[ReplicaMetrics],
+    drain_factor: float = 1.0,
 ) -> float:
     if sq is None or (sq.queue_size == 0 and sq.queue_bytes == 0):
         return 0.0
@@ -108,7 +110,10 @@ def estimate_scheduler_queue_demand(
     input_tokens = max(sq.queue_bytes / BYTES_PER_TOKEN, sq.queue_size * avg_input)
     input_tokens *= 1 - avg_hit
     output_tokens = sq.queue_size * avg_output
-    return input_tokens + output_tokens
+    # drain_factor < 1 models that queued requests occupy capacity only for
+    # their service time within the optimization interval (Little's law),
+    # not as a standing concurrent population — see SaturationScalingConfig.
+    return (input_tokens + output_tokens) * drain_factor
 
 
 def _median(values: List[int]) -> int:
@@ -188,7 +193,9 @@ class SaturationAnalyzerV2:
                 (vc.replica_count + vc.pending_replicas) * vc.per_replica_capacity
             )
         total_demand += estimate_scheduler_queue_demand(
-            input.scheduler_queue, input.replica_metrics
+            input.scheduler_queue,
+            input.replica_metrics,
+            getattr(cfg, "scheduler_queue_drain_factor", 1.0),
         )
 
         utilization = total_demand / total_supply if total_supply > 0 else 0.0

@@ -31,6 +31,14 @@ class SaturationScalingConfig:
     analyzer_name: str = ""  # "" → V1 percentage, "saturation" → V2 token-based
     scale_up_threshold: float = 0.0
     scale_down_boundary: float = 0.0
+    # Fraction of the scheduler (EPP flow-control) queue's token footprint
+    # counted as concurrent demand. 1.0 = reference parity (the whole
+    # backlog held concurrently, engine_v2.go); < 1.0 models that queued
+    # requests drain within the optimization interval — set it to
+    # min(1, avg request service time / optimization interval). Improvement
+    # over the reference for fast accelerators (MI355X drains hundreds of
+    # requests per second per replica).
+    scheduler_queue_drain_factor: float = 1.0
     # Per-model override entries (keyed by "modelID|namespace"); populated by
     # the ConfigMap parser from the `overrides` list.
     overrides: Dict[str, "SaturationScalingConfig"] = field(default_factory=dict)
@@ -109,6 +117,9 @@ class SaturationScalingConfig:
             analyzer_name=d.get("analyzerName", "") or "",
             scale_up_threshold=float(d.get("scaleUpThreshold", 0.0)),
             scale_down_boundary=float(d.get("scaleDownBoundary", 0.0)),
+            scheduler_queue_drain_factor=float(
+                d.get("schedulerQueueDrainFactor", 1.0)
+            ),
         )
         cfg.apply_defaults()
         for entry in d.get("overrides") or []:
@@ -136,6 +147,8 @@ class SaturationScalingConfig:
             d["scaleUpThreshold"] = self.scale_up_threshold
         if self.scale_down_boundary:
             d["scaleDownBoundary"] = self.scale_down_boundary
+        if self.scheduler_queue_drain_factor != 1.0:
+            d["schedulerQueueDrainFactor"] = self.scheduler_queue_drain_factor
         if self.overrides:
             d["overrides"] = [o.to_dict() for o in self.overrides.values()]
         return d
