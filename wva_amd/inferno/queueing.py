@@ -95,8 +95,11 @@ class MM1KModel:
     def solve(self, lam: float, mu: float) -> None:
         self.lam = lam
         self.mu = mu
+        if lam < 0 or mu <= 0:
+            self.is_valid = False
+            return
         self.rho = self._compute_rho()
-        if self.rho < 0 or self.rho >= self._rho_max() or lam < 0 or mu <= 0:
+        if self.rho < 0 or self.rho >= self._rho_max():
             self.is_valid = False
             return
         self.is_valid = True

@@ -1,0 +1,222 @@
+"""Inferno SystemData JSON schema.
+
+Parity: reference pkg/config/types.go:6-157 and defaults.go:12-35 — same
+field names on the wire (JSON camelCase) so reference system-data files
+load unchanged. Defaults: SLOPercentile 0.95, MaxQueueToBatchRatio 10,
+AcceleratorSwitchFactor (transition penalty) 0.1.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+# defaults (pkg/config/defaults.go)
+SLO_PERCENTILE = 0.95
+MAX_QUEUE_TO_BATCH_RATIO = 10
+ACCELERATOR_SWITCH_FACTOR = 0.1
+
+# best-effort saturation policies (solver/greedy.go bestEffort)
+POLICY_PRIORITY_EXHAUSTIVE = "PriorityExhaustive"
+POLICY_PRIORITY_ROUND_ROBIN = "PriorityRoundRobin"
+POLICY_ROUND_ROBIN = "RoundRobin"
+POLICY_NONE = "None"
+
+
+@dataclass
+class AcceleratorSpec:
+    name: str = ""
+    type: str = ""  # capacity pool type (e.g. "MI355X")
+    multiplicity: int = 1  # devices per unit (xGMI hive size for TP variants)
+    mem_size: int = 0  # GB
+    mem_bw: int = 0  # GB/s
+    power: float = 0.0
+    cost: float = 0.0
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "AcceleratorSpec":
+        return cls(
+            name=d.get("name", ""),
+            type=d.get("type", d.get("name", "")),
+            multiplicity=int(d.get("multiplicity", 1)),
+            mem_size=int(d.get("memSize", 0)),
+            mem_bw=int(d.get("memBW", 0)),
+            power=float(d.get("power", 0.0)),
+            cost=float(d.get("cost", 0.0)),
+        )
+
+
+@dataclass
+class ServiceParmsSpec:
+    alpha: float = 0.0
+    beta: float = 0.0
+    gamma: float = 0.0
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "ServiceParmsSpec":
+        return cls(
+            alpha=float(d.get("alpha", 0.0)),
+            beta=float(d.get("beta", 0.0)),
+            gamma=float(d.get("gamma", 0.0)),
+        )
+
+
+@dataclass
+class ModelAcceleratorPerfData:
+    name: str = ""  # model name
+    acc: str = ""  # accelerator name
+    acc_count: int = 1  # instances of accelerator per replica
+    max_batch_size: int = 0
+    at_tokens: int = 0  # token count at which maxBatchSize was measured
+    service_parms: ServiceParmsSpec = field(default_factory=ServiceParmsSpec)
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "ModelAcceleratorPerfData":
+        return cls(
+            name=d.get("name", ""),
+            acc=d.get("acc", ""),
+            acc_count=int(d.get("accCount", 1)),
+            max_batch_size=int(d.get("maxBatchSize", 0)),
+            at_tokens=int(d.get("atTokens", 0)),
+            service_parms=ServiceParmsSpec.from_dict(d.get("decodeParms") or d.get("serviceParms") or {}),
+        )
+
+
+@dataclass
+class ModelTarget:
+    model: str = ""
+    slo_itl: float = 0.0  # msec
+    slo_ttft: float = 0.0  # msec (queueing + prefill)
+    slo_tps: float = 0.0  # tokens/sec
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "ModelTarget":
+        return cls(
+            model=d.get("model", ""),
+            slo_itl=float(d.get("slo-itl", d.get("SLO_ITL", 0.0)) or 0.0),
+            slo_ttft=float(d.get("slo-ttw", d.get("slo-ttft", d.get("SLO_TTFT", 0.0))) or 0.0),
+            slo_tps=float(d.get("slo-tps", d.get("SLO_TPS", 0.0)) or 0.0),
+        )
+
+
+@dataclass
+class ServiceClassSpec:
+    name: str = ""
+    priority: int = 0  # lower value = higher priority
+    model_targets: List[ModelTarget] = field(default_factory=list)
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "ServiceClassSpec":
+        return cls(
+            name=d.get("name", ""),
+            priority=int(d.get("priority", 0)),
+            model_targets=[ModelTarget.from_dict(t) for t in d.get("data", [])],
+        )
+
+
+@dataclass
+class ServerLoadSpec:
+    arrival_rate: float = 0.0  # requests/min
+    avg_in_tokens: int = 0
+    avg_out_tokens: int = 0
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "ServerLoadSpec":
+        return cls(
+            arrival_rate=float(d.get("arrivalRate", 0.0)),
+            avg_in_tokens=int(d.get("avgInTokens", d.get("avgInputTokens", 0))),
+            avg_out_tokens=int(d.get("avgOutTokens", d.get("avgOutputTokens", 0))),
+        )
+
+
+@dataclass
+class ServerSpec:
+    name: str = ""
+    service_class: str = ""
+    model: str = ""
+    load: ServerLoadSpec = field(default_factory=ServerLoadSpec)
+    max_batch_size: int = 0  # 0 = derive from perf data
+    min_num_replicas: int = 0
+    current_accelerator: str = ""
+    current_num_replicas: int = 0
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "ServerSpec":
+        return cls(
+            name=d.get("name", ""),
+            service_class=d.get("class", d.get("serviceClass", "")),
+            model=d.get("model", ""),
+            load=ServerLoadSpec.from_dict(d.get("load") or {}),
+            max_batch_size=int(d.get("maxBatchSize", 0)),
+            min_num_replicas=int(d.get("minNumReplicas", 0)),
+            current_accelerator=d.get("currentAccelerator", ""),
+            current_num_replicas=int(d.get("currentNumReplicas", 0)),
+        )
+
+
+@dataclass
+class CapacitySpec:
+    """Available accelerator units per type."""
+
+    counts: Dict[str, int] = field(default_factory=dict)
+
+
+@dataclass
+class OptimizerSpec:
+    unlimited: bool = False
+    delayed_best_effort: bool = False
+    saturation_policy: str = POLICY_NONE
+    use_cplex: bool = False  # parity placeholder; no MIP path
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "OptimizerSpec":
+        return cls(
+            unlimited=bool(d.get("unlimited", False)),
+            delayed_best_effort=bool(d.get("delayedBestEffort", False)),
+            saturation_policy=d.get("saturationPolicy", POLICY_NONE),
+        )
+
+
+@dataclass
+class SystemData:
+    accelerators: List[AcceleratorSpec] = field(default_factory=list)
+    models: List[ModelAcceleratorPerfData] = field(default_factory=list)
+    service_classes: List[ServiceClassSpec] = field(default_factory=list)
+    servers: List[ServerSpec] = field(default_factory=list)
+    capacity: Dict[str, int] = field(default_factory=dict)
+    optimizer: OptimizerSpec = field(default_factory=OptimizerSpec)
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any]) -> "SystemData":
+        spec = d.get("spec", d)
+        return cls(
+            accelerators=[
+                AcceleratorSpec.from_dict(a)
+                for a in (spec.get("accelerators") or {}).get("data", [])
+            ] if isinstance(spec.get("accelerators"), dict) else [
+                AcceleratorSpec.from_dict(a) for a in spec.get("accelerators", [])
+            ],
+            models=[
+                ModelAcceleratorPerfData.from_dict(m)
+                for m in (spec.get("models") or {}).get("data", [])
+            ] if isinstance(spec.get("models"), dict) else [
+                ModelAcceleratorPerfData.from_dict(m) for m in spec.get("models", [])
+            ],
+            service_classes=[
+                ServiceClassSpec.from_dict(s)
+                for s in (spec.get("serviceClasses") or {}).get("data", [])
+            ] if isinstance(spec.get("serviceClasses"), dict) else [
+                ServiceClassSpec.from_dict(s) for s in spec.get("serviceClasses", [])
+            ],
+            servers=[
+                ServerSpec.from_dict(s)
+                for s in (spec.get("servers") or {}).get("data", [])
+            ] if isinstance(spec.get("servers"), dict) else [
+                ServerSpec.from_dict(s) for s in spec.get("servers", [])
+            ],
+            capacity=dict(
+                (spec.get("capacity") or {}).get("count", {})
+                if isinstance(spec.get("capacity"), dict)
+                else {}
+            ),
+            optimizer=OptimizerSpec.from_dict(spec.get("optimizer") or {}),
+        )
