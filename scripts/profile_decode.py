@@ -1,0 +1,54 @@
+#!/usr/bin/env python3
+"""Decode-step workload for rocprofv3 profiling on MI355X.
+
+Runs Llama-3-8B decode iterations at a fixed batch so the kernel trace
+captures the steady-state mix (HIP kernels from wva_amd.ops + hipBLASLt
+GEMMs). Keep iterations small — rocprof multiplies overhead.
+
+    cd /tmp && export TMPDIR=/tmp
+    rocprofv3 --kernel-trace --stats -d gpurun_out/prof -- \
+        python scripts/profile_decode.py --batch 64 --iters 10
+"""
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--batch", type=int, default=64)
+    p.add_argument("--iters", type=int, default=10)
+    p.add_argument("--context", type=int, default=512)
+    p.add_argument("--model", default="8b", choices=["8b", "tiny"])
+    args = p.parse_args()
+
+    import torch
+    from wva_amd.calibration.model import LLAMA_3_8B, TINY, LlamaDecodeModel
+
+    cfg = LLAMA_3_8B if args.model == "8b" else TINY
+    model = LlamaDecodeModel(cfg, max_batch=args.batch, max_seq=args.context + args.iters + 8)
+    model.reset(args.batch, args.context)
+    tokens = torch.randint(0, cfg.vocab_size, (args.batch,), device="cuda")
+
+    # warmup
+    for _ in range(3):
+        model.decode_step(tokens)
+    torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    for _ in range(args.iters):
+        model.decode_step(tokens)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    itl_ms = dt * 1000.0 / args.iters
+    print(
+        f"decode: batch={args.batch} ctx={args.context} iters={args.iters} "
+        f"ITL={itl_ms:.3f} ms  decode_tps={args.batch / (itl_ms / 1000.0):.0f}"
+    )
+
+
+if __name__ == "__main__":
+    main()
