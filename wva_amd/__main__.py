@@ -1,0 +1,85 @@
+"""Process entry point — `python -m wva_amd`.
+
+Parity: reference cmd/main.go flag surface (metrics/probe addresses,
+leader election timings, watch namespace, verbosity) + env/config-file
+loading via the same keys (PROMETHEUS_BASE_URL, GLOBAL_OPT_INTERVAL,
+WVA_SCALE_TO_ZERO, ...). In this environment there is no Kubernetes API
+server; `--emulated` starts the full controller stack against an in-memory
+cluster seeded from a scenario file, which is the supported mode (the
+FakeCluster implements the client surface a REST-backed client would).
+"""
+from __future__ import annotations
+
+import argparse
+import signal
+import sys
+import time
+
+from .app import build_app
+from .config.loader import load_config
+from .emulator.cluster_sim import ClusterSim
+from .emulator.sim_source import SimMetricsSource
+from .emulator.vllm_sim import ServiceProfile
+from .kube.fake import FakeCluster
+from .utils.logging import get_logger, setup_logging
+
+log = get_logger("main")
+
+
+def parse_flags(argv=None):
+    p = argparse.ArgumentParser(prog="wva-amd")
+    p.add_argument("--metrics-bind-address", default=None)
+    p.add_argument("--health-probe-bind-address", default=None)
+    p.add_argument("--leader-elect", action="store_true", default=None)
+    p.add_argument("--watch-namespace", default=None)
+    p.add_argument("--config", default=None, help="YAML config file path")
+    p.add_argument("--v", type=int, default=None, help="log verbosity (0-5)")
+    p.add_argument(
+        "--emulated",
+        action="store_true",
+        help="run against the in-memory emulated cluster",
+    )
+    return p.parse_args(argv)
+
+
+def main(argv=None) -> int:
+    args = parse_flags(argv)
+    flags = {
+        "METRICS_BIND_ADDRESS": args.metrics_bind_address,
+        "HEALTH_PROBE_BIND_ADDRESS": args.health_probe_bind_address,
+        "LEADER_ELECT": args.leader_elect,
+        "WATCH_NAMESPACE": args.watch_namespace,
+        "V": args.v,
+    }
+    config = load_config(
+        flags=flags,
+        config_path=args.config,
+        require_prometheus=not args.emulated,
+    )
+    setup_logging(config.infra.logger_verbosity or 2)
+
+    cluster = FakeCluster()
+    source = None
+    if args.emulated:
+        sim = ClusterSim(cluster)
+        source = SimMetricsSource(sim)
+        log.info("running in emulated mode (in-memory cluster)")
+
+    app = build_app(cluster, config, source=source)
+    app.configmap_reconciler.bootstrap_initial_configmaps()
+    app.start()
+    log.info("manager started (leader_elect=%s)", config.infra.enable_leader_election)
+
+    stop = []
+    signal.signal(signal.SIGINT, lambda *_: stop.append(1))
+    signal.signal(signal.SIGTERM, lambda *_: stop.append(1))
+    try:
+        while not stop:
+            time.sleep(0.5)
+    finally:
+        app.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
