@@ -1,0 +1,429 @@
+"""Runtime + controller tests: executor retry, manager dispatch + leader
+election, datastore, indexers, predicates, ConfigMap reconciler, metrics
+emitter, config loader. Mirrors the reference's envtest/unit coverage for
+cmd/main wiring, internal/controller and internal/config.
+"""
+import os
+import threading
+import time
+
+import pytest
+from prometheus_client import CollectorRegistry
+
+from wva_amd.api.types import (
+    CrossVersionObjectReference,
+    ObjectMeta,
+    VariantAutoscaling,
+    VariantAutoscalingSpec,
+)
+from wva_amd.config.config import Config
+from wva_amd.config.loader import load_config
+from wva_amd.config.validation import ConfigLoadError
+from wva_amd.controllers.configmap import (
+    ConfigMapReconciler,
+    parse_saturation_configmap,
+)
+from wva_amd.controllers.predicates import (
+    configmap_predicate,
+    deployment_predicate,
+    variant_autoscaling_predicate,
+)
+from wva_amd.datastore.datastore import Datastore
+from wva_amd.engines.common import DecisionCache, DecisionTrigger
+from wva_amd.kube.fake import ADDED, DELETED, MODIFIED, FakeCluster, WatchEvent
+from wva_amd.kube.indexers import VAIndex
+from wva_amd.kube.objects import ConfigMap, Deployment, EndpointPool
+from wva_amd.metrics.metrics import MetricsEmitter
+from wva_amd.runtime.executor import PollingExecutor
+from wva_amd.runtime.manager import LeaderElector, Manager
+
+
+def make_va(name="va1", ns="default", target=None, labels=None):
+    return VariantAutoscaling(
+        metadata=ObjectMeta(name=name, namespace=ns, labels=labels or {}),
+        spec=VariantAutoscalingSpec(
+            scale_target_ref=CrossVersionObjectReference(name=target or name),
+            model_id="m",
+        ),
+    )
+
+
+class TestPollingExecutor:
+    def test_tick_retry_until_success(self):
+        calls = []
+
+        def tick():
+            calls.append(1)
+            if len(calls) < 3:
+                raise RuntimeError("boom")
+
+        ex = PollingExecutor(999, tick, name="t")
+        ex.INITIAL_RETRY_BACKOFF_SECONDS = 0.01
+        # run_once retries internally with backoff until success
+        import wva_amd.runtime.executor as exmod
+
+        old = exmod.INITIAL_RETRY_BACKOFF_SECONDS
+        exmod.INITIAL_RETRY_BACKOFF_SECONDS = 0.01
+        try:
+            ex.run_once()
+        finally:
+            exmod.INITIAL_RETRY_BACKOFF_SECONDS = old
+        assert len(calls) == 3
+
+    def test_periodic_loop(self):
+        calls = []
+        ex = PollingExecutor(0.02, lambda: calls.append(1), name="t")
+        ex.start()
+        time.sleep(0.15)
+        ex.stop()
+        assert len(calls) >= 3
+
+    def test_stop_interrupts_backoff(self):
+        ex = PollingExecutor(10, lambda: (_ for _ in ()).throw(RuntimeError()), name="t")
+        ex.start()
+        time.sleep(0.05)
+        t0 = time.time()
+        ex.stop()
+        assert time.time() - t0 < 3.0
+
+
+class TestLeaderElection:
+    def test_acquire_and_renew(self):
+        c = FakeCluster()
+        e1 = LeaderElector(c, "test-lease", identity="a", lease_duration=1.0)
+        assert e1.try_acquire_or_renew()
+        assert e1.try_acquire_or_renew()  # renew
+
+    def test_second_elector_blocked_until_expiry(self):
+        c = FakeCluster()
+        e1 = LeaderElector(c, "test-lease", identity="a", lease_duration=0.1)
+        e2 = LeaderElector(c, "test-lease", identity="b", lease_duration=0.1)
+        assert e1.try_acquire_or_renew()
+        assert not e2.try_acquire_or_renew()
+        time.sleep(0.15)  # lease expires
+        assert e2.try_acquire_or_renew()
+
+    def test_release_on_cancel_fast_failover(self):
+        c = FakeCluster()
+        e1 = LeaderElector(c, "test-lease", identity="a", lease_duration=60.0)
+        e2 = LeaderElector(c, "test-lease", identity="b", lease_duration=60.0)
+        assert e1.try_acquire_or_renew()
+        e1.release()
+        assert e2.try_acquire_or_renew()  # no wait for 60s expiry
+
+
+class TestManagerDispatch:
+    def test_watch_event_drives_reconcile(self):
+        c = FakeCluster()
+        config = Config()
+        config.mark_bootstrap_complete()
+        mgr = Manager(c, config)
+        seen = []
+        mgr.register_reconciler(
+            ["VariantAutoscaling"],
+            lambda e: e.type == ADDED,
+            lambda ns, name: seen.append((ns, name)),
+        )
+        mgr.start()
+        try:
+            c.create(make_va("v1"))
+            deadline = time.time() + 2
+            while not seen and time.time() < deadline:
+                time.sleep(0.02)
+            assert seen == [("default", "v1")]
+        finally:
+            mgr.stop()
+
+    def test_decision_trigger_feeds_va_reconciler(self):
+        c = FakeCluster()
+        config = Config()
+        config.mark_bootstrap_complete()
+        trigger = DecisionTrigger()
+        mgr = Manager(c, config, trigger)
+        seen = []
+        mgr.register_reconciler(
+            ["VariantAutoscaling"],
+            lambda e: False,
+            lambda ns, name: seen.append((ns, name)),
+            is_va_reconciler=True,
+        )
+        mgr.start()
+        try:
+            trigger.push("ns1", "va9")
+            deadline = time.time() + 2
+            while not seen and time.time() < deadline:
+                time.sleep(0.02)
+            assert seen == [("ns1", "va9")]
+        finally:
+            mgr.stop()
+
+    def test_leader_gates_runnables(self):
+        c = FakeCluster()
+        config = Config()
+        config.infra.enable_leader_election = True
+        config.infra.retry_period_seconds = 0.05
+        config.mark_bootstrap_complete()
+        started = threading.Event()
+
+        class R:
+            def start(self):
+                started.set()
+
+            def stop(self):
+                pass
+
+        mgr = Manager(c, config)
+        mgr.add_runnable(R())
+        mgr.start()
+        try:
+            assert started.wait(timeout=2.0)
+            assert mgr.is_leader()
+        finally:
+            mgr.stop()
+
+
+class TestDatastore:
+    def test_pool_registry(self):
+        c = FakeCluster()
+        ds = Datastore(c)
+        ds.pool_set(EndpointPool(name="p", namespace="ns", selector={"a": "b"}))
+        assert ds.pool_get("ns", "p") is not None
+        assert ds.pool_source("ns", "p") is not None
+        assert ds.pool_get_from_labels("ns", {"a": "b", "x": "y"}).name == "p"
+        assert ds.pool_get_from_labels("ns", {"a": "wrong"}) is None
+        assert ds.pool_get_from_labels("other", {"a": "b"}) is None
+        ds.pool_delete("ns", "p")
+        assert ds.pool_get("ns", "p") is None
+
+    def test_namespace_tracking(self):
+        ds = Datastore(FakeCluster())
+        ds.namespace_track("a")
+        assert ds.namespace_is_tracked("a")
+        assert not ds.namespace_is_tracked("b")
+        ds.namespace_untrack("a")
+        assert not ds.namespace_is_tracked("a")
+
+
+class TestIndexers:
+    def test_find_va_for_deployment(self):
+        c = FakeCluster()
+        c.create(make_va("va1", target="deploy1"))
+        idx = VAIndex(c)
+        va = idx.find_va_for_deployment("default", "deploy1")
+        assert va is not None and va.name == "va1"
+        assert idx.find_va_for_deployment("default", "other") is None
+
+    def test_duplicate_target_rejected(self):
+        c = FakeCluster()
+        c.create(make_va("va1", target="deploy1"))
+        c.create(make_va("va2", target="deploy1"))
+        idx = VAIndex(c)
+        with pytest.raises(ValueError):
+            idx.find_va_for_deployment("default", "deploy1")
+        assert idx.validate_unique_targets()
+
+
+class TestPredicates:
+    def test_va_predicate_create_only(self):
+        c = FakeCluster()
+        pred = variant_autoscaling_predicate(c)
+        va = make_va()
+        assert pred(WatchEvent(type=ADDED, kind="VariantAutoscaling", obj=va))
+        assert not pred(WatchEvent(type=MODIFIED, kind="VariantAutoscaling", obj=va))
+        assert pred(WatchEvent(type=DELETED, kind="VariantAutoscaling", obj=va))
+
+    def test_controller_instance_isolation(self):
+        c = FakeCluster()
+        pred = variant_autoscaling_predicate(c)
+        labeled = make_va(labels={"wva.llmd.ai/controller-instance": "other"})
+        assert not pred(WatchEvent(type=ADDED, kind="VariantAutoscaling", obj=labeled))
+        os.environ["CONTROLLER_INSTANCE"] = "other"
+        try:
+            assert pred(WatchEvent(type=ADDED, kind="VariantAutoscaling", obj=labeled))
+            assert not pred(WatchEvent(type=ADDED, kind="VariantAutoscaling", obj=make_va()))
+        finally:
+            del os.environ["CONTROLLER_INSTANCE"]
+
+    def test_deployment_predicate(self):
+        pred = deployment_predicate()
+        d = Deployment(metadata=ObjectMeta(name="d", namespace="ns"))
+        assert pred(WatchEvent(type=ADDED, kind="Deployment", obj=d))
+        assert not pred(WatchEvent(type=MODIFIED, kind="Deployment", obj=d))
+        assert pred(WatchEvent(type=DELETED, kind="Deployment", obj=d))
+
+    def test_configmap_predicate_well_known_names(self):
+        pred = configmap_predicate()
+        good = ConfigMap(metadata=ObjectMeta(
+            name="wva-saturation-scaling-config", namespace="ns"))
+        bad = ConfigMap(metadata=ObjectMeta(name="random", namespace="ns"))
+        assert pred(WatchEvent(type=MODIFIED, kind="ConfigMap", obj=good))
+        assert not pred(WatchEvent(type=MODIFIED, kind="ConfigMap", obj=bad))
+
+
+class TestConfigMapReconciler:
+    def _setup(self, controller_ns="wva-system"):
+        os.environ["POD_NAMESPACE"] = controller_ns
+        c = FakeCluster()
+        config = Config()
+        ds = Datastore(c)
+        rec = ConfigMapReconciler(c, config, ds)
+        return c, config, ds, rec
+
+    def test_global_saturation_config(self):
+        c, config, ds, rec = self._setup()
+        c.create(ConfigMap(
+            metadata=ObjectMeta(
+                name="wva-saturation-scaling-config", namespace="wva-system"
+            ),
+            data={"default": "kvCacheThreshold: 0.7\nqueueLengthThreshold: 9\n"},
+        ))
+        rec.reconcile("wva-system", "wva-saturation-scaling-config")
+        cfg = config.saturation_config()
+        assert cfg.kv_cache_threshold == 0.7
+        assert cfg.queue_length_threshold == 9
+
+    def test_namespace_local_override(self):
+        c, config, ds, rec = self._setup()
+        ds.namespace_track("team-a")
+        c.create(ConfigMap(
+            metadata=ObjectMeta(
+                name="wva-saturation-scaling-config", namespace="team-a"
+            ),
+            data={"default": "kvCacheThreshold: 0.5\n"},
+        ))
+        rec.reconcile("team-a", "wva-saturation-scaling-config")
+        assert config.saturation_config_for_namespace("team-a").kv_cache_threshold == 0.5
+        assert config.saturation_config_for_namespace("other").kv_cache_threshold == 0.8
+
+    def test_untracked_namespace_ignored(self):
+        c, config, ds, rec = self._setup()
+        c.create(ConfigMap(
+            metadata=ObjectMeta(
+                name="wva-saturation-scaling-config", namespace="stranger"
+            ),
+            data={"default": "kvCacheThreshold: 0.5\n"},
+        ))
+        rec.reconcile("stranger", "wva-saturation-scaling-config")
+        assert config.saturation_config_for_namespace("stranger").kv_cache_threshold == 0.8
+
+    def test_deletion_removes_override(self):
+        c, config, ds, rec = self._setup()
+        ds.namespace_track("team-a")
+        c.create(ConfigMap(
+            metadata=ObjectMeta(
+                name="wva-saturation-scaling-config", namespace="team-a"
+            ),
+            data={"default": "kvCacheThreshold: 0.5\n"},
+        ))
+        rec.reconcile("team-a", "wva-saturation-scaling-config")
+        c.delete("ConfigMap", "team-a", "wva-saturation-scaling-config")
+        rec.reconcile("team-a", "wva-saturation-scaling-config")
+        assert config.saturation_config_for_namespace("team-a").kv_cache_threshold == 0.8
+
+    def test_invalid_config_keeps_previous(self):
+        c, config, ds, rec = self._setup()
+        cm = ConfigMap(
+            metadata=ObjectMeta(
+                name="wva-saturation-scaling-config", namespace="wva-system"
+            ),
+            data={"default": "kvCacheThreshold: 0.7\n"},
+        )
+        c.create(cm)
+        rec.reconcile("wva-system", "wva-saturation-scaling-config")
+        cm.data = {"default": "kvCacheThreshold: 7.0\n"}  # invalid > 1
+        c.update(cm)
+        rec.reconcile("wva-system", "wva-saturation-scaling-config")
+        assert config.saturation_config().kv_cache_threshold == 0.7
+
+    def test_per_model_overrides(self):
+        parsed = parse_saturation_configmap({
+            "default": "kvCacheThreshold: 0.8\n",
+            "llama": "model_id: m1\nnamespace: ns\nkvCacheThreshold: 0.6\n",
+        })
+        assert parsed.for_model("m1", "ns").kv_cache_threshold == 0.6
+        assert parsed.for_model("other", "ns").kv_cache_threshold == 0.8
+
+    def test_bootstrap_marks_ready(self):
+        c, config, ds, rec = self._setup()
+        assert not config.is_bootstrap_complete()
+        rec.bootstrap_initial_configmaps()
+        assert config.is_bootstrap_complete()
+
+
+class TestMetricsEmitter:
+    def test_labels_and_ratio(self):
+        reg = CollectorRegistry()
+        e = MetricsEmitter(registry=reg, controller_instance="")
+        e.emit_replica_metrics("v", "ns", current=2, desired=4, accelerator_type="MI355X")
+        samples = {}
+        for fam in reg.collect():
+            for s in fam.samples:
+                samples[s.name] = s
+        assert samples["wva_desired_replicas"].value == 4
+        assert samples["wva_current_replicas"].value == 2
+        assert samples["wva_desired_ratio"].value == 2.0
+        assert samples["wva_desired_replicas"].labels == {
+            "variant_name": "v", "namespace": "ns", "accelerator_type": "MI355X"
+        }
+
+    def test_zero_current_ratio_is_desired(self):
+        reg = CollectorRegistry()
+        e = MetricsEmitter(registry=reg, controller_instance="")
+        e.emit_replica_metrics("v", "ns", current=0, desired=3, accelerator_type="A")
+        for fam in reg.collect():
+            if fam.name == "wva_desired_ratio":
+                assert fam.samples[0].value == 3.0
+
+    def test_scaling_counter(self):
+        reg = CollectorRegistry()
+        e = MetricsEmitter(registry=reg, controller_instance="c1")
+        e.emit_replica_scaling_metrics("v", "ns", "up", "saturation")
+        for fam in reg.collect():
+            if fam.name == "wva_replica_scaling":  # counter family name
+                s = fam.samples[0]
+                assert s.labels["direction"] == "up"
+                assert s.labels["controller_instance"] == "c1"
+
+
+class TestConfigLoader:
+    def test_defaults(self):
+        cfg = load_config(env={}, require_prometheus=False)
+        assert cfg.infra.optimization_interval_seconds == 60.0
+        assert cfg.infra.lease_duration_seconds == 60.0
+        assert cfg.scale_from_zero_max_concurrency() == 10
+        assert cfg.cache.ttl_seconds == 30.0
+
+    def test_env_overrides_defaults(self):
+        cfg = load_config(
+            env={"GLOBAL_OPT_INTERVAL": "30s", "WVA_SCALE_TO_ZERO": "true"},
+            require_prometheus=False,
+        )
+        assert cfg.infra.optimization_interval_seconds == 30.0
+        assert cfg.scale_to_zero_enabled()
+
+    def test_flags_override_env(self):
+        cfg = load_config(
+            flags={"V": 4},
+            env={"V": "1"},
+            require_prometheus=False,
+        )
+        assert cfg.infra.logger_verbosity == 4
+
+    def test_prometheus_required(self):
+        with pytest.raises(ConfigLoadError):
+            load_config(env={}, require_prometheus=True)
+        cfg = load_config(
+            env={"PROMETHEUS_BASE_URL": "http://prom:9090"},
+            require_prometheus=True,
+        )
+        assert cfg.prometheus.base_url == "http://prom:9090"
+
+    def test_invalid_leader_timings(self):
+        with pytest.raises(ConfigLoadError):
+            load_config(
+                env={
+                    "LEADER_ELECTION_LEASE_DURATION": "10s",
+                    "LEADER_ELECTION_RENEW_DEADLINE": "50s",
+                },
+                require_prometheus=False,
+            )

@@ -215,7 +215,7 @@ class Lease:
 
     metadata: ObjectMeta = field(default_factory=ObjectMeta)
     holder_identity: str = ""
-    lease_duration_seconds: int = 60
+    lease_duration_seconds: float = 60
     acquire_time: Optional[Any] = None
     renew_time: Optional[Any] = None
 

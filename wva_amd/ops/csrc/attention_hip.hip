@@ -3,14 +3,26 @@
 // contiguous KV cache — MI355X (gfx950).
 //
 // Shape contract (Llama-3 family): head_dim = 128, G = Hq/Hk ≤ 8 query
-// heads share one KV head. One 4-wave workgroup per (batch, kv_head);
-// the G query heads of the group are processed together so K/V stream
-// from HBM exactly once per group (the op is HBM-bound: 2·ctx·D·2 bytes
-// per (b,kv_head)). Lanes hold 2 contiguous elements (lane l → elements
-// 2l, 2l+1) so every K/V row is one coalesced 256 B wave read (dword per
-// lane). Waves take interleaved position tiles; partial (max, sum, acc)
-// merge across waves flash-decoding style through LDS. fp32 softmax and
-// accumulation throughout.
+// heads share one KV head. One 4-wave workgroup per (batch, kv_head).
+//
+// Structure (v3 — shared tile, head-per-wave):
+//   * Each iteration stages ONE shared 64-position K/V tile into LDS
+//     cooperatively (all 4 waves, coalesced row loads), rows padded to 65
+//     dwords so every position-per-lane sweep is bank-conflict-free
+//     (bank = (65·row + d) % 32 = (row + d) % 32).
+//   * Query heads are OWNED by waves (head g → wave g % 4): each wave
+//     runs the full online softmax for its heads, so there is no
+//     cross-wave flash-decoding merge at all — the owning wave writes the
+//     final output directly.
+//   * Scores: lane ↔ position (full 64-lane utilization), K swept with
+//     ds_read_b128; ONE max+sum wave-reduction per (tile, head) — the v1
+//     kernel did 6-shuffle reductions per position per head and ran at
+//     ~10% of HBM speed (profiles/decode8b_b64_ctx512_kernel_stats.txt).
+//   * PV: p staged to LDS (broadcast reads), lane ↔ output dword.
+// fp32 softmax and accumulation throughout.
+//
+// LDS: K+V tile 2×64×260 B ≈ 33 KiB + q 4 KiB + p 2 KiB ≈ 39 KiB/WG
+// → 4 workgroups/CU admissible; grid B×Hk.
 //
 // Layouts: q [B, Hq, 128], k/v cache [B, S_max, Hk, 128], out [B, Hq, 128].
 
@@ -18,9 +30,13 @@
 
 #define HEAD_DIM 128
 #define MAX_G 8
-#define TILE 8  // positions per wave inner iteration
+#define TILE 64            // positions per shared tile (= wave width)
+#define NUM_WAVES 4
+#define ROW_DW 65          // LDS row stride in dwords (64 + 1 pad)
 
-__global__ __launch_bounds__(256) void gqa_decode_attn_kernel(
+typedef __attribute__((ext_vector_type(4))) unsigned int uint4_t;
+
+__global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
     bf16* __restrict__ out,            // [B, Hq, 128]
     const bf16* __restrict__ q,        // [B, Hq, 128]
     const bf16* __restrict__ k_cache,  // [B, S_max, Hk, 128]
@@ -37,92 +53,120 @@ __global__ __launch_bounds__(256) void gqa_decode_attn_kernel(
 
   const int lane = threadIdx.x & (WAVE_SIZE - 1);
   const int wave = threadIdx.x / WAVE_SIZE;
-  const int num_waves = blockDim.x / WAVE_SIZE;
 
-  // Per-lane fragment of each query head in the group (2 elements).
-  float qf[MAX_G][2];
-  for (int g = 0; g < G; ++g) {
-    const bf16x2* qrow = reinterpret_cast<const bf16x2*>(
-        q + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM);
-    bf16x2 v = qrow[lane];
-    qf[g][0] = bf2f(v.x) * scale;
-    qf[g][1] = bf2f(v.y) * scale;
+  __shared__ float q_smem[MAX_G][HEAD_DIM];
+  __shared__ unsigned int k_smem[TILE * ROW_DW];  // bf16x2-packed rows
+  __shared__ unsigned int v_smem[TILE * ROW_DW];
+  __shared__ float p_smem[MAX_G][TILE];
+
+  // Stage scaled q into LDS (fp32).
+  for (int idx = threadIdx.x; idx < G * (HEAD_DIM / 2); idx += blockDim.x) {
+    const int g = idx / (HEAD_DIM / 2);
+    const int d2 = idx % (HEAD_DIM / 2);
+    const bf16x2 v2 = reinterpret_cast<const bf16x2*>(
+        q + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM)[d2];
+    q_smem[g][2 * d2] = bf2f(v2.x) * scale;
+    q_smem[g][2 * d2 + 1] = bf2f(v2.y) * scale;
   }
 
-  float m[MAX_G], s[MAX_G], acc[MAX_G][2];
-  for (int g = 0; g < G; ++g) {
-    m[g] = -INFINITY;
-    s[g] = 0.0f;
-    acc[g][0] = acc[g][1] = 0.0f;
+  // Per-wave running softmax state for its owned heads (g = wave + j*4).
+  const int heads_mine = (G > wave) ? (G - wave + NUM_WAVES - 1) / NUM_WAVES : 0;
+  float m[2], s[2], acc[2][2];  // at most 2 heads per wave (G ≤ 8)
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    m[j] = -INFINITY;
+    s[j] = 0.0f;
+    acc[j][0] = acc[j][1] = 0.0f;
   }
 
-  const long kv_row_stride = (long)num_kv_heads * HEAD_DIM;
-  const bf16x2* k_base = reinterpret_cast<const bf16x2*>(
-      k_cache + (long)b * max_seq * kv_row_stride + (long)kvh * HEAD_DIM);
-  const bf16x2* v_base = reinterpret_cast<const bf16x2*>(
-      v_cache + (long)b * max_seq * kv_row_stride + (long)kvh * HEAD_DIM);
-  const long row2 = kv_row_stride / 2;  // bf16x2 stride between positions
+  const long kv_row_dw = (long)num_kv_heads * (HEAD_DIM / 2);
+  const unsigned int* k_base = reinterpret_cast<const unsigned int*>(
+      k_cache + (long)b * max_seq * num_kv_heads * HEAD_DIM +
+      (long)kvh * HEAD_DIM);
+  const unsigned int* v_base = reinterpret_cast<const unsigned int*>(
+      v_cache + (long)b * max_seq * num_kv_heads * HEAD_DIM +
+      (long)kvh * HEAD_DIM);
 
-  // Interleaved tiles: wave w takes tiles w, w+num_waves, ...
-  for (int t0 = wave * TILE; t0 < ctx; t0 += num_waves * TILE) {
+  __syncthreads();  // q_smem visible
+
+  for (int t0 = 0; t0 < ctx; t0 += TILE) {
     const int tn = min(TILE, ctx - t0);
-#pragma unroll
-    for (int ti = 0; ti < TILE; ++ti) {
-      if (ti >= tn) break;
-      const int t = t0 + ti;
-      const bf16x2 kv = k_base[(long)t * row2 + lane];
-      const float k0 = bf2f(kv.x), k1 = bf2f(kv.y);
-      const bf16x2 vv = v_base[(long)t * row2 + lane];
-      const float v0 = bf2f(vv.x), v1 = bf2f(vv.y);
-#pragma unroll
-      for (int g = 0; g < MAX_G; ++g) {
-        if (g >= G) break;
-        float partial = fmaf(qf[g][0], k0, qf[g][1] * k1);
-        const float score = wave_reduce_sum(partial);
-        // online softmax update
-        const float m_new = fmaxf(m[g], score);
-        const float corr = __expf(m[g] - m_new);
-        const float p = __expf(score - m_new);
-        s[g] = s[g] * corr + p;
-        acc[g][0] = fmaf(acc[g][0], corr, p * v0);
-        acc[g][1] = fmaf(acc[g][1], corr, p * v1);
-        m[g] = m_new;
+
+    // --- cooperative staging: 128 row-loads (K+V) over 4 waves ---
+    // row-load r in [0, 2·tn): even→K row r/2, odd→V row r/2.
+    for (int r = wave; r < 2 * tn; r += NUM_WAVES) {
+      const int row = r >> 1;
+      const long src = (long)(t0 + row) * kv_row_dw + lane;
+      if (r & 1) {
+        v_smem[row * ROW_DW + lane] = v_base[src];
+      } else {
+        k_smem[row * ROW_DW + lane] = k_base[src];
       }
     }
-  }
+    __syncthreads();
 
-  // Merge partials across waves through LDS.
-  // Layout per (wave, g): [m, s] scalars + 128 acc floats.
-  __shared__ float lds_ms[4][MAX_G][2];
-  __shared__ float lds_acc[4][MAX_G][HEAD_DIM];
-  for (int g = 0; g < G; ++g) {
-    if (lane == 0) {
-      lds_ms[wave][g][0] = m[g];
-      lds_ms[wave][g][1] = s[g];
-    }
-    lds_acc[wave][g][2 * lane] = acc[g][0];
-    lds_acc[wave][g][2 * lane + 1] = acc[g][1];
-  }
-  __syncthreads();
-
-  if (wave == 0) {
-    for (int g = 0; g < G; ++g) {
-      float M = -INFINITY;
-      for (int w = 0; w < num_waves; ++w) M = fmaxf(M, lds_ms[w][g][0]);
-      float S = 0.0f, o0 = 0.0f, o1 = 0.0f;
-      for (int w = 0; w < num_waves; ++w) {
-        const float mw = lds_ms[w][g][0];
-        if (mw == -INFINITY) continue;
-        const float f = __expf(mw - M);
-        S += lds_ms[w][g][1] * f;
-        o0 = fmaf(lds_acc[w][g][2 * lane], f, o0);
-        o1 = fmaf(lds_acc[w][g][2 * lane + 1], f, o1);
+    // --- scores + softmax + p staging, per owned head ---
+    const bool live = lane < tn;
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      float score = -INFINITY;
+      if (live) {
+        score = 0.0f;
+        const unsigned int* krow = &k_smem[lane * ROW_DW];
+        const float* qg = q_smem[g];
+#pragma unroll 4
+        for (int d4 = 0; d4 < HEAD_DIM / 8; ++d4) {
+          uint4_t kv4 = *reinterpret_cast<const uint4_t*>(&krow[d4 * 4]);
+          unsigned int kw[4] = {kv4[0], kv4[1], kv4[2], kv4[3]};
+          float partial = 0.0f;
+#pragma unroll
+          for (int w2 = 0; w2 < 4; ++w2) {
+            const bf16x2 p2 = *reinterpret_cast<const bf16x2*>(&kw[w2]);
+            partial = fmaf(bf2f(p2.x), qg[d4 * 8 + 2 * w2], partial);
+            partial = fmaf(bf2f(p2.y), qg[d4 * 8 + 2 * w2 + 1], partial);
+          }
+          score += partial;
+        }
       }
-      const float inv = S > 0.0f ? 1.0f / S : 0.0f;
-      bf16x2* orow = reinterpret_cast<bf16x2*>(
-          out + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM);
-      orow[lane] = bf16x2{f2bf(o0 * inv), f2bf(o1 * inv)};
+      const float tile_max = wave_reduce_max(score);
+      const float m_new = fmaxf(m[j], tile_max);
+      const float p = live ? __expf(score - m_new) : 0.0f;
+      const float tile_sum = wave_reduce_sum(p);
+      const float corr = (m[j] == -INFINITY) ? 0.0f : __expf(m[j] - m_new);
+      s[j] = s[j] * corr + tile_sum;
+      acc[j][0] *= corr;
+      acc[j][1] *= corr;
+      m[j] = m_new;
+      p_smem[g][lane] = p;  // lane ↔ position, in-bounds (TILE == 64)
     }
+    // Waves read only their OWN heads' p rows and their own K/V-view; the
+    // barrier below orders the NEXT tile's staging against this compute.
+
+    // --- PV: lane ↔ output dword, sweep tile positions ---
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      const float* pg = p_smem[g];
+      float a0 = acc[j][0], a1 = acc[j][1];
+      for (int t = 0; t < tn; ++t) {
+        const unsigned int vw = v_smem[t * ROW_DW + lane];
+        const bf16x2 vv = *reinterpret_cast<const bf16x2*>(&vw);
+        const float p = pg[t];  // broadcast
+        a0 = fmaf(p, bf2f(vv.x), a0);
+        a1 = fmaf(p, bf2f(vv.y), a1);
+      }
+      acc[j][0] = a0;
+      acc[j][1] = a1;
+    }
+    __syncthreads();  // compute done before next tile overwrites LDS
+  }
+
+  // --- finalize: owning wave writes its heads directly ---
+  for (int j = 0; j < heads_mine; ++j) {
+    const int g = wave + j * NUM_WAVES;
+    const float inv = s[j] > 0.0f ? 1.0f / s[j] : 0.0f;
+    bf16x2* orow = reinterpret_cast<bf16x2*>(
+        out + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM);
+    orow[lane] = bf16x2{f2bf(acc[j][0] * inv), f2bf(acc[j][1] * inv)};
   }
 }
 

@@ -80,7 +80,7 @@ class LeaderElector:
                     name=self.lease_name, namespace=self.LEASE_NAMESPACE
                 ),
                 holder_identity=self.identity,
-                lease_duration_seconds=int(self.lease_duration),
+                lease_duration_seconds=self.lease_duration,
                 acquire_time=now,
                 renew_time=now,
             )
