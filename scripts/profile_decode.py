@@ -52,3 +52,30 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def attn_bench():
+    """Standalone attention microbench: bytes-based bandwidth estimate."""
+    import torch
+    from wva_amd import ops
+
+    for batch, hq, hk, ctx in [(64, 32, 8, 512), (64, 32, 8, 2048),
+                               (256, 32, 8, 512), (8, 64, 8, 4096)]:
+        S = ctx
+        q = torch.randn(batch, hq, 128, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(batch, S, hk, 128, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(batch, S, hk, 128, device="cuda", dtype=torch.bfloat16)
+        lens = torch.full((batch,), ctx, device="cuda", dtype=torch.int32)
+        for _ in range(3):
+            ops.gqa_decode_attn(q, k, v, lens)
+        torch.cuda.synchronize()
+        import time as _t
+        n = 20
+        t0 = _t.perf_counter()
+        for _ in range(n):
+            ops.gqa_decode_attn(q, k, v, lens)
+        torch.cuda.synchronize()
+        us = (_t.perf_counter() - t0) / n * 1e6
+        kv_bytes = 2 * batch * ctx * hk * 128 * 2
+        print(f"attn b={batch} hq={hq} hk={hk} ctx={ctx}: {us:.1f} us, "
+              f"KV {kv_bytes/1e6:.1f} MB, {kv_bytes/us/1e3:.2f} TB/s")
