@@ -35,8 +35,14 @@
 
 typedef __attribute__((ext_vector_type(4))) unsigned int uint4_t;
 
+// Split-KV: when the (B × Hk) grid is too small to fill 256 CUs (small
+// batch / TP-8 70B decode), the context is additionally partitioned across
+// blockIdx.z splits; each split writes (m, s, acc[128]) partials to a
+// workspace [B, Hk, G, splits, 2 + 128] and a second kernel merges.
+// splits == 1 takes the direct-store fast path.
 __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
     bf16* __restrict__ out,            // [B, Hq, 128]
+    float* __restrict__ workspace,     // [B, Hk, G, splits, 2+128] or null
     const bf16* __restrict__ q,        // [B, Hq, 128]
     const bf16* __restrict__ k_cache,  // [B, S_max, Hk, 128]
     const bf16* __restrict__ v_cache,  // [B, S_max, Hk, 128]
@@ -47,6 +53,8 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
     const float scale) {
   const int b = blockIdx.x;
   const int kvh = blockIdx.y;
+  const int split = blockIdx.z;
+  const int num_splits = gridDim.z;
   const int G = num_q_heads / num_kv_heads;
   const int ctx = context_lens[b];
 
@@ -88,7 +96,8 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
 
   __syncthreads();  // q_smem visible
 
-  for (int t0 = 0; t0 < ctx; t0 += TILE) {
+  // Interleave tiles across splits: split s takes tiles s, s+S, s+2S, ...
+  for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
     const int tn = min(TILE, ctx - t0);
 
     // --- cooperative staging: 128 row-loads (K+V) over 4 waves ---
@@ -159,24 +168,97 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
     __syncthreads();  // compute done before next tile overwrites LDS
   }
 
-  // --- finalize: owning wave writes its heads directly ---
-  for (int j = 0; j < heads_mine; ++j) {
-    const int g = wave + j * NUM_WAVES;
-    const float inv = s[j] > 0.0f ? 1.0f / s[j] : 0.0f;
-    bf16x2* orow = reinterpret_cast<bf16x2*>(
-        out + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM);
-    orow[lane] = bf16x2{f2bf(acc[j][0] * inv), f2bf(acc[j][1] * inv)};
+  // --- finalize ---
+  if (num_splits == 1) {
+    // fast path: owning wave writes its heads directly
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      const float inv = s[j] > 0.0f ? 1.0f / s[j] : 0.0f;
+      bf16x2* orow = reinterpret_cast<bf16x2*>(
+          out + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM);
+      orow[lane] = bf16x2{f2bf(acc[j][0] * inv), f2bf(acc[j][1] * inv)};
+    }
+  } else {
+    // partials → workspace [B, Hk, G, splits, 2+128]
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      float* wsp = workspace +
+          ((((long)b * num_kv_heads + kvh) * G + g) * num_splits + split) *
+              (2 + HEAD_DIM);
+      if (lane == 0) {
+        wsp[0] = m[j];
+        wsp[1] = s[j];
+      }
+      wsp[2 + 2 * lane] = acc[j][0];
+      wsp[2 + 2 * lane + 1] = acc[j][1];
+    }
   }
 }
 
+// Merge split-KV partials: one 64-thread wave per (b, q_head).
+__global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
+    bf16* __restrict__ out,          // [B, Hq, 128]
+    const float* __restrict__ workspace,  // [B, Hk, G, splits, 2+128]
+    const int num_q_heads,
+    const int num_kv_heads,
+    const int num_splits) {
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;  // query head
+  const int G = num_q_heads / num_kv_heads;
+  const int kvh = h / G;
+  const int g = h % G;
+  const int lane = threadIdx.x;
+
+  const float* base = workspace +
+      ((((long)b * num_kv_heads + kvh) * G + g) * (long)num_splits) *
+          (2 + HEAD_DIM);
+  float M = -INFINITY;
+  for (int sp = 0; sp < num_splits; ++sp) {
+    M = fmaxf(M, base[sp * (2 + HEAD_DIM)]);
+  }
+  float S = 0.0f, o0 = 0.0f, o1 = 0.0f;
+  for (int sp = 0; sp < num_splits; ++sp) {
+    const float* wsp = base + sp * (2 + HEAD_DIM);
+    const float mw = wsp[0];
+    if (mw == -INFINITY) continue;
+    const float f = __expf(mw - M);
+    S += wsp[1] * f;
+    o0 = fmaf(wsp[2 + 2 * lane], f, o0);
+    o1 = fmaf(wsp[2 + 2 * lane + 1], f, o1);
+  }
+  const float inv = S > 0.0f ? 1.0f / S : 0.0f;
+  bf16x2* orow = reinterpret_cast<bf16x2*>(
+      out + ((long)b * num_q_heads + h) * HEAD_DIM);
+  orow[lane] = bf16x2{f2bf(o0 * inv), f2bf(o1 * inv)};
+}
+
+extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
+                                          int max_ctx_hint) {
+  // Fill ~2 workgroups per CU; cap so each split still gets >= 2 tiles.
+  const int base = batch * num_kv_heads;
+  if (base >= 512) return 1;
+  int splits = 512 / base;
+  if (splits > 16) splits = 16;
+  const int max_useful = max_ctx_hint > 0 ? (max_ctx_hint + 2 * TILE - 1) / (2 * TILE) : splits;
+  if (splits > max_useful && max_useful >= 1) splits = max_useful;
+  return splits < 1 ? 1 : splits;
+}
+
 extern "C" void launch_gqa_decode_attn(
-    void* out, const void* q, const void* k_cache, const void* v_cache,
-    const int* context_lens, int batch, int num_q_heads, int num_kv_heads,
-    int max_seq, float scale, hipStream_t stream) {
-  dim3 grid(batch, num_kv_heads);
+    void* out, void* workspace, const void* q, const void* k_cache,
+    const void* v_cache, const int* context_lens, int batch, int num_q_heads,
+    int num_kv_heads, int max_seq, int num_splits, float scale,
+    hipStream_t stream) {
+  dim3 grid(batch, num_kv_heads, num_splits);
   dim3 block(256);
   hipLaunchKernelGGL(gqa_decode_attn_kernel, grid, block, 0, stream,
-                     (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
-                     (const bf16*)v_cache, context_lens, num_q_heads,
-                     num_kv_heads, max_seq, scale);
+                     (bf16*)out, (float*)workspace, (const bf16*)q,
+                     (const bf16*)k_cache, (const bf16*)v_cache, context_lens,
+                     num_q_heads, num_kv_heads, max_seq, scale);
+  if (num_splits > 1) {
+    dim3 mgrid(batch, num_q_heads);
+    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64), 0,
+                       stream, (bf16*)out, (const float*)workspace,
+                       num_q_heads, num_kv_heads, num_splits);
+  }
 }

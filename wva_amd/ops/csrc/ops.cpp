@@ -16,13 +16,15 @@ extern "C" void launch_rope(void* q, void* k, const int* positions, int tokens,
                             float theta, hipStream_t stream);
 extern "C" void launch_silu_mul(void* out, const void* gate, const void* up,
                                 long n, hipStream_t stream);
-extern "C" void launch_gqa_decode_attn(void* out, const void* q,
-                                       const void* k_cache,
+extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
+                                          int max_ctx_hint);
+extern "C" void launch_gqa_decode_attn(void* out, void* workspace,
+                                       const void* q, const void* k_cache,
                                        const void* v_cache,
                                        const int* context_lens, int batch,
                                        int num_q_heads, int num_kv_heads,
-                                       int max_seq, float scale,
-                                       hipStream_t stream);
+                                       int max_seq, int num_splits,
+                                       float scale, hipStream_t stream);
 
 namespace {
 
@@ -112,10 +114,22 @@ torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
   TORCH_CHECK(num_q_heads / num_kv_heads <= 8, "GQA group size must be <= 8");
   TORCH_CHECK(v_cache.sizes() == k_cache.sizes(), "k/v cache mismatch");
   auto out = torch::empty_like(q);
-  launch_gqa_decode_attn(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
-                         v_cache.data_ptr(), context_lens.data_ptr<int>(),
-                         batch, num_q_heads, num_kv_heads, max_seq,
-                         (float)scale, current_stream());
+  const int num_splits =
+      gqa_decode_attn_num_splits(batch, num_kv_heads, max_seq);
+  torch::Tensor workspace;
+  void* ws_ptr = nullptr;
+  if (num_splits > 1) {
+    const int G = num_q_heads / num_kv_heads;
+    workspace = torch::empty(
+        {(long)batch * num_kv_heads * G * num_splits * (2 + head_dim)},
+        q.options().dtype(torch::kFloat32));
+    ws_ptr = workspace.data_ptr();
+  }
+  launch_gqa_decode_attn(out.data_ptr(), ws_ptr, q.data_ptr(),
+                         k_cache.data_ptr(), v_cache.data_ptr(),
+                         context_lens.data_ptr<int>(), batch, num_q_heads,
+                         num_kv_heads, max_seq, num_splits, (float)scale,
+                         current_stream());
   return out;
 }
 
