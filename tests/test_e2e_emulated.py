@@ -304,3 +304,45 @@ class TestEndToEnd:
             sim.reconcile_deployments()
         deploy = cluster.get("Deployment", NS, VARIANT)
         assert deploy.replicas >= 2
+
+
+class TestInfernoAnalyzerPath:
+    def test_inferno_engine_path(self):
+        """analyzerName: inferno — SLO-derived capacity drives decisions."""
+        from wva_amd.analyzers.modelanalyzer import InfernoAnalyzer
+        from wva_amd.inferno.system import System
+        from wva_amd.inferno.types import (
+            AcceleratorSpec,
+            ModelAcceleratorPerfData,
+            ModelTarget,
+            ServiceClassSpec,
+            ServiceParmsSpec,
+            SystemData,
+        )
+
+        prof = ServiceProfile(
+            alpha_ms=50.0, beta_ms=2.0, max_num_seqs=8, num_gpu_blocks=500
+        )
+        cluster, sim, app = make_stack(
+            replicas=1, profile=prof, analyzer="inferno"
+        )
+        system = System(SystemData(
+            accelerators=[AcceleratorSpec(name="MI355X", type="MI355X", cost=50)],
+            models=[ModelAcceleratorPerfData(
+                name=MODEL, acc="MI355X", max_batch_size=8, at_tokens=50,
+                service_parms=ServiceParmsSpec(alpha=50.0, beta=2.0),
+            )],
+            service_classes=[ServiceClassSpec(
+                name="default", priority=1,
+                model_targets=[ModelTarget(model=MODEL, slo_itl=80.0,
+                                           slo_ttft=2000.0)],
+            )],
+        ))
+        app.saturation_engine.inferno_analyzer = InfernoAnalyzer(system)
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=20, seconds=30)
+        app.saturation_engine.optimize()
+        d = app.decision_cache.get(NS, VARIANT)
+        assert d is not None
+        # one tiny replica cannot serve 20 qps within SLO → scale up
+        assert d.target_replicas >= 2

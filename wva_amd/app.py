@@ -105,6 +105,7 @@ def build_app(
     source_registry.register(source)
     reg.register_saturation_queries(source_registry)
     reg.register_scale_to_zero_queries(source_registry)
+    reg.register_arrival_rate_query(source_registry)
 
     datastore = Datastore(
         cluster,

@@ -33,6 +33,7 @@ QUERY_PREFIX_CACHE_HIT_RATE = "prefix_cache_hit_rate"
 QUERY_SCHEDULER_QUEUE_SIZE = "scheduler_queue_size"
 QUERY_SCHEDULER_QUEUE_BYTES = "scheduler_queue_bytes"
 QUERY_MODEL_REQUEST_COUNT = "model_request_count"
+QUERY_MODEL_ARRIVAL_RATE = "model_arrival_rate"
 
 
 def register_saturation_queries(source_registry: SourceRegistry) -> None:
@@ -102,6 +103,35 @@ def register_saturation_queries(source_registry: SourceRegistry) -> None:
         params=[PARAM_MODEL_ID],
         description="Total bytes queued in scheduler flow control for this model",
     ))
+
+
+def register_arrival_rate_query(source_registry: SourceRegistry) -> None:
+    """Arrival (completion) rate for the Inferno SLO analyzer: at steady
+    state sum(rate(request_success_total)) equals the offered rate."""
+    src = source_registry.get(PROMETHEUS_SOURCE_NAME)
+    if src is None:
+        return
+    src.query_list().must_register(QueryTemplate(
+        name=QUERY_MODEL_ARRIVAL_RATE,
+        type=QUERY_TYPE_PROMQL,
+        template='sum(rate(vllm:request_success_total{namespace="{{.namespace}}",model_name="{{.modelID}}"}[2m]))',
+        params=[PARAM_NAMESPACE, PARAM_MODEL_ID],
+        description="Model-level request completion rate (req/s, 2m window)",
+    ))
+
+
+def collect_model_arrival_rate(
+    metrics_source: MetricsSource, model_id: str, namespace: str
+) -> Optional[float]:
+    """Observed request rate (req/s) or None when unavailable."""
+    results = metrics_source.refresh(RefreshSpec(
+        queries=[QUERY_MODEL_ARRIVAL_RATE],
+        params={PARAM_MODEL_ID: model_id, PARAM_NAMESPACE: namespace},
+    ))
+    result = results.get(QUERY_MODEL_ARRIVAL_RATE)
+    if result is None or result.has_error() or not result.values:
+        return None
+    return result.first_value().value
 
 
 def register_scale_to_zero_queries(source_registry: SourceRegistry) -> None:

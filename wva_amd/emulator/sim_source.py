@@ -121,6 +121,13 @@ class SimMetricsSource:
         if model is None:
             return []
 
+        if query == reg.QUERY_MODEL_ARRIVAL_RATE:
+            window = 120.0
+            cutoff = self.sim.now - window
+            count = sum(1 for c in model.completed if c.finish_time >= cutoff)
+            elapsed = min(window, self.sim.now) or 1.0
+            return [MetricValue(value=count / elapsed, timestamp=ts)]
+
         if query == reg.QUERY_MODEL_REQUEST_COUNT:
             retention = parse_go_duration(params.get("retentionPeriod", "10m"))
             cutoff = self.sim.now - retention

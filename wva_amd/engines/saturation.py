@@ -122,6 +122,10 @@ class SaturationEngine:
         )
         self.v1_analyzer = SaturationAnalyzerV1()
         self.v2_analyzer = SaturationAnalyzerV2(self.capacity_store)
+        # Optional Inferno SLO analyzer (analyzerName: "inferno"); the
+        # reference ships its Inferno library dormant — here it is a
+        # first-class engine path (wva_amd/analyzers/modelanalyzer.py)
+        self.inferno_analyzer = None
         self.optimizer = CostAwareOptimizer()
         self.executor = PollingExecutor(
             interval_seconds, self.optimize, name="saturation-engine"
@@ -155,9 +159,9 @@ class SaturationEngine:
         va_map = {f"{va.namespace}/{va.name}": va for va in active_vas}
 
         sat_cfg = self.config.saturation_config().apply_defaults()
-        use_v2 = sat_cfg.analyzer_name == "saturation"
-
-        if use_v2:
+        if sat_cfg.analyzer_name in ("saturation", "inferno"):
+            # token-based V2 and the Inferno SLO analyzer share the
+            # optimizer + enforcer-bridge flow; only the analyzer differs
             all_decisions = self._optimize_v2(model_groups)
         else:
             all_decisions = self._optimize_v1(model_groups)
@@ -332,7 +336,30 @@ class SaturationEngine:
                 namespace, model_id, va.name, accelerator, gpu_count, deploy
             )
         scheduler_queue = self.collector.collect_scheduler_queue_metrics(model_id)
-        return self.v2_analyzer.analyze(
+        analyzer = self.v2_analyzer
+        if config.analyzer_name == "inferno":
+            if self.inferno_analyzer is None:
+                log.error(
+                    "analyzerName=inferno but no Inferno system configured; "
+                    "falling back to the V2 token analyzer"
+                )
+            else:
+                analyzer = self.inferno_analyzer
+                from ..collector import registration as _reg
+
+                rate = _reg.collect_model_arrival_rate(
+                    self.collector.source, model_id, namespace
+                )
+                if rate is not None:
+                    avg_in = avg_out = 0.0
+                    ins = [m.avg_input_tokens for m in data.replica_metrics
+                           if m.avg_input_tokens > 0]
+                    outs = [m.avg_output_tokens for m in data.replica_metrics
+                            if m.avg_output_tokens > 0]
+                    avg_in = sum(ins) / len(ins) if ins else 100.0
+                    avg_out = sum(outs) / len(outs) if outs else 50.0
+                    analyzer.observe_load(model_id, rate, avg_in, avg_out)
+        return analyzer.analyze(
             AnalyzerInput(
                 model_id=model_id,
                 namespace=namespace,
