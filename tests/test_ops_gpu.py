@@ -136,3 +136,31 @@ class TestDecodeModel:
         )
         assert result.alpha_ms > 0
         assert profile.num_gpu_blocks > 0
+
+
+class TestFusedDecodeOps:
+    def test_rope_append_kv_matches_unfused(self, dev):
+        B, Hq, Hk, D, S = 5, 32, 8, 128, 64
+        qkv = torch.randn(B, (Hq + 2 * Hk) * D, device=dev, dtype=torch.bfloat16)
+        k_cache = torch.zeros(B, S, Hk, D, device=dev, dtype=torch.bfloat16)
+        v_cache = torch.zeros_like(k_cache)
+        pos = torch.tensor([0, 3, 10, 31, 63], device=dev, dtype=torch.int32)
+        q = ops.rope_append_kv(qkv, k_cache, v_cache, pos, Hq, Hk, 500000.0)
+
+        # unfused reference on CPU
+        qkv_c = qkv.cpu()
+        kc = torch.zeros(B, S, Hk, D, dtype=torch.bfloat16)
+        vc = torch.zeros_like(kc)
+        q_ref = ops.rope_append_kv(qkv_c, kc, vc, pos.cpu(), Hq, Hk, 500000.0)
+        torch.testing.assert_close(q.float().cpu(), q_ref.float(),
+                                   atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(k_cache.float().cpu(), kc.float(),
+                                   atol=5e-2, rtol=5e-2)
+        torch.testing.assert_close(v_cache.float().cpu(), vc.float(),
+                                   atol=1e-3, rtol=1e-3)
+
+    def test_silu_mul_fused_matches_split(self, dev):
+        gu = torch.randn(16, 2 * 14336, device=dev, dtype=torch.bfloat16)
+        out = ops.silu_mul_fused(gu)
+        ref = ops.silu_mul_ref(gu[:, :14336], gu[:, 14336:])
+        torch.testing.assert_close(out.float(), ref, atol=2e-2, rtol=2e-2)

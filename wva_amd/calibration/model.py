@@ -174,7 +174,6 @@ class LlamaDecodeModel:
         x = self.embed.index_select(0, token_ids)  # [B, H]
         residual: Optional[torch.Tensor] = None
 
-        batch_idx = torch.arange(B, device=self.device)
         for li, layer in enumerate(self.layers):
             if residual is None:
                 residual = x.clone()
@@ -183,20 +182,11 @@ class LlamaDecodeModel:
                 h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
 
             qkv = h @ layer.wqkv.t()  # hipBLASLt GEMM
-            q = qkv[:, : cfg.q_size].reshape(B, cfg.num_q_heads, cfg.head_dim)
-            k = qkv[:, cfg.q_size : cfg.q_size + cfg.kv_size].reshape(
-                B, cfg.num_kv_heads, cfg.head_dim
+            # fused: RoPE on the strided qkv row + KV-cache append
+            q = ops.rope_append_kv(
+                qkv, self.k_cache[li][:B], self.v_cache[li][:B],
+                positions, cfg.num_q_heads, cfg.num_kv_heads, cfg.rope_theta,
             )
-            v = qkv[:, cfg.q_size + cfg.kv_size :].reshape(
-                B, cfg.num_kv_heads, cfg.head_dim
-            )
-            q = q.contiguous()
-            k = k.contiguous()
-            ops.rope(q, k, positions, cfg.rope_theta)
-
-            # append KV at the current position
-            self.k_cache[li][batch_idx, positions.long()] = k
-            self.v_cache[li][batch_idx, positions.long()] = v
 
             ctx = positions + 1  # includes the new token
             attn = ops.gqa_decode_attn(
@@ -206,9 +196,7 @@ class LlamaDecodeModel:
 
             h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
             gate_up = h2 @ layer.w_gate_up.t()
-            gate = gate_up[:, : cfg.intermediate_size].contiguous()
-            up = gate_up[:, cfg.intermediate_size :].contiguous()
-            act = ops.silu_mul(gate, up)
+            act = ops.silu_mul_fused(gate_up)
             x = act @ layer.w_down.t()
 
         final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)

@@ -166,6 +166,50 @@ def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     return silu_mul_ref(gate, up).to(gate.dtype)
 
 
+def rope_append_kv(
+    qkv: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    positions: torch.Tensor,
+    num_q_heads: int,
+    num_kv_heads: int,
+    theta: float = 500000.0,
+) -> torch.Tensor:
+    """Decode fast path: strided qkv row -> RoPE'd q + KV-cache append.
+    Returns contiguous q [B, Hq, D]."""
+    if qkv.is_cuda:
+        return _require_ext().rope_append_kv(
+            qkv, k_cache, v_cache, positions.to(torch.int32),
+            num_q_heads, num_kv_heads, theta,
+        )
+    B = qkv.shape[0]
+    D = k_cache.shape[3]
+    q = qkv[:, : num_q_heads * D].reshape(B, num_q_heads, D).contiguous()
+    k = (
+        qkv[:, num_q_heads * D : (num_q_heads + num_kv_heads) * D]
+        .reshape(B, num_kv_heads, D)
+        .contiguous()
+    )
+    v = qkv[:, (num_q_heads + num_kv_heads) * D :].reshape(B, num_kv_heads, D)
+    qr, kr = rope_ref(q, k, positions, theta)
+    q.copy_(qr.to(q.dtype))
+    k.copy_(kr.to(k.dtype))
+    idx = torch.arange(B)
+    k_cache[idx, positions.long()] = k
+    v_cache[idx, positions.long()] = v.to(v_cache.dtype)
+    return q
+
+
+def silu_mul_fused(gate_up: torch.Tensor) -> torch.Tensor:
+    """SwiGLU on the fused [rows, 2*inter] gate_up GEMM output."""
+    if gate_up.is_cuda:
+        return _require_ext().silu_mul_fused(gate_up)
+    inter = gate_up.shape[1] // 2
+    return silu_mul_ref(gate_up[:, :inter], gate_up[:, inter:]).to(
+        gate_up.dtype
+    )
+
+
 def gqa_decode_attn(
     q: torch.Tensor,
     k_cache: torch.Tensor,

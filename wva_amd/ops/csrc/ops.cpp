@@ -16,6 +16,15 @@ extern "C" void launch_rope(void* q, void* k, const int* positions, int tokens,
                             float theta, hipStream_t stream);
 extern "C" void launch_silu_mul(void* out, const void* gate, const void* up,
                                 long n, hipStream_t stream);
+extern "C" void launch_silu_mul_fused(void* out, const void* gate_up,
+                                      long rows, long inter,
+                                      hipStream_t stream);
+extern "C" void launch_rope_append_kv(const void* qkv, void* q_out,
+                                      void* k_cache, void* v_cache,
+                                      const int* positions, int batch,
+                                      int num_q_heads, int num_kv_heads,
+                                      int head_dim, int max_seq, float theta,
+                                      hipStream_t stream);
 extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
                                           int max_ctx_hint);
 extern "C" void launch_gqa_decode_attn(void* out, void* workspace,
@@ -93,6 +102,44 @@ torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
   return out;
 }
 
+// Decode fast path: strided qkv row → RoPE'd q_out + cache append.
+torch::Tensor rope_append_kv(torch::Tensor qkv, torch::Tensor k_cache,
+                             torch::Tensor v_cache, torch::Tensor positions,
+                             int64_t num_q_heads, int64_t num_kv_heads,
+                             double theta) {
+  check_bf16_contig(qkv, "qkv");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(positions.is_cuda() && positions.scalar_type() == torch::kInt32,
+              "positions must be int32 on GPU");
+  TORCH_CHECK(qkv.dim() == 2, "qkv must be [B, (Hq+2Hk)*D]");
+  TORCH_CHECK(k_cache.dim() == 4, "k_cache must be [B, S, Hk, D]");
+  const int batch = qkv.size(0);
+  const int head_dim = k_cache.size(3);
+  const int max_seq = k_cache.size(1);
+  TORCH_CHECK(k_cache.size(2) == num_kv_heads, "Hk mismatch");
+  TORCH_CHECK(qkv.size(1) == (num_q_heads + 2 * num_kv_heads) * head_dim,
+              "qkv width mismatch");
+  auto q_out = torch::empty({batch, num_q_heads, head_dim}, qkv.options());
+  launch_rope_append_kv(qkv.data_ptr(), q_out.data_ptr(), k_cache.data_ptr(),
+                        v_cache.data_ptr(), positions.data_ptr<int>(), batch,
+                        (int)num_q_heads, (int)num_kv_heads, head_dim, max_seq,
+                        (float)theta, current_stream());
+  return q_out;
+}
+
+torch::Tensor silu_mul_fused(torch::Tensor gate_up) {
+  check_bf16_contig(gate_up, "gate_up");
+  TORCH_CHECK(gate_up.dim() == 2 && gate_up.size(1) % 4 == 0,
+              "gate_up must be [rows, 2*inter], inter even");
+  const long rows = gate_up.size(0);
+  const long inter = gate_up.size(1) / 2;
+  auto out = torch::empty({rows, inter}, gate_up.options());
+  launch_silu_mul_fused(out.data_ptr(), gate_up.data_ptr(), rows, inter,
+                        current_stream());
+  return out;
+}
+
 torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
                               torch::Tensor v_cache,
                               torch::Tensor context_lens, double scale) {
@@ -144,6 +191,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("theta") = 500000.0);
   m.def("silu_mul", &silu_mul, "Fused SwiGLU silu(gate)*up",
         py::arg("gate"), py::arg("up"));
+  m.def("silu_mul_fused", &silu_mul_fused,
+        "SwiGLU on the fused [rows, 2*inter] gate_up buffer",
+        py::arg("gate_up"));
+  m.def("rope_append_kv", &rope_append_kv,
+        "RoPE on strided qkv row + KV-cache append; returns contiguous q",
+        py::arg("qkv"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("positions"), py::arg("num_q_heads"), py::arg("num_kv_heads"),
+        py::arg("theta") = 500000.0);
   m.def("gqa_decode_attn", &gqa_decode_attn,
         "GQA decode attention over contiguous KV cache",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
