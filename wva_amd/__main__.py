@@ -19,7 +19,6 @@ from .app import build_app
 from .config.loader import load_config
 from .emulator.cluster_sim import ClusterSim
 from .emulator.sim_source import SimMetricsSource
-from .emulator.vllm_sim import ServiceProfile
 from .kube.fake import FakeCluster
 from .utils.logging import get_logger, setup_logging
 

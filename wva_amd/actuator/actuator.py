@@ -7,7 +7,6 @@ the tick.
 """
 from __future__ import annotations
 
-from typing import Optional
 
 from ..api.types import VariantAutoscaling
 from ..kube.fake import FakeCluster, NotFoundError

@@ -14,7 +14,7 @@ saturation-derived.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from ..api.types import VariantAutoscaling, utcnow
 from ..config.saturation import SaturationScalingConfig

@@ -13,7 +13,7 @@ import concurrent.futures
 import math
 import threading
 import time
-from typing import Callable, Dict, List, Optional
+from typing import Dict, List, Optional
 
 import requests
 

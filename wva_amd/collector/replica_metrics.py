@@ -12,7 +12,6 @@ guard needed at MI355X scale), TokensInUse = round(kv × capacity) clamped.
 from __future__ import annotations
 
 import math
-import time
 from typing import Dict, List, Optional
 
 from ..analyzers.interfaces import (

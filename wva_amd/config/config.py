@@ -9,14 +9,13 @@ holding the lock.
 from __future__ import annotations
 
 import threading
-from dataclasses import dataclass, field
-from typing import Dict, Optional
+from dataclasses import dataclass
+from typing import Dict
 
 from .saturation import SaturationScalingConfig
 from .scale_to_zero import (
     ScaleToZeroConfigData,
     is_scale_to_zero_enabled,
-    min_num_replicas,
     scale_to_zero_retention_seconds,
 )
 

@@ -14,8 +14,8 @@ values from wva_amd.calibration when run on hardware.
 from __future__ import annotations
 
 import random
-from dataclasses import dataclass, field
-from typing import Deque, Dict, List, Optional
+from dataclasses import dataclass
+from typing import Deque, Dict, List
 from collections import deque
 
 

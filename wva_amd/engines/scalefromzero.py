@@ -13,7 +13,6 @@ DecisionTrigger.
 from __future__ import annotations
 
 import concurrent.futures
-from typing import Optional
 
 from ..actuator.direct import DirectActuator
 from ..analyzers.interfaces import ACTION_SCALE_UP, VariantDecision

@@ -9,7 +9,7 @@ indicator.
 from __future__ import annotations
 
 import math
-from typing import Callable, List, Optional, Sequence, Tuple
+from typing import Callable, List, Sequence, Tuple
 
 EPSILON = 1e-6
 MAX_ITERATIONS = 100

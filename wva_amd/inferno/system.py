@@ -14,8 +14,8 @@ minNumReplicas at base cost. Transition penalty multiplies value by
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from dataclasses import dataclass
+from typing import Dict, List
 
 from .queue_analyzer import (
     Configuration,

@@ -8,7 +8,7 @@ AcceleratorSwitchFactor (transition penalty) 0.1.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 # defaults (pkg/config/defaults.go)
 SLO_PERCENTILE = 0.95

@@ -13,7 +13,6 @@ import math
 from typing import Dict, List, Optional
 
 from ..analyzers.interfaces import (
-    ACTION_NONE,
     ACTION_SCALE_DOWN,
     ACTION_SCALE_UP,
     AnalyzerResult,

@@ -23,7 +23,7 @@ from ..config.config import Config
 from ..engines.common import DecisionTrigger
 from ..kube.fake import FakeCluster, NotFoundError, WatchEvent
 from ..kube.objects import Lease
-from ..api.types import ObjectMeta, utcnow
+from ..api.types import ObjectMeta
 from ..utils.logging import get_logger
 
 log = get_logger("runtime.manager")
