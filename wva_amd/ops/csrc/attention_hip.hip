@@ -63,9 +63,8 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
   const int wave = threadIdx.x / WAVE_SIZE;
 
   __shared__ float q_smem[MAX_G][HEAD_DIM];
-  // double-buffered tiles: staging of tile i+1 overlaps compute on tile i
-  __shared__ unsigned int k_smem[2][TILE * ROW_DW];  // bf16x2-packed rows
-  __shared__ unsigned int v_smem[2][TILE * ROW_DW];
+  __shared__ unsigned int k_smem[TILE * ROW_DW];  // bf16x2-packed rows
+  __shared__ unsigned int v_smem[TILE * ROW_DW];
   __shared__ float p_smem[MAX_G][TILE];
 
   // Stage scaled q into LDS (fp32).
@@ -96,61 +95,24 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
       v_cache + (long)b * max_seq * num_kv_heads * HEAD_DIM +
       (long)kvh * HEAD_DIM);
 
-  const int tile_stride = num_splits * TILE;
+  __syncthreads();  // q_smem visible
 
-  // Each wave owns 32 of the 128 row-loads per tile (2·TILE rows of 64
-  // dwords, 4 waves): r = wave + j·4, j in [0,32); even r → K row r/2,
-  // odd → V row r/2. Loads land in registers first so the NEXT tile's
-  // HBM traffic overlaps THIS tile's compute (software pipelining).
-#define ROWS_PER_WAVE (2 * TILE / NUM_WAVES)
-  unsigned int stage_reg[ROWS_PER_WAVE];
-
-  auto issue_tile_loads = [&](int t0, int tn, unsigned int* dst_reg) {
-#pragma unroll 8
-    for (int j = 0; j < ROWS_PER_WAVE; ++j) {
-      const int r = wave + j * NUM_WAVES;
-      const int row = r >> 1;
-      if (row < tn) {
-        const long src = (long)(t0 + row) * kv_row_dw + lane;
-        dst_reg[j] = (r & 1) ? v_base[src] : k_base[src];
-      }
-    }
-  };
-  auto commit_tile = [&](int tn, unsigned int* src_reg, int buf) {
-#pragma unroll 8
-    for (int j = 0; j < ROWS_PER_WAVE; ++j) {
-      const int r = wave + j * NUM_WAVES;
-      const int row = r >> 1;
-      if (row < tn) {
-        if (r & 1) {
-          v_smem[buf][row * ROW_DW + lane] = src_reg[j];
-        } else {
-          k_smem[buf][row * ROW_DW + lane] = src_reg[j];
-        }
-      }
-    }
-  };
-
-  // Prologue: stage the first tile into buffer 0.
-  {
-    const int t0 = split * TILE;
-    if (t0 < ctx) {
-      const int tn = min(TILE, ctx - t0);
-      issue_tile_loads(t0, tn, stage_reg);
-      commit_tile(tn, stage_reg, 0);
-    }
-  }
-  __syncthreads();  // q_smem + first tile visible
-
-  int buf = 0;
-  for (int t0 = split * TILE; t0 < ctx; t0 += tile_stride) {
+  // Interleave tiles across splits: split s takes tiles s, s+S, s+2S, ...
+  for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
     const int tn = min(TILE, ctx - t0);
-    const int t0_next = t0 + tile_stride;
 
-    // issue next tile's global loads (registers; no wait) — overlapped
-    if (t0_next < ctx) {
-      issue_tile_loads(t0_next, min(TILE, ctx - t0_next), stage_reg);
+    // --- cooperative staging: 128 row-loads (K+V) over 4 waves ---
+    // row-load r in [0, 2·tn): even→K row r/2, odd→V row r/2.
+    for (int r = wave; r < 2 * tn; r += NUM_WAVES) {
+      const int row = r >> 1;
+      const long src = (long)(t0 + row) * kv_row_dw + lane;
+      if (r & 1) {
+        v_smem[row * ROW_DW + lane] = v_base[src];
+      } else {
+        k_smem[row * ROW_DW + lane] = k_base[src];
+      }
     }
+    __syncthreads();
 
     // --- scores + softmax + p staging, per owned head ---
     const bool live = lane < tn;
@@ -159,7 +121,7 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
       float score = -INFINITY;
       if (live) {
         score = 0.0f;
-        const unsigned int* krow = &k_smem[buf][lane * ROW_DW];
+        const unsigned int* krow = &k_smem[lane * ROW_DW];
         const float* qg = q_smem[g];
 #pragma unroll 4
         for (int d4 = 0; d4 < HEAD_DIM / 8; ++d4) {
@@ -195,7 +157,7 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
       const float* pg = p_smem[g];
       float a0 = acc[j][0], a1 = acc[j][1];
       for (int t = 0; t < tn; ++t) {
-        const unsigned int vw = v_smem[buf][t * ROW_DW + lane];
+        const unsigned int vw = v_smem[t * ROW_DW + lane];
         const bf16x2 vv = *reinterpret_cast<const bf16x2*>(&vw);
         const float p = pg[t];  // broadcast
         a0 = fmaf(p, bf2f(vv.x), a0);
@@ -204,13 +166,7 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
       acc[j][0] = a0;
       acc[j][1] = a1;
     }
-
-    // commit the overlapped loads into the other buffer and flip
-    if (t0_next < ctx) {
-      commit_tile(min(TILE, ctx - t0_next), stage_reg, buf ^ 1);
-    }
-    buf ^= 1;
-    __syncthreads();  // staged tile visible; compute of this tile done
+    __syncthreads();  // compute done before next tile overwrites LDS
   }
 
   // --- finalize ---
