@@ -98,6 +98,11 @@ class VariantAutoscalingReconciler:
                 REASON_TARGET_NOT_FOUND,
                 f"Scale target {va.get_scale_target_name()} not found",
             )
+            self.cluster.record_event(
+                va, "Warning", REASON_TARGET_NOT_FOUND,
+                f"scale target Deployment {va.get_scale_target_name()} "
+                "not found",
+            )
             self._patch_status(va)
             return  # no requeue — Deployment create watch will retrigger
 

@@ -51,6 +51,18 @@ def _meta(obj: Any):
     return obj.metadata
 
 
+@dataclass
+class Event:
+    """core/v1 Event subset (Recorder.Eventf analog)."""
+
+    namespace: str
+    involved_kind: str
+    involved_name: str
+    type: str  # "Normal" | "Warning"
+    reason: str
+    message: str
+
+
 class FakeCluster:
     """Thread-safe in-memory object store with watches."""
 
@@ -60,6 +72,7 @@ class FakeCluster:
         self._objects: Dict[Tuple[str, str, str], Any] = {}
         self._rv = 0
         self._watchers: List[Tuple[Optional[set], "queue.Queue[WatchEvent]"]] = []
+        self.events: List[Event] = []
 
     # --- internals ---
 
@@ -181,6 +194,22 @@ class FakeCluster:
             _meta(obj).resource_version = self._rv
             self._notify(MODIFIED, obj)
             return copy.deepcopy(obj)
+
+    # --- events (Recorder.Eventf analog) ---
+
+    def record_event(
+        self, obj: Any, event_type: str, reason: str, message: str
+    ) -> None:
+        m = _meta(obj)
+        with self._lock:
+            self.events.append(Event(
+                namespace=m.namespace,
+                involved_kind=_kind_of(obj),
+                involved_name=m.name,
+                type=event_type,
+                reason=reason,
+                message=message,
+            ))
 
     # --- watches ---
 
