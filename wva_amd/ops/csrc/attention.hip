@@ -234,10 +234,12 @@ __global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
 
 extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
                                           int max_ctx_hint) {
-  // Fill ~2 workgroups per CU; cap so each split still gets >= 2 tiles.
+  // Fill 4 resident workgroups per CU (the v3 kernel's admissible
+  // residency: 39 KiB LDS, 109 VGPRs) = 1024 WGs; cap so each split
+  // still gets >= 2 tiles of work.
   const int base = batch * num_kv_heads;
-  if (base >= 512) return 1;
-  int splits = 512 / base;
+  if (base >= 1024) return 1;
+  int splits = 1024 / base;
   if (splits > 16) splits = 16;
   const int max_useful = max_ctx_hint > 0 ? (max_ctx_hint + 2 * TILE - 1) / (2 * TILE) : splits;
   if (splits > max_useful && max_useful >= 1) splits = max_useful;
