@@ -111,12 +111,15 @@ def build_app(
         cluster,
         epp_bearer_token=config.epp_metric_reader_bearer_token(),
         scrape_fetch=scrape_fetch,
+        source_registry=source_registry,
     )
     emitter = MetricsEmitter(registry=metrics_registry)
     decision_cache = DecisionCache()
     decision_trigger = DecisionTrigger()
 
-    collector = ReplicaMetricsCollector(source, PodVAMapper(cluster))
+    collector = ReplicaMetricsCollector(
+        source, PodVAMapper(cluster), freshness=config.freshness
+    )
     actuator = Actuator(cluster, emitter)
     enforcer = Enforcer(
         lambda model_id, namespace, retention: reg.collect_model_request_count(

@@ -59,9 +59,15 @@ def _finite(v: float) -> bool:
 
 
 class ReplicaMetricsCollector:
-    def __init__(self, source: MetricsSource, pod_va_mapper: PodVAMapper):
+    def __init__(
+        self,
+        source: MetricsSource,
+        pod_va_mapper: PodVAMapper,
+        freshness=None,  # config.FreshnessThresholds (1m/2m/5m ladder)
+    ):
         self.source = source
         self.pod_va_mapper = pod_va_mapper
+        self.freshness = freshness
 
     def collect_replica_metrics(
         self,
@@ -90,6 +96,8 @@ class ReplicaMetricsCollector:
                 return None
             return pod_data.setdefault(name, _PodMetricData())
 
+        sample_ages: Dict[str, float] = {}
+
         kv = results.get(reg.QUERY_KV_CACHE_USAGE)
         if kv is not None:
             if kv.has_error():
@@ -99,6 +107,9 @@ class ReplicaMetricsCollector:
                 if d is not None:
                     d.kv_usage = v.value
                     d.has_kv = True
+                    name = _pod_name(v.labels)
+                    if name:
+                        sample_ages[name] = v.age_seconds()
 
         q = results.get(reg.QUERY_QUEUE_LENGTH)
         if q is not None:
@@ -204,8 +215,14 @@ class ReplicaMetricsCollector:
                     prefix_cache_hit_rate=data.prefix_cache_hit_rate,
                     metadata=ReplicaMetricsMetadata(
                         collected_at=collected_at,
-                        age_seconds=0.0,
-                        freshness_status="fresh",
+                        age_seconds=sample_ages.get(pod_name, 0.0),
+                        freshness_status=(
+                            self.freshness.determine_status(
+                                sample_ages.get(pod_name, 0.0)
+                            )
+                            if self.freshness is not None
+                            else "fresh"
+                        ),
                     ),
                 )
             )

@@ -21,10 +21,15 @@ class Datastore:
         cluster: FakeCluster,
         epp_bearer_token: str = "",
         scrape_fetch: Optional[FetchFunc] = None,
+        source_registry=None,
     ):
         self.cluster = cluster
         self.epp_bearer_token = epp_bearer_token
         self.scrape_fetch = scrape_fetch
+        # Optional collector SourceRegistry: pool sources register there
+        # under their pool name (reference datastore.go registers one
+        # PodScrapingSource per InferencePool in the registry)
+        self.source_registry = source_registry
         self._lock = threading.RLock()
         self._pools: Dict[str, EndpointPool] = {}
         self._pool_sources: Dict[str, PodScrapingSource] = {}
@@ -41,12 +46,15 @@ class Datastore:
         with self._lock:
             key = self._pool_key(pool.namespace, pool.name)
             self._pools[key] = pool
-            self._pool_sources[key] = PodScrapingSource(
+            source = PodScrapingSource(
                 self.cluster,
                 pool,
                 bearer_token=self.epp_bearer_token,
                 fetch=self.scrape_fetch,
             )
+            self._pool_sources[key] = source
+            if self.source_registry is not None:
+                self.source_registry.register(source)
 
     def pool_get(self, namespace: str, name: str) -> Optional[EndpointPool]:
         with self._lock:
@@ -56,7 +64,9 @@ class Datastore:
         with self._lock:
             key = self._pool_key(namespace, name)
             self._pools.pop(key, None)
-            self._pool_sources.pop(key, None)
+            source = self._pool_sources.pop(key, None)
+            if source is not None and self.source_registry is not None:
+                self.source_registry.unregister(source.name())
 
     def pool_source(
         self, namespace: str, name: str

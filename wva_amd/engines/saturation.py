@@ -25,7 +25,11 @@ from __future__ import annotations
 from typing import Dict, List, Optional
 
 from ..actuator.actuator import Actuator
-from ..analyzers.capacity_store import CapacityKnowledgeStore
+from ..analyzers.capacity_store import (
+    CAPACITY_EVICTION_TIMEOUT_S,
+    HISTORY_EVICTION_TIMEOUT_S,
+    CapacityKnowledgeStore,
+)
 from ..analyzers.interfaces import (
     ACTION_SCALE_DOWN,
     ACTION_SCALE_UP,
@@ -142,6 +146,11 @@ class SaturationEngine:
     # --- per-tick optimization ---
 
     def optimize(self) -> None:
+        # periodic eviction of stale capacity knowledge and k2 history
+        # (reference constants: 7d capacity, 24h history)
+        self.capacity_store.evict_stale(CAPACITY_EVICTION_TIMEOUT_S)
+        self.v2_analyzer.evict_stale_history(HISTORY_EVICTION_TIMEOUT_S)
+
         active_vas = active_variant_autoscalings(self.cluster)
         if not active_vas:
             log.debug("no active VariantAutoscalings found, skipping optimization")
