@@ -100,7 +100,11 @@ def build_app(
             config.prometheus.base_url,
             cache_ttl_seconds=config.cache.ttl_seconds,
             bearer_token=config.prometheus.bearer_token,
+            token_path=config.prometheus.token_path,
             verify_tls=not config.prometheus.insecure_skip_verify,
+            ca_cert_path=config.prometheus.ca_cert_path,
+            client_cert_path=config.prometheus.client_cert_path,
+            client_key_path=config.prometheus.client_key_path,
         )
     source_registry.register(source)
     reg.register_saturation_queries(source_registry)

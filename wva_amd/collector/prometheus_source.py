@@ -89,17 +89,34 @@ class PrometheusSource:
         query_timeout_seconds: float = DEFAULT_QUERY_TIMEOUT_SECONDS,
         cache_ttl_seconds: float = DEFAULT_CACHE_TTL_SECONDS,
         bearer_token: str = "",
+        token_path: str = "",
         verify_tls: bool = True,
+        ca_cert_path: str = "",
+        client_cert_path: str = "",
+        client_key_path: str = "",
         session: Optional[requests.Session] = None,
         max_retries: int = 3,
     ):
+        """TLS parity with reference internal/utils/tls.go: CA bundle,
+        mTLS client cert/key, bearer token (inline or mounted file)."""
         self.base_url = base_url.rstrip("/")
         self.query_timeout_seconds = query_timeout_seconds
         self._query_list = QueryList()
         self._cache = TTLCache(ttl_seconds=cache_ttl_seconds)
         self._session = session or requests.Session()
+        if not bearer_token and token_path:
+            try:
+                with open(token_path) as f:
+                    bearer_token = f.read().strip()
+            except OSError:
+                pass
         self._bearer_token = bearer_token
-        self._verify_tls = verify_tls
+        # requests `verify` accepts False or a CA bundle path
+        self._verify_tls = (
+            ca_cert_path if (verify_tls and ca_cert_path) else verify_tls
+        )
+        if client_cert_path and client_key_path:
+            self._session.cert = (client_cert_path, client_key_path)
         self._max_retries = max_retries
         self._lock = threading.Lock()
 
