@@ -335,10 +335,12 @@ def main() -> None:
 
     ms_per_step = elapsed * 1000.0 / args.steps
     if dist is not None:
-        t = torch.tensor([ms_per_step])
+        # NCCL all-reduces need device tensors; gloo wants CPU
+        red_dev = "cuda" if has_gpu else "cpu"
+        t = torch.tensor([ms_per_step], device=red_dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         ms_per_step = float(t[0])
-        s = torch.tensor([score, accuracy, slo])
+        s = torch.tensor([score, accuracy, slo], device=red_dev)
         dist.all_reduce(s, op=dist.ReduceOp.SUM)
         score, accuracy, slo = (float(v) / world_size for v in s)
 
