@@ -40,7 +40,7 @@ typedef __attribute__((ext_vector_type(4))) unsigned int uint4_t;
 // blockIdx.z splits; each split writes (m, s, acc[128]) partials to a
 // workspace [B, Hk, G, splits, 2 + 128] and a second kernel merges.
 // splits == 1 takes the direct-store fast path.
-__global__ __launch_bounds__(256, 6) void gqa_decode_attn_kernel(
+__global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
     bf16* __restrict__ out,            // [B, Hq, 128]
     float* __restrict__ workspace,     // [B, Hk, G, splits, 2+128] or null
     const bf16* __restrict__ q,        // [B, Hq, 128]
@@ -62,12 +62,8 @@ __global__ __launch_bounds__(256, 6) void gqa_decode_attn_kernel(
   const int wave = threadIdx.x / WAVE_SIZE;
 
   __shared__ float q_smem[MAX_G][HEAD_DIM];
-  // Only K is LDS-staged (read G times, once per head). V is read exactly
-  // once per workgroup in the PV sweep, in the same coalesced
-  // row-per-iteration pattern as staging would use — streaming it straight
-  // from global halves LDS (39 → 22 KiB) and raises residency to 6
-  // workgroups/CU (with the launch-bounds cap below).
   __shared__ unsigned int k_smem[TILE * ROW_DW];  // bf16x2-packed rows
+  __shared__ unsigned int v_smem[TILE * ROW_DW];
   __shared__ float p_smem[MAX_G][TILE];
 
   // Stage scaled q into LDS (fp32).
@@ -104,9 +100,16 @@ __global__ __launch_bounds__(256, 6) void gqa_decode_attn_kernel(
   for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
     const int tn = min(TILE, ctx - t0);
 
-    // --- cooperative K staging: 64 row-loads over 4 waves ---
-    for (int row = wave; row < tn; row += NUM_WAVES) {
-      k_smem[row * ROW_DW + lane] = k_base[(long)(t0 + row) * kv_row_dw + lane];
+    // --- cooperative staging: 128 row-loads (K+V) over 4 waves ---
+    // row-load r in [0, 2·tn): even→K row r/2, odd→V row r/2.
+    for (int r = wave; r < 2 * tn; r += NUM_WAVES) {
+      const int row = r >> 1;
+      const long src = (long)(t0 + row) * kv_row_dw + lane;
+      if (r & 1) {
+        v_smem[row * ROW_DW + lane] = v_base[src];
+      } else {
+        k_smem[row * ROW_DW + lane] = k_base[src];
+      }
     }
     __syncthreads();
 
@@ -153,8 +156,7 @@ __global__ __launch_bounds__(256, 6) void gqa_decode_attn_kernel(
       const float* pg = p_smem[g];
       float a0 = acc[j][0], a1 = acc[j][1];
       for (int t = 0; t < tn; ++t) {
-        // V streamed from global: one coalesced 256 B row per iteration
-        const unsigned int vw = v_base[(long)(t0 + t) * kv_row_dw + lane];
+        const unsigned int vw = v_smem[t * ROW_DW + lane];
         const bf16x2 vv = *reinterpret_cast<const bf16x2*>(&vw);
         const float p = pg[t];  // broadcast
         a0 = fmaf(p, bf2f(vv.x), a0);
@@ -236,8 +238,8 @@ extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
   // residency: 39 KiB LDS, 109 VGPRs) = 1024 WGs; cap so each split
   // still gets >= 2 tiles of work.
   const int base = batch * num_kv_heads;
-  if (base >= 1536) return 1;
-  int splits = 1536 / base;
+  if (base >= 1024) return 1;
+  int splits = 1024 / base;
   if (splits > 16) splits = 16;
   const int max_useful = max_ctx_hint > 0 ? (max_ctx_hint + 2 * TILE - 1) / (2 * TILE) : splits;
   if (splits > max_useful && max_useful >= 1) splits = max_useful;
