@@ -79,3 +79,22 @@ def attn_bench():
         kv_bytes = 2 * batch * ctx * hk * 128 * 2
         print(f"attn b={batch} hq={hq} hk={hk} ctx={ctx}: {us:.1f} us, "
               f"KV {kv_bytes/1e6:.1f} MB, {kv_bytes/us/1e3:.2f} TB/s")
+
+
+def calibrate_70b():
+    """Llama-3.1-70B on ONE MI355X: 140 GB bf16 weights fit in 288 GB HBM3E
+    (impossible on 192 GB MI300X without TP) — BASELINE config #5 evidence."""
+    import json
+    import torch
+    from wva_amd.calibration.itl_benchmark import calibrate_service_profile
+    from wva_amd.calibration.model import LLAMA_3_70B
+
+    profile, result = calibrate_service_profile(
+        LLAMA_3_70B, batch_sizes=[1, 4, 8, 16], context_len=256,
+        max_seq=512, iters=4,
+    )
+    print("llama-70b calibration:", result.to_json())
+    free, total = torch.cuda.mem_get_info()
+    print(f"hbm total={total/2**30:.0f}GiB")
+    with open("gpurun_out/calibration_70b.json", "w") as f:
+        f.write(result.to_json())
