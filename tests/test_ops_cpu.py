@@ -89,3 +89,23 @@ class TestReferenceOps:
         # → ≈ (0.9*288GB − ~18GB) / 128KiB ≈ 1.9M tokens
         assert 1_500_000 < tokens < 2_200_000
         assert blocks == tokens // 16
+
+
+class TestMoEReference:
+    def test_router_top2_weights_normalized(self):
+        import torch
+        from wva_amd.calibration.moe_model import TINY_MOE, MixtralDecodeModel
+
+        m = MixtralDecodeModel(TINY_MOE, max_batch=2, max_seq=16, device="cpu")
+        h2 = torch.randn(2, TINY_MOE.hidden_size, dtype=torch.bfloat16)
+        out = m._moe_mlp(m.layers[0], h2)
+        assert out.shape == h2.shape
+        assert torch.isfinite(out.float()).all()
+
+    def test_weight_bytes_mixtral_fits_mi355x(self):
+        from wva_amd.calibration.moe_model import MIXTRAL_8X7B
+
+        gb = MIXTRAL_8X7B.weight_bytes() / 2**30
+        # Mixtral-8x7B ≈ 47B params ≈ 87-94 GiB bf16: resident on 288 GB
+        # MI355X, NOT on 192 GB MI300X together with a useful KV budget
+        assert 80 < gb < 100

@@ -1,0 +1,215 @@
+"""Mixtral-family (sparse MoE) decode engine for MI355X calibration.
+
+BASELINE config #4 pairs Llama-3-8B with Mixtral-8×7B under service-class
+priorities; this engine provides the measured service profile for the MoE
+family the same way calibration/model.py does for dense Llama. Mixtral
+8×7B bf16 is ≈ 94 GB of weights — resident on a single 288 GB MI355X (a
+192 GB MI300X needs TP=2), which is itself an MI355X capacity statement
+the cost solver exploits.
+
+Architecture: Llama-style attention (same HIP kernels: fused add-RMSNorm,
+rope_append_kv, GQA decode attention) + top-2 routed expert MLPs. Expert
+GEMMs run per active expert through hipBLASLt (token-gather loop — the
+right shape for decode batches where each expert sees B·2/8 tokens).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+
+from .. import ops
+
+
+@dataclass
+class MixtralConfig:
+    name: str = "mistralai/Mixtral-8x7B"
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_q_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    vocab_size: int = 32000
+    rope_theta: float = 1000000.0
+    rms_eps: float = 1e-5
+    num_experts: int = 8
+    top_k: int = 2
+
+    @property
+    def q_size(self) -> int:
+        return self.num_q_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.num_kv_heads * self.head_dim
+
+    def kv_bytes_per_token(self, dtype_bytes: int = 2) -> int:
+        return 2 * self.num_layers * self.kv_size * dtype_bytes
+
+    def weight_bytes(self, dtype_bytes: int = 2) -> int:
+        per_layer = (
+            self.hidden_size * (self.q_size + 2 * self.kv_size)
+            + self.q_size * self.hidden_size
+            + self.num_experts * 3 * self.hidden_size * self.intermediate_size
+            + self.hidden_size * self.num_experts  # router
+            + 2 * self.hidden_size
+        )
+        return dtype_bytes * (
+            self.num_layers * per_layer
+            + 2 * self.vocab_size * self.hidden_size
+            + self.hidden_size
+        )
+
+
+MIXTRAL_8X7B = MixtralConfig()
+
+TINY_MOE = MixtralConfig(
+    name="tiny-mixtral",
+    hidden_size=512,
+    intermediate_size=1024,
+    num_layers=2,
+    num_q_heads=4,
+    num_kv_heads=1,
+    vocab_size=1000,
+    num_experts=4,
+    top_k=2,
+)
+
+
+class _MoELayer:
+    def __init__(self, cfg: MixtralConfig, device, dtype, gen):
+        h, std = cfg.hidden_size, 0.02
+
+        def w(rows, cols):
+            return torch.empty(rows, cols, device=device, dtype=dtype).normal_(
+                0.0, std, generator=gen
+            )
+
+        self.input_norm = torch.ones(h, device=device, dtype=dtype)
+        self.post_attn_norm = torch.ones(h, device=device, dtype=dtype)
+        self.wqkv = w(cfg.q_size + 2 * cfg.kv_size, h)
+        self.wo = w(h, cfg.q_size)
+        self.w_router = w(cfg.num_experts, h)
+        # experts: fused gate_up [E, 2I, H] and down [E, H, I]
+        self.w_gate_up = [
+            w(2 * cfg.intermediate_size, h) for _ in range(cfg.num_experts)
+        ]
+        self.w_down = [
+            w(h, cfg.intermediate_size) for _ in range(cfg.num_experts)
+        ]
+
+
+class MixtralDecodeModel:
+    """Decode-only MoE engine; same KV-cache layout as LlamaDecodeModel."""
+
+    def __init__(
+        self,
+        cfg: MixtralConfig,
+        max_batch: int = 256,
+        max_seq: int = 2048,
+        device: str = "cuda",
+        seed: int = 0,
+    ):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        self.dtype = torch.bfloat16
+        self.max_batch = max_batch
+        self.max_seq = max_seq
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed)
+
+        self.embed = torch.empty(
+            cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=self.dtype
+        ).normal_(0.0, 0.02, generator=gen)
+        self.layers: List[_MoELayer] = [
+            _MoELayer(cfg, self.device, self.dtype, gen)
+            for _ in range(cfg.num_layers)
+        ]
+        self.final_norm = torch.ones(
+            cfg.hidden_size, device=self.device, dtype=self.dtype
+        )
+        self.lm_head = torch.empty(
+            cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=self.dtype
+        ).normal_(0.0, 0.02, generator=gen)
+
+        self.k_cache = [
+            torch.zeros(
+                max_batch, max_seq, cfg.num_kv_heads, cfg.head_dim,
+                device=self.device, dtype=self.dtype,
+            )
+            for _ in range(cfg.num_layers)
+        ]
+        self.v_cache = [
+            torch.zeros_like(self.k_cache[0]) for _ in range(cfg.num_layers)
+        ]
+        self.context_lens = torch.zeros(
+            max_batch, dtype=torch.int32, device=self.device
+        )
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+
+    def reset(self, batch: int, context_len: int) -> None:
+        self.context_lens.zero_()
+        self.context_lens[:batch] = context_len
+        for layer in range(self.cfg.num_layers):
+            self.k_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
+            self.v_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
+
+    def _moe_mlp(self, layer: _MoELayer, h2: torch.Tensor) -> torch.Tensor:
+        cfg = self.cfg
+        B = h2.shape[0]
+        router_logits = (h2.float() @ layer.w_router.t().float())  # [B, E]
+        weights, selected = torch.topk(router_logits, cfg.top_k, dim=-1)
+        weights = torch.softmax(weights, dim=-1).to(h2.dtype)  # [B, K]
+
+        out = torch.zeros_like(h2)
+        for e in range(cfg.num_experts):
+            # tokens with expert e among their top-k
+            mask = selected == e  # [B, K]
+            token_idx, k_idx = mask.nonzero(as_tuple=True)
+            if token_idx.numel() == 0:
+                continue
+            x_e = h2.index_select(0, token_idx)
+            gate_up = x_e @ layer.w_gate_up[e].t()
+            act = ops.silu_mul_fused(gate_up)
+            y = act @ layer.w_down[e].t()
+            w_e = weights[token_idx, k_idx].unsqueeze(-1)
+            out.index_add_(0, token_idx, (y * w_e).to(out.dtype))
+        return out
+
+    @torch.no_grad()
+    def decode_step(self, token_ids: torch.Tensor) -> torch.Tensor:
+        cfg = self.cfg
+        B = token_ids.shape[0]
+        positions = self.context_lens[:B].clone()
+
+        x = self.embed.index_select(0, token_ids)
+        residual: Optional[torch.Tensor] = None
+
+        for li, layer in enumerate(self.layers):
+            if residual is None:
+                residual = x.clone()
+                h = ops.rmsnorm(x, layer.input_norm, None, cfg.rms_eps)
+            else:
+                h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
+
+            qkv = h @ layer.wqkv.t()
+            q = ops.rope_append_kv(
+                qkv, self.k_cache[li][:B], self.v_cache[li][:B],
+                positions, cfg.num_q_heads, cfg.num_kv_heads, cfg.rope_theta,
+            )
+            ctx = positions + 1
+            attn = ops.gqa_decode_attn(
+                q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
+            )
+            x = attn.reshape(B, cfg.q_size) @ layer.wo.t()
+
+            h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
+            x = self._moe_mlp(layer, h2)
+
+        final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)
+        logits = final @ self.lm_head.t()
+        self.context_lens[:B] += 1
+        return logits
