@@ -29,10 +29,15 @@ class Actuator:
             )
 
         deploy = retry_with_backoff(fetch, retry_on=NotFoundError, max_attempts=2)
-        if deploy.status.replicas >= 0:
+        # status first (actual state); a zero status with a non-zero spec
+        # means the controller hasn't caught up — use spec, matching the
+        # engine's BuildVariantStates fallback (engine.go:525-528)
+        if deploy.status.replicas > 0:
             return deploy.status.replicas
-        if deploy.replicas is not None:
+        if deploy.replicas:
             return deploy.replicas
+        if deploy.status.replicas == 0 and deploy.replicas == 0:
+            return 0
         return 1
 
     def emit_metrics(self, va: VariantAutoscaling) -> None:
