@@ -523,9 +523,12 @@ class SaturationEngine:
                 accelerator_name = update_va.status.desired_optimized_alloc.accelerator
                 reason = "No scaling decision (optimization loop)"
 
-            if not accelerator_name:
-                # Fallback: VA accelerator label (keeps MI355X-labeled VAs
-                # emitting metrics even before first metrics arrive)
+            if decision is not None and not accelerator_name:
+                # Fallback for real decisions only: VA accelerator label.
+                # No-decision refresh passes must NOT fabricate an
+                # accelerator — the reference skips status+metric emission
+                # then (engine.go:895-912), which keeps safety-net metrics
+                # from being overwritten with zeros.
                 accelerator_name = va.metadata.labels.get(ACCELERATOR_LABEL_KEY, "")
 
             if not accelerator_name:
