@@ -203,3 +203,17 @@ class TestMoEDecodeModel:
         torch.testing.assert_close(
             lg.float().cpu(), lc.float(), atol=0.5, rtol=0.1
         )
+
+
+class TestTPModel:
+    def test_tp1_degenerate_matches_shapes(self, dev):
+        """TP=1 (no process group): runs the sharded engine end to end."""
+        from wva_amd.calibration.model import TINY
+        from wva_amd.calibration.tp_model import TPLlamaDecodeModel
+
+        model = TPLlamaDecodeModel(TINY, max_batch=2, max_seq=32)
+        model.reset(2, 8)
+        tokens = torch.randint(0, TINY.vocab_size, (2,), device=dev)
+        logits = model.decode_step(tokens)
+        assert logits.shape == (2, TINY.vocab_size)
+        assert torch.isfinite(logits.float()).all()

@@ -1,0 +1,52 @@
+"""TP decode model: 2-process gloo test on CPU (world_size 2) — the
+multi-process coverage for the RCCL/xGMI TP path, per the environment's
+gloo-based distributed testing contract.
+"""
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = '''
+import os, sys, torch, torch.distributed as dist
+sys.path.insert(0, {repo!r})
+from wva_amd.calibration.model import TINY
+from wva_amd.calibration.tp_model import TPLlamaDecodeModel
+dist.init_process_group("gloo")
+model = TPLlamaDecodeModel(TINY, max_batch=2, max_seq=32, device="cpu")
+model.reset(2, 8)
+torch.manual_seed(0)
+tokens = torch.randint(0, TINY.vocab_size, (2,))
+logits = model.decode_step(tokens)
+assert logits.shape == (2, TINY.vocab_size)
+assert torch.isfinite(logits.float()).all()
+# shard sanity: each rank holds half the heads
+assert model.shard.num_q_heads == TINY.num_q_heads // 2
+if dist.get_rank() == 0:
+    print("TP2_OK")
+dist.destroy_process_group()
+'''
+
+
+@pytest.mark.timeout(300)
+def test_tp2_gloo_decode():
+    script = WORKER.format(repo=REPO)
+    import tempfile
+
+    with tempfile.NamedTemporaryFile("w", suffix=".py", delete=False) as f:
+        f.write(script)
+        path = f.name
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29542",
+            path,
+        ],
+        capture_output=True, text=True, timeout=280,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "TP2_OK" in out.stdout
