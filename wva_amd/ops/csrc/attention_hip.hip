@@ -32,7 +32,13 @@
 #define MAX_G 8
 #define TILE 64            // positions per shared tile (= wave width)
 #define NUM_WAVES 4
-#define ROW_DW 65          // LDS row stride in dwords (64 + 1 pad)
+// LDS row stride in dwords: 66 = 64 + 2 pad. Even stride keeps every
+// 8-byte access provably aligned (ds_read/write_b64) and banking stays
+// conflict-free: b64 groups see banks (2·row + 2·d) mod 64 — distinct
+// per lane; b32 PV reads see (2·t + lane) mod 32 — distinct per lane.
+// (A 65-dword pad is conflict-free too but leaves every odd row 4-byte
+// aligned, so the compiler silently splits vector LDS accesses.)
+#define ROW_DW 66
 
 typedef __attribute__((ext_vector_type(4))) unsigned int uint4_t;
 
@@ -101,15 +107,27 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
   for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
     const int tn = min(TILE, ctx - t0);
 
-    // --- cooperative staging: 128 row-loads (K+V) over 4 waves ---
-    // row-load r in [0, 2·tn): even→K row r/2, odd→V row r/2.
-    for (int r = wave; r < 2 * tn; r += NUM_WAVES) {
-      const int row = r >> 1;
-      const long src = (long)(t0 + row) * kv_row_dw + lane;
-      if (r & 1) {
-        v_smem[row * ROW_DW + lane] = v_base[src];
-      } else {
-        k_smem[row * ROW_DW + lane] = k_base[src];
+    // --- cooperative staging, dwordx2-wide: each wave instruction moves
+    // TWO rows (lane l → row pair member l/32, dwords 2·(l%32)); 128 K+V
+    // rows over 4 waves = 16 wide loads per wave per tile (vs 32 narrow)
+    // with 8 B per lane in flight per instruction ---
+    {
+      const int sub_row = lane >> 5;        // 0..1 within the pair
+      const int d2 = (lane & 31) * 2;       // dword offset in the row
+      const int pairs = (2 * tn + 1) >> 1;  // pairs of rows across K+V
+      for (int pr = wave; pr < pairs; pr += NUM_WAVES) {
+        // flat row index in [0, 2·tn): first tn rows = K, next tn = V
+        const int flat = pr * 2 + sub_row;
+        if (flat < 2 * tn) {
+          const bool is_v = flat >= tn;
+          const int row = is_v ? flat - tn : flat;
+          const unsigned int* base = is_v ? v_base : k_base;
+          typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
+          const uint2_t val = *reinterpret_cast<const uint2_t*>(
+              &base[(long)(t0 + row) * kv_row_dw + d2]);
+          unsigned int* dst = (is_v ? v_smem : k_smem) + row * ROW_DW + d2;
+          *reinterpret_cast<uint2_t*>(dst) = val;
+        }
       }
     }
     __syncthreads();
@@ -123,16 +141,18 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
         score = 0.0f;
         const unsigned int* krow = &k_smem[lane * ROW_DW];
         const float* qg = q_smem[g];
-#pragma unroll 4
-        for (int d4 = 0; d4 < HEAD_DIM / 8; ++d4) {
-          uint4_t kv4 = *reinterpret_cast<const uint4_t*>(&krow[d4 * 4]);
-          unsigned int kw[4] = {kv4[0], kv4[1], kv4[2], kv4[3]};
+        typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
+#pragma unroll 8
+        for (int d2 = 0; d2 < HEAD_DIM / 4; ++d2) {
+          // aligned ds_read_b64: 4 bf16 K elements
+          const uint2_t kv2 = *reinterpret_cast<const uint2_t*>(&krow[d2 * 2]);
+          unsigned int kw[2] = {kv2[0], kv2[1]};
           float partial = 0.0f;
 #pragma unroll
-          for (int w2 = 0; w2 < 4; ++w2) {
+          for (int w2 = 0; w2 < 2; ++w2) {
             const bf16x2 p2 = *reinterpret_cast<const bf16x2*>(&kw[w2]);
-            partial = fmaf(bf2f(p2.x), qg[d4 * 8 + 2 * w2], partial);
-            partial = fmaf(bf2f(p2.y), qg[d4 * 8 + 2 * w2 + 1], partial);
+            partial = fmaf(bf2f(p2.x), qg[d2 * 4 + 2 * w2], partial);
+            partial = fmaf(bf2f(p2.y), qg[d2 * 4 + 2 * w2 + 1], partial);
           }
           score += partial;
         }
