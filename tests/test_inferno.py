@@ -373,3 +373,40 @@ class TestTuner:
         # stash consumed, covariance widened — next good update accepted
         assert tuner._stash is None
         assert tuner.update(self._observe(true, 100))
+
+
+class TestServiceClassConfigMap:
+    def test_reference_chart_format_loads(self):
+        """Byte-format parity with the reference chart's service-class
+        ConfigMap (slo-tpot/slo-ttft keys, priority, per-model data)."""
+        from wva_amd.inferno.types import parse_service_class_configmap
+
+        data = {
+            "premium.yaml": (
+                "name: Premium\n"
+                "priority: 1\n"
+                "data:\n"
+                "  - model: default/default\n"
+                "    slo-tpot: 24\n"
+                "    slo-ttft: 500\n"
+                "  - model: meta/llama0-70b\n"
+                "    slo-tpot: 80\n"
+                "    slo-ttft: 500\n"
+            ),
+            "freemium.yaml": (
+                "name: Freemium\n"
+                "priority: 10\n"
+                "data:\n"
+                "  - model: ibm/granite-13b\n"
+                "    slo-tpot: 200\n"
+                "    slo-ttft: 2000\n"
+            ),
+        }
+        classes = parse_service_class_configmap(data)
+        by_name = {c.name: c for c in classes}
+        assert by_name["Premium"].priority == 1
+        assert by_name["Freemium"].priority == 10
+        t = by_name["Premium"].model_targets[0]
+        assert t.model == "default/default"
+        assert t.slo_itl == 24.0
+        assert t.slo_ttft == 500.0

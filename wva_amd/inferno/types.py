@@ -90,10 +90,14 @@ class ModelTarget:
 
     @classmethod
     def from_dict(cls, d: Dict[str, Any]) -> "ModelTarget":
+        # accepts the reference chart's service-class keys: slo-tpot
+        # (time-per-output-token = ITL), slo-ttft, slo-tps
+        itl = d.get("slo-tpot", d.get("slo-itl", d.get("SLO_ITL", 0.0)))
+        ttft = d.get("slo-ttft", d.get("slo-ttw", d.get("SLO_TTFT", 0.0)))
         return cls(
             model=d.get("model", ""),
-            slo_itl=float(d.get("slo-itl", d.get("SLO_ITL", 0.0)) or 0.0),
-            slo_ttft=float(d.get("slo-ttw", d.get("slo-ttft", d.get("SLO_TTFT", 0.0))) or 0.0),
+            slo_itl=float(itl or 0.0),
+            slo_ttft=float(ttft or 0.0),
             slo_tps=float(d.get("slo-tps", d.get("SLO_TPS", 0.0)) or 0.0),
         )
 
@@ -220,3 +224,21 @@ class SystemData:
             ),
             optimizer=OptimizerSpec.from_dict(spec.get("optimizer") or {}),
         )
+
+
+def parse_service_class_configmap(data: Dict[str, str]):
+    """Parse the reference's service-classes ConfigMap data section: each
+    key is a YAML doc {name, priority, data: [{model, slo-tpot, slo-ttft}]}
+    (charts/.../wva-configmap-service-class.yaml)."""
+    import yaml
+
+    out = []
+    for key in sorted(data or {}):
+        try:
+            parsed = yaml.safe_load(data[key]) or {}
+        except Exception:  # noqa: BLE001
+            continue
+        if not isinstance(parsed, dict) or "name" not in parsed:
+            continue
+        out.append(ServiceClassSpec.from_dict(parsed))
+    return out
