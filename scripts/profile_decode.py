@@ -63,8 +63,8 @@ def attn_bench():
                                (256, 32, 8, 512), (8, 64, 8, 4096)]:
         S = ctx
         q = torch.randn(batch, hq, 128, device="cuda", dtype=torch.bfloat16)
-        k = torch.randn(batch, S, hk, 128, device="cuda", dtype=torch.bfloat16)
-        v = torch.randn(batch, S, hk, 128, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(batch, hk, S, 128, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(batch, hk, S, 128, device="cuda", dtype=torch.bfloat16)
         lens = torch.full((batch,), ctx, device="cuda", dtype=torch.int32)
         for _ in range(3):
             ops.gqa_decode_attn(q, k, v, lens)

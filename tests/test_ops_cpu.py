@@ -43,8 +43,8 @@ class TestReferenceOps:
         # all V rows identical → output equals that row regardless of scores
         B, S, Hk, Hq, D = 1, 16, 2, 4, 128
         q = torch.randn(B, Hq, D)
-        k = torch.randn(B, S, Hk, D)
-        v = torch.ones(B, S, Hk, D) * 0.5
+        k = torch.randn(B, Hk, S, D)
+        v = torch.ones(B, Hk, S, D) * 0.5
         lens = torch.tensor([10], dtype=torch.int32)
         out = ops.gqa_decode_attn_ref(q, k, v, lens, 1.0 / math.sqrt(D))
         torch.testing.assert_close(
@@ -55,11 +55,11 @@ class TestReferenceOps:
         B, S, Hk, Hq, D = 1, 8, 1, 1, 128
         q = torch.zeros(B, Hq, D)
         q[0, 0, 0] = 10.0
-        k = torch.zeros(B, S, Hk, D)
-        v = torch.zeros(B, S, Hk, D)
+        k = torch.zeros(B, Hk, S, D)
+        v = torch.zeros(B, Hk, S, D)
         # position 5 (beyond ctx=4) has huge value — must not leak
-        k[0, 5, 0, 0] = 100.0
-        v[0, 5, 0, :] = 999.0
+        k[0, 0, 5, 0] = 100.0
+        v[0, 0, 5, :] = 999.0
         lens = torch.tensor([4], dtype=torch.int32)
         out = ops.gqa_decode_attn_ref(q, k, v, lens, 1.0)
         assert out.abs().max() < 1.0

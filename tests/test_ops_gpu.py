@@ -96,8 +96,8 @@ class TestDecodeAttention:
     def test_matches_reference(self, dev, batch, hq, hk, ctx):
         D, S = 128, 512
         q = torch.randn(batch, hq, D, device=dev, dtype=torch.bfloat16)
-        k = torch.randn(batch, S, hk, D, device=dev, dtype=torch.bfloat16)
-        v = torch.randn(batch, S, hk, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(batch, hk, S, D, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(batch, hk, S, D, device=dev, dtype=torch.bfloat16)
         lens = torch.full((batch,), ctx, device=dev, dtype=torch.int32)
         scale = 1.0 / math.sqrt(D)
         out = ops.gqa_decode_attn(q, k, v, lens, scale)
@@ -107,8 +107,8 @@ class TestDecodeAttention:
     def test_varied_context_lens(self, dev):
         D, S, batch = 128, 256, 3
         q = torch.randn(batch, 32, D, device=dev, dtype=torch.bfloat16)
-        k = torch.randn(batch, S, 8, D, device=dev, dtype=torch.bfloat16)
-        v = torch.randn(batch, S, 8, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(batch, 8, S, D, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(batch, 8, S, D, device=dev, dtype=torch.bfloat16)
         lens = torch.tensor([1, 100, 256], device=dev, dtype=torch.int32)
         out = ops.gqa_decode_attn(q, k, v, lens)
         ref = ops.gqa_decode_attn_ref(q, k, v, lens, 1.0 / math.sqrt(D))
@@ -142,14 +142,14 @@ class TestFusedDecodeOps:
     def test_rope_append_kv_matches_unfused(self, dev):
         B, Hq, Hk, D, S = 5, 32, 8, 128, 64
         qkv = torch.randn(B, (Hq + 2 * Hk) * D, device=dev, dtype=torch.bfloat16)
-        k_cache = torch.zeros(B, S, Hk, D, device=dev, dtype=torch.bfloat16)
+        k_cache = torch.zeros(B, Hk, S, D, device=dev, dtype=torch.bfloat16)
         v_cache = torch.zeros_like(k_cache)
         pos = torch.tensor([0, 3, 10, 31, 63], device=dev, dtype=torch.int32)
         q = ops.rope_append_kv(qkv, k_cache, v_cache, pos, Hq, Hk, 500000.0)
 
         # unfused reference on CPU
         qkv_c = qkv.cpu()
-        kc = torch.zeros(B, S, Hk, D, dtype=torch.bfloat16)
+        kc = torch.zeros(B, Hk, S, D, dtype=torch.bfloat16)
         vc = torch.zeros_like(kc)
         q_ref = ops.rope_append_kv(qkv_c, kc, vc, pos.cpu(), Hq, Hk, 500000.0)
         torch.testing.assert_close(q.float().cpu(), q_ref.float(),

@@ -140,10 +140,12 @@ class LlamaDecodeModel:
             cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=self.dtype
         ).normal_(0.0, 0.02, generator=gen)
 
-        # Contiguous KV cache per layer: [B, S, Hk, D]
+        # Head-major contiguous KV cache per layer: [B, Hk, S, D] — each
+        # (sequence, kv-head) is one sequential HBM stream for the
+        # attention kernel's tile staging (docs/mi355x-kernels.md)
         self.k_cache = [
             torch.zeros(
-                max_batch, max_seq, cfg.num_kv_heads, cfg.head_dim,
+                max_batch, cfg.num_kv_heads, max_seq, cfg.head_dim,
                 device=self.device, dtype=self.dtype,
             )
             for _ in range(cfg.num_layers)
@@ -159,8 +161,8 @@ class LlamaDecodeModel:
         self.context_lens.zero_()
         self.context_lens[:batch] = context_len
         for layer in range(self.cfg.num_layers):
-            self.k_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
-            self.v_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
+            self.k_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+            self.v_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
 
     @torch.no_grad()
     def decode_step(self, token_ids: torch.Tensor) -> torch.Tensor:

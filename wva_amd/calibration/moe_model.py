@@ -137,7 +137,7 @@ class MixtralDecodeModel:
 
         self.k_cache = [
             torch.zeros(
-                max_batch, max_seq, cfg.num_kv_heads, cfg.head_dim,
+                max_batch, cfg.num_kv_heads, max_seq, cfg.head_dim,
                 device=self.device, dtype=self.dtype,
             )
             for _ in range(cfg.num_layers)
@@ -154,8 +154,8 @@ class MixtralDecodeModel:
         self.context_lens.zero_()
         self.context_lens[:batch] = context_len
         for layer in range(self.cfg.num_layers):
-            self.k_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
-            self.v_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
+            self.k_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+            self.v_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
 
     def _moe_mlp(self, layer: _MoELayer, h2: torch.Tensor) -> torch.Tensor:
         cfg = self.cfg

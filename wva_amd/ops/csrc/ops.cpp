@@ -113,11 +113,11 @@ torch::Tensor rope_append_kv(torch::Tensor qkv, torch::Tensor k_cache,
   TORCH_CHECK(positions.is_cuda() && positions.scalar_type() == torch::kInt32,
               "positions must be int32 on GPU");
   TORCH_CHECK(qkv.dim() == 2, "qkv must be [B, (Hq+2Hk)*D]");
-  TORCH_CHECK(k_cache.dim() == 4, "k_cache must be [B, S, Hk, D]");
+  TORCH_CHECK(k_cache.dim() == 4, "k_cache must be [B, Hk, S, D]");
   const int batch = qkv.size(0);
   const int head_dim = k_cache.size(3);
-  const int max_seq = k_cache.size(1);
-  TORCH_CHECK(k_cache.size(2) == num_kv_heads, "Hk mismatch");
+  const int max_seq = k_cache.size(2);
+  TORCH_CHECK(k_cache.size(1) == num_kv_heads, "Hk mismatch");
   TORCH_CHECK(qkv.size(1) == (num_q_heads + 2 * num_kv_heads) * head_dim,
               "qkv width mismatch");
   auto q_out = torch::empty({batch, num_q_heads, head_dim}, qkv.options());
@@ -150,12 +150,13 @@ torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
                   context_lens.scalar_type() == torch::kInt32,
               "context_lens must be int32 on GPU");
   TORCH_CHECK(q.dim() == 3, "q must be [B, Hq, D]");
-  TORCH_CHECK(k_cache.dim() == 4, "k_cache must be [B, S, Hk, D]");
+  TORCH_CHECK(k_cache.dim() == 4,
+              "k_cache must be [B, Hk, S, D] (head-major)");
   const int batch = q.size(0);
   const int num_q_heads = q.size(1);
   const int head_dim = q.size(2);
-  const int max_seq = k_cache.size(1);
-  const int num_kv_heads = k_cache.size(2);
+  const int num_kv_heads = k_cache.size(1);
+  const int max_seq = k_cache.size(2);
   TORCH_CHECK(head_dim == 128, "head_dim must be 128 (Llama-3 family)");
   TORCH_CHECK(num_q_heads % num_kv_heads == 0, "Hq must divide by Hk");
   TORCH_CHECK(num_q_heads / num_kv_heads <= 8, "GQA group size must be <= 8");

@@ -57,8 +57,8 @@ __global__ void rope_kernel(
 __global__ void rope_append_kv_kernel(
     const bf16* __restrict__ qkv,   // [B, (Hq + 2·Hk) · D]
     bf16* __restrict__ q_out,       // [B, Hq, D]
-    bf16* __restrict__ k_cache,     // [B, S_max, Hk, D]
-    bf16* __restrict__ v_cache,     // [B, S_max, Hk, D]
+    bf16* __restrict__ k_cache,     // [B, Hk, S_max, D] (head-major)
+    bf16* __restrict__ v_cache,     // [B, Hk, S_max, D]
     const int* __restrict__ positions,  // [B] append position per sequence
     const int num_q_heads,
     const int num_kv_heads,
@@ -84,7 +84,7 @@ __global__ void rope_append_kv_kernel(
       const int kh = head - num_q_heads;
       const bf16* src = qkv + qkv_row + ((long)num_q_heads + kh) * head_dim;
       bf16* dst = k_cache +
-          (((long)b * max_seq + pos_i) * num_kv_heads + kh) * head_dim;
+          (((long)b * num_kv_heads + kh) * max_seq + pos_i) * head_dim;
       rotate_pair(src, dst, d, half, head_dim, pos, theta);
     }
   }
@@ -96,7 +96,7 @@ __global__ void rope_append_kv_kernel(
     const int kh = idx / half;
     const int d2 = idx % half;
     bf16x2* dst = reinterpret_cast<bf16x2*>(
-        v_cache + (((long)b * max_seq + pos_i) * num_kv_heads + kh) * head_dim);
+        v_cache + (((long)b * num_kv_heads + kh) * max_seq + pos_i) * head_dim);
     dst[d2] = vsrc[(long)kh * half + d2];
   }
 }
