@@ -177,6 +177,25 @@ def build_app(
         inferencepool_predicate(),
         inferencepool_reconciler.reconcile,
     )
+
+    # ServiceMonitor deletion → observability warning (controller:330-367):
+    # without the monitor, Prometheus stops scraping vLLM pods and the
+    # saturation metrics go stale.
+    def _servicemonitor_warning(ns: str, name: str) -> None:
+        if cluster.try_get("ServiceMonitor", ns, name) is None:
+            log.warning(
+                "ServiceMonitor %s/%s deleted — vLLM metrics scraping will "
+                "stop and saturation decisions will degrade to the safety net",
+                ns, name,
+            )
+
+    from .kube.fake import DELETED as _DELETED
+
+    manager.register_reconciler(
+        ["ServiceMonitor"],
+        lambda e: e.kind == "ServiceMonitor" and e.type == _DELETED,
+        _servicemonitor_warning,
+    )
     if start_engines:
         manager.add_runnable(saturation_engine)
         manager.add_runnable(scale_from_zero_engine)

@@ -210,6 +210,27 @@ class EndpointPool:
 
 
 @dataclass
+class ServiceMonitor:
+    """monitoring.coreos.com/v1 ServiceMonitor marker object (watched for
+    deletion only — losing it breaks Prometheus scraping of vLLM pods,
+    reference variantautoscaling_controller.go:330-367)."""
+
+    metadata: ObjectMeta = field(default_factory=ObjectMeta)
+    selector: Dict[str, str] = field(default_factory=dict)
+
+    kind: str = "ServiceMonitor"
+    api_version: str = "monitoring.coreos.com/v1"
+
+    @property
+    def name(self) -> str:
+        return self.metadata.name
+
+    @property
+    def namespace(self) -> str:
+        return self.metadata.namespace
+
+
+@dataclass
 class Lease:
     """coordination.k8s.io/v1 Lease used for leader election."""
 
