@@ -1,0 +1,91 @@
+"""Property-based tests (hypothesis) for the parsing and queueing-math
+surfaces where adversarial inputs are realistic: vLLM arg strings,
+durations, PromQL escaping, queueing-model invariants.
+"""
+from hypothesis import given, settings, strategies as st
+
+from wva_amd.analyzers.deployment_parser import (
+    parse_vllm_args,
+    split_shell_string,
+)
+from wva_amd.collector.query_template import escape_promql_value
+from wva_amd.config.scale_to_zero import parse_go_duration
+from wva_amd.inferno.queueing import MM1KModel, MM1StateDependentModel
+from wva_amd.kube.objects import Container, Deployment, PodTemplateSpec
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.text(max_size=200))
+def test_split_shell_string_never_raises(s):
+    tokens = split_shell_string(s)
+    assert all(isinstance(t, str) for t in tokens)
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.lists(st.text(max_size=40), max_size=12))
+def test_parse_vllm_args_never_raises(args):
+    d = Deployment(template=PodTemplateSpec(containers=[Container(args=args)]))
+    p = parse_vllm_args(d)
+    assert p.effective_max_batched_tokens > 0
+    assert p.block_size > 0
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.text(max_size=100))
+def test_promql_escape_roundtrip_safe(s):
+    escaped = escape_promql_value(s)
+    # escaped value must not contain an unescaped double quote
+    i = 0
+    while i < len(escaped):
+        if escaped[i] == "\\":
+            i += 2
+            continue
+        assert escaped[i] != '"'
+        i += 1
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    st.integers(min_value=1, max_value=50),
+    st.floats(min_value=0.001, max_value=10.0, allow_nan=False),
+    st.floats(min_value=0.01, max_value=10.0, allow_nan=False),
+)
+def test_mm1k_invariants(K, lam, mu):
+    m = MM1KModel(K)
+    m.solve(lam, mu)
+    if not m.is_valid:
+        return
+    assert abs(sum(m.p) - 1.0) < 1e-6
+    assert 0 <= m.avg_num_in_system <= K
+    assert m.throughput <= lam + 1e-9
+    assert m.throughput <= mu + 1e-9  # single server can't beat service rate
+    assert m.avg_wait_time >= 0
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    st.integers(min_value=1, max_value=30),
+    st.lists(
+        st.floats(min_value=0.01, max_value=5.0, allow_nan=False),
+        min_size=1, max_size=8,
+    ),
+    st.floats(min_value=0.0, max_value=20.0, allow_nan=False),
+)
+def test_state_dependent_invariants(K, rates, lam):
+    m = MM1StateDependentModel(K, rates)
+    m.solve(lam)
+    if not m.is_valid:
+        return
+    assert abs(sum(m.p) - 1.0) < 1e-6
+    assert 0 <= m.avg_num_in_servers <= min(len(rates), K) + 1e-9
+    assert m.throughput <= lam + 1e-9
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.text(alphabet="0123456789hms.", max_size=12))
+def test_parse_duration_never_hangs(s):
+    try:
+        v = parse_go_duration(s)
+        assert isinstance(v, float)
+    except ValueError:
+        pass
