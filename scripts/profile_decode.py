@@ -98,3 +98,50 @@ def calibrate_70b():
     print(f"hbm total={total/2**30:.0f}GiB")
     with open("gpurun_out/calibration_70b.json", "w") as f:
         f.write(result.to_json())
+
+
+def calibrate_mixtral():
+    """Mixtral-8x7B (MoE) on ONE MI355X: ~94 GB bf16 weights resident
+    (BASELINE config #4's second model family)."""
+    import json
+    import time as _t
+
+    import torch
+
+    from wva_amd.calibration.itl_benchmark import fit_itl_curve
+    from wva_amd.calibration.moe_model import MIXTRAL_8X7B, MixtralDecodeModel
+
+    batches = [1, 8, 32, 64]
+    model = MixtralDecodeModel(MIXTRAL_8X7B, max_batch=max(batches), max_seq=1024)
+    itls = []
+    for b in batches:
+        model.reset(b, 512)
+        tokens = torch.randint(0, MIXTRAL_8X7B.vocab_size, (b,), device="cuda")
+        for _ in range(3):
+            model.decode_step(tokens)
+        torch.cuda.synchronize()
+        t0 = _t.perf_counter()
+        for _ in range(5):
+            model.decode_step(tokens)
+        torch.cuda.synchronize()
+        itls.append((_t.perf_counter() - t0) * 1000.0 / 5)
+    alpha, beta, r2 = fit_itl_curve(batches, itls)
+    rec = {
+        "model": MIXTRAL_8X7B.name,
+        "gpu_count": 1,
+        "alpha_ms": alpha,
+        "beta_ms": beta,
+        "r_squared": r2,
+        "batch_sizes": batches,
+        "itl_ms": itls,
+        "weights_gib": MIXTRAL_8X7B.weight_bytes() / 2**30,
+        "decode_tokens_per_s_peak": max(
+            b / (t / 1000.0) for b, t in zip(batches, itls)
+        ),
+    }
+    print("mixtral calibration:", json.dumps(rec, indent=2))
+    import os
+
+    os.makedirs("gpurun_out", exist_ok=True)
+    with open("gpurun_out/calibration_mixtral.json", "w") as f:
+        json.dump(rec, f, indent=2)
