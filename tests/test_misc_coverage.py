@@ -1,0 +1,164 @@
+"""Coverage for remaining corners: solver best-effort policies, local
+discovery label mapping, CRD manifest, backoff helper.
+"""
+import pytest
+import yaml
+
+from wva_amd.api.crd import crd_yaml, variantautoscaling_crd
+from wva_amd.discovery.local import LocalGPU, _product_label, node_labels_for_local_gpus
+from wva_amd.inferno import Solver, System
+from wva_amd.inferno.types import (
+    AcceleratorSpec,
+    ModelAcceleratorPerfData,
+    ModelTarget,
+    OptimizerSpec,
+    POLICY_PRIORITY_ROUND_ROBIN,
+    POLICY_ROUND_ROBIN,
+    ServerLoadSpec,
+    ServerSpec,
+    ServiceClassSpec,
+    ServiceParmsSpec,
+    SystemData,
+)
+from wva_amd.utils.backoff import retry_with_backoff
+
+
+def big_demand_system(capacity):
+    data = SystemData(
+        accelerators=[AcceleratorSpec(name="MI355X", type="MI355X", cost=50.0)],
+        models=[ModelAcceleratorPerfData(
+            name="m", acc="MI355X", acc_count=1, max_batch_size=256,
+            at_tokens=50, service_parms=ServiceParmsSpec(alpha=11.3, beta=0.015),
+        )],
+        service_classes=[ServiceClassSpec(
+            name="premium", priority=1,
+            model_targets=[ModelTarget(model="m", slo_itl=24.0, slo_ttft=500.0)],
+        )],
+        servers=[
+            ServerSpec(
+                name=f"srv-{i}", service_class="premium", model="m",
+                load=ServerLoadSpec(arrival_rate=200000.0, avg_in_tokens=100,
+                                    avg_out_tokens=50),
+            )
+            for i in range(3)
+        ],
+        capacity={"MI355X": capacity},
+    )
+    return System(data)
+
+
+class TestBestEffortPolicies:
+    def _want(self, sys_):
+        sys_.generate_all_allocations()
+        return min(
+            a.num_replicas
+            for s in sys_.servers.values()
+            for a in s.all_allocations.values()
+        )
+
+    def test_round_robin_shares_equally(self):
+        sys_ = big_demand_system(capacity=6)
+        want = self._want(sys_)
+        assert want > 2  # each server wants more than its share
+        solver = Solver(OptimizerSpec(saturation_policy=POLICY_ROUND_ROBIN))
+        solver.solve(sys_)
+        got = [s.allocation.num_replicas if s.allocation else 0
+               for s in sys_.servers.values()]
+        assert sum(got) <= 6
+        # round robin: equal shares
+        assert max(got) - min(got) <= 1
+        assert all(g >= 1 for g in got)
+
+    def test_priority_round_robin(self):
+        sys_ = big_demand_system(capacity=6)
+        self._want(sys_)
+        solver = Solver(OptimizerSpec(
+            saturation_policy=POLICY_PRIORITY_ROUND_ROBIN
+        ))
+        solver.solve(sys_)
+        total = sum(s.allocation.num_replicas if s.allocation else 0
+                    for s in sys_.servers.values())
+        assert 0 < total <= 6
+
+    def test_none_policy_leaves_unallocated(self):
+        sys_ = big_demand_system(capacity=1)
+        self._want(sys_)
+        solver = Solver(OptimizerSpec(saturation_policy="None"))
+        solver.solve(sys_)
+        # nobody's full requirement fits and best-effort is off
+        assert all(s.allocation is None for s in sys_.servers.values())
+
+
+class TestLocalDiscovery:
+    def test_product_label_mi355x(self):
+        gpu = LocalGPU(index=0, name="AMD Instinct MI355X", memory_mib=294912)
+        assert _product_label(gpu) == "AMD-Instinct-MI355X-288GB"
+
+    def test_product_label_from_gfx_arch(self):
+        gpu = LocalGPU(index=0, name="AMD Radeon Graphics",
+                       memory_mib=294912, gfx_arch="gfx950")
+        assert _product_label(gpu) == "AMD-Instinct-MI355X-288GB"
+
+    def test_node_labels(self):
+        gpus = [
+            LocalGPU(index=i, name="AMD Instinct MI355X", memory_mib=294912)
+            for i in range(8)
+        ]
+        labels = node_labels_for_local_gpus(gpus)
+        assert labels["amd.com/gpu.product"] == "AMD-Instinct-MI355X-288GB"
+        assert labels["amd.com/gpu.memory"] == "294912"
+        assert labels["amd.com/gpu.count"] == "8"
+
+    def test_no_gpus(self):
+        assert node_labels_for_local_gpus([]) == {}
+
+
+class TestCRDManifest:
+    def test_yaml_valid_and_complete(self):
+        doc = yaml.safe_load(crd_yaml())
+        assert doc["metadata"]["name"] == "variantautoscalings.llmd.ai"
+        v = doc["spec"]["versions"][0]
+        assert v["name"] == "v1alpha1"
+        assert v["subresources"] == {"status": {}}
+        spec_schema = v["schema"]["openAPIV3Schema"]["properties"]["spec"]
+        assert set(spec_schema["required"]) == {"modelID", "scaleTargetRef"}
+        assert spec_schema["properties"]["variantCost"]["default"] == "10.0"
+        cols = {c["name"] for c in v["additionalPrinterColumns"]}
+        assert cols == {"Target", "Model", "Optimized", "MetricsReady", "Age"}
+        assert doc["spec"]["names"]["shortNames"] == ["va"]
+
+    def test_condition_list_map(self):
+        doc = variantautoscaling_crd()
+        conds = (doc["spec"]["versions"][0]["schema"]["openAPIV3Schema"]
+                 ["properties"]["status"]["properties"]["conditions"])
+        assert conds["x-kubernetes-list-map-keys"] == ["type"]
+
+
+class TestBackoff:
+    def test_succeeds_after_retries(self):
+        calls = []
+
+        def fn():
+            calls.append(1)
+            if len(calls) < 3:
+                raise ValueError("x")
+            return 42
+
+        assert retry_with_backoff(fn, sleep=lambda s: None) == 42
+        assert len(calls) == 3
+
+    def test_raises_after_exhaustion(self):
+        def fn():
+            raise ValueError("always")
+
+        with pytest.raises(ValueError):
+            retry_with_backoff(fn, max_attempts=2, sleep=lambda s: None)
+
+    def test_should_retry_gate(self):
+        def fn():
+            raise KeyError("nope")
+
+        with pytest.raises(KeyError):
+            retry_with_backoff(
+                fn, should_retry=lambda e: False, sleep=lambda s: None
+            )
