@@ -1,6 +1,8 @@
 """Emulator unit tests: ReplicaSim queueing dynamics, ClusterSim
 deployment controller, EPP admission bound, event recording.
 """
+import pytest
+
 from wva_amd.api.types import ObjectMeta
 from wva_amd.api.types import (
     CrossVersionObjectReference,
@@ -170,3 +172,49 @@ class _NullSource:
 
     def get(self, query, params):
         return None
+
+
+class TestITLTableProfile:
+    """Measured per-batch ITL tables (MoE concave curves) — see
+    docs/calibration.md Mixtral section."""
+
+    MIXTRAL = ([1, 8, 32, 64], [23.86, 42.56, 47.15, 48.58])
+
+    def test_interpolation_exact_at_knots(self):
+        p = ServiceProfile.from_itl_table(*self.MIXTRAL)
+        for b, t in zip(*self.MIXTRAL):
+            assert p.itl_ms(b) == pytest.approx(t)
+
+    def test_interpolation_between_knots(self):
+        p = ServiceProfile.from_itl_table(*self.MIXTRAL)
+        mid = p.itl_ms(20)  # between 8 and 32
+        assert 42.56 < mid < 47.15
+
+    def test_extrapolation_uses_last_slope(self):
+        p = ServiceProfile.from_itl_table(*self.MIXTRAL)
+        slope = (48.58 - 47.15) / (64 - 32)
+        assert p.itl_ms(128) == pytest.approx(48.58 + slope * 64)
+
+    def test_below_first_knot_clamps(self):
+        p = ServiceProfile.from_itl_table(*self.MIXTRAL)
+        assert p.itl_ms(0) == pytest.approx(23.86)
+
+    def test_linear_parms_derived(self):
+        p = ServiceProfile.from_itl_table(*self.MIXTRAL)
+        assert p.alpha_ms > 0
+        assert p.beta_ms > 0
+        # linear model must at least be within the table's range at B=32
+        assert 20 < p.alpha_ms + p.beta_ms * 32 < 60
+
+    def test_without_table_linear(self):
+        p = ServiceProfile(alpha_ms=10.0, beta_ms=0.5)
+        assert p.itl_ms(20) == pytest.approx(20.0)
+
+    def test_rejects_short_table(self):
+        with pytest.raises(ValueError):
+            ServiceProfile.from_itl_table([1], [10.0])
+
+    def test_unsorted_input_sorted(self):
+        p = ServiceProfile.from_itl_table([32, 1, 8], [47.15, 23.86, 42.56])
+        assert p.itl_ms(1) == pytest.approx(23.86)
+        assert p.itl_ms(32) == pytest.approx(47.15)

@@ -15,7 +15,7 @@ from __future__ import annotations
 
 import random
 from dataclasses import dataclass
-from typing import Deque, Dict, List
+from typing import Deque, Dict, List, Optional, Sequence, Tuple
 from collections import deque
 
 
@@ -35,13 +35,68 @@ class ServiceProfile:
     block_size: int = 16
     prefill_tokens_per_s: float = 50_000.0
     prefix_cache_hit_rate: float = 0.0
+    # optional measured per-batch ITL table: [(batch, itl_ms), ...] sorted
+    # by batch. When set, itl_ms() interpolates the table instead of the
+    # linear α+β·B model — required for MoE decode, whose ITL is concave
+    # (all experts active by B≈8; see docs/calibration.md, Mixtral section).
+    itl_table: Optional[Tuple[Tuple[int, float], ...]] = None
 
     @property
     def kv_capacity_tokens(self) -> int:
         return self.num_gpu_blocks * self.block_size
 
     def itl_ms(self, batch: int) -> float:
-        return self.alpha_ms + self.beta_ms * max(batch, 1)
+        batch = max(batch, 1)
+        if self.itl_table:
+            return self._interp_itl(batch)
+        return self.alpha_ms + self.beta_ms * batch
+
+    def _interp_itl(self, batch: int) -> float:
+        table = self.itl_table
+        if batch <= table[0][0]:
+            return table[0][1]
+        for (b0, t0), (b1, t1) in zip(table, table[1:]):
+            if batch <= b1:
+                frac = (batch - b0) / (b1 - b0)
+                return t0 + frac * (t1 - t0)
+        # beyond the measured range: extend with the last segment's slope
+        (b0, t0), (b1, t1) = table[-2], table[-1]
+        slope = (t1 - t0) / (b1 - b0)
+        return t1 + slope * (batch - b1)
+
+    @classmethod
+    def from_itl_table(
+        cls,
+        batches: Sequence[int],
+        itl_ms: Sequence[float],
+        **kwargs,
+    ) -> "ServiceProfile":
+        """Profile from a measured per-batch ITL curve.
+
+        `batches`/`itl_ms` come straight from a calibration JSON
+        (profiles/calibration_*.json: keys `batch_sizes` / `itl_ms`).
+        α/β are also derived by least squares over the table so consumers
+        that only understand the linear model (the V2 analyzer's
+        ITLAtBatch and Inferno's ServiceParmsSpec) get the best linear
+        approximation, while the simulator itself interpolates exactly.
+        """
+        if len(batches) != len(itl_ms) or len(batches) < 2:
+            raise ValueError("need >=2 (batch, itl) pairs of equal length")
+        pairs = tuple(sorted(zip((int(b) for b in batches), itl_ms)))
+        n = len(pairs)
+        sx = sum(b for b, _ in pairs)
+        sy = sum(t for _, t in pairs)
+        sxx = sum(b * b for b, _ in pairs)
+        sxy = sum(b * t for b, t in pairs)
+        denom = n * sxx - sx * sx
+        beta = (n * sxy - sx * sy) / denom if denom else 0.0
+        alpha = (sy - beta * sx) / n
+        return cls(
+            alpha_ms=max(alpha, 0.0),
+            beta_ms=max(beta, 0.0),
+            itl_table=pairs,
+            **kwargs,
+        )
 
 
 @dataclass
