@@ -145,3 +145,32 @@ def calibrate_mixtral():
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/calibration_mixtral.json", "w") as f:
         json.dump(rec, f, indent=2)
+
+
+def graph_ab():
+    """Eager vs hipGraph-captured decode A/B on Llama-3-8B: the launch
+    overhead shows up in alpha of ITL = alpha + beta*batch."""
+    import json
+
+    from wva_amd.calibration.itl_benchmark import fit_itl_curve, measure_itl
+    from wva_amd.calibration.model import LLAMA_3_8B, LlamaDecodeModel
+
+    batches = [1, 8, 32, 64]
+    model = LlamaDecodeModel(LLAMA_3_8B, max_batch=max(batches), max_seq=1024)
+    rec = {}
+    for mode, use_graph in [("eager", False), ("graph", True)]:
+        itls = [
+            measure_itl(model, b, 512, iters=8, use_graph=use_graph)
+            for b in batches
+        ]
+        alpha, beta, r2 = fit_itl_curve(batches, itls)
+        rec[mode] = {
+            "itl_ms": itls, "alpha_ms": alpha, "beta_ms": beta, "r2": r2,
+            "peak_tps": max(b / (t / 1000.0) for b, t in zip(batches, itls)),
+        }
+        print(f"{mode}: itl={['%.3f' % t for t in itls]} "
+              f"alpha={alpha:.3f} beta={beta:.4f} peak={rec[mode]['peak_tps']:.0f}")
+    import os
+    os.makedirs("gpurun_out", exist_ok=True)
+    with open("gpurun_out/graph_ab.json", "w") as f:
+        json.dump(rec, f, indent=2)

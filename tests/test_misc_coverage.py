@@ -162,3 +162,14 @@ class TestBackoff:
             retry_with_backoff(
                 fn, should_retry=lambda e: False, sleep=lambda s: None
             )
+
+
+class TestGraphedDecoderCPU:
+    def test_requires_gpu(self):
+        import torch
+        if torch.cuda.is_available():
+            pytest.skip("GPU present")
+        from wva_amd.calibration.graph import GraphedDecoder
+
+        with pytest.raises(RuntimeError, match="requires a GPU"):
+            GraphedDecoder(engine=None, batch=1)

@@ -66,19 +66,34 @@ def measure_itl(
     context_len: int = 512,
     iters: int = 8,
     warmup: int = 3,
+    use_graph: bool = False,
 ) -> float:
-    """Median decode-iteration wall time (ms) at the given batch size."""
+    """Median decode-iteration wall time (ms) at the given batch size.
+
+    With `use_graph` the step is replayed as one captured hipGraph
+    (calibration/graph.py) — measuring the launch-overhead-free serving
+    configuration, which is what a production engine would run.
+    """
     model.reset(batch, context_len)
     tokens = torch.randint(
         0, model.cfg.vocab_size, (batch,), device=model.device
     )
-    for _ in range(warmup):
-        model.decode_step(tokens)
+    if use_graph:
+        from .graph import GraphedDecoder
+
+        dec = GraphedDecoder(model, batch, warmup_steps=max(warmup, 1))
+        dec.reset_to(batch, context_len)
+        step = dec.decode_step
+        step(tokens)  # one replay to settle
+    else:
+        step = model.decode_step
+        for _ in range(warmup):
+            step(tokens)
     torch.cuda.synchronize()
     times = []
     for _ in range(iters):
         t0 = time.perf_counter()
-        model.decode_step(tokens)
+        step(tokens)
         torch.cuda.synchronize()
         times.append((time.perf_counter() - t0) * 1000.0)
     times.sort()
@@ -109,6 +124,7 @@ def calibrate_service_profile(
     max_seq: int = 1024,
     gpu_count: int = 1,
     iters: int = 8,
+    use_graph: bool = False,
 ) -> Tuple[ServiceProfile, CalibrationResult]:
     """Measure on the current GPU and return an emulator ServiceProfile
     + the raw calibration record."""
@@ -118,7 +134,9 @@ def calibrate_service_profile(
     model = LlamaDecodeModel(cfg, max_batch=max_batch, max_seq=max_seq)
     itl: List[float] = []
     for b in batch_sizes:
-        itl.append(measure_itl(model, b, context_len, iters=iters))
+        itl.append(
+            measure_itl(model, b, context_len, iters=iters, use_graph=use_graph)
+        )
     alpha, beta, r2 = fit_itl_curve(batch_sizes, itl)
     blocks, kv_tokens = derive_kv_capacity(cfg)
     peak_tps = max(
