@@ -174,3 +174,41 @@ def graph_ab():
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/graph_ab.json", "w") as f:
         json.dump(rec, f, indent=2)
+
+
+def tunableop_ab():
+    """TunableOp A/B: autotune hipBLASLt GEMM solutions for the decode
+    shapes, persist the tuning table, measure eager ITL before/after."""
+    import json
+    import os
+
+    import torch
+
+    from wva_amd.calibration.itl_benchmark import fit_itl_curve, measure_itl
+    from wva_amd.calibration.model import LLAMA_3_8B, LlamaDecodeModel
+
+    os.makedirs("gpurun_out", exist_ok=True)
+    batches = [1, 8, 32, 64]
+    model = LlamaDecodeModel(LLAMA_3_8B, max_batch=max(batches), max_seq=1024)
+
+    base = [measure_itl(model, b, 512, iters=8) for b in batches]
+    a0, b0, _ = fit_itl_curve(batches, base)
+
+    tun = torch.cuda.tunable
+    tun.enable(True)
+    tun.tuning_enable(True)
+    tun.set_filename("gpurun_out/tunableop_8b.csv")
+    for b in batches:  # tuning happens on first call per GEMM shape
+        measure_itl(model, b, 512, iters=2)
+    tun.write_file() if hasattr(tun, "write_file") else None
+    tun.tuning_enable(False)
+
+    tuned = [measure_itl(model, b, 512, iters=8) for b in batches]
+    a1, b1, _ = fit_itl_curve(batches, tuned)
+    rec = {
+        "baseline": {"itl_ms": base, "alpha_ms": a0, "beta_ms": b0},
+        "tunableop": {"itl_ms": tuned, "alpha_ms": a1, "beta_ms": b1},
+    }
+    print("tunableop A/B:", json.dumps(rec, indent=2))
+    with open("gpurun_out/tunableop_ab.json", "w") as f:
+        json.dump(rec, f, indent=2)
