@@ -173,3 +173,53 @@ class TestGraphedDecoderCPU:
 
         with pytest.raises(RuntimeError, match="requires a GPU"):
             GraphedDecoder(engine=None, batch=1)
+
+
+class TestProbeServer:
+    def test_probes_and_metrics(self):
+        import urllib.request
+
+        from prometheus_client import CollectorRegistry
+        from wva_amd.metrics.metrics import MetricsEmitter
+        from wva_amd.runtime.http import ProbeServer
+
+        registry = CollectorRegistry()
+        emitter = MetricsEmitter(registry=registry)
+        emitter.emit("va1", "ns1", "MI355X", current=2, desired=3)
+        ready = {"ok": False}
+        srv = ProbeServer(
+            "127.0.0.1:0",
+            healthz=lambda: True,
+            readyz=lambda: ready["ok"],
+            registry=registry,
+        )
+        srv.start()
+        try:
+            base = f"http://127.0.0.1:{srv.port}"
+
+            def get(path):
+                try:
+                    r = urllib.request.urlopen(base + path, timeout=5)
+                    return r.status, r.read()
+                except urllib.error.HTTPError as e:
+                    return e.code, e.read()
+
+            assert get("/healthz")[0] == 200
+            assert get("/readyz")[0] == 503  # bootstrap not complete
+            ready["ok"] = True
+            assert get("/readyz")[0] == 200
+            code, body = get("/metrics")
+            assert code == 200
+            text = body.decode()
+            assert 'wva_desired_replicas{' in text
+            assert 'accelerator_type="MI355X"' in text
+            assert get("/nope")[0] == 404
+        finally:
+            srv.stop()
+
+    def test_parse_bind(self):
+        from wva_amd.runtime.http import _parse_bind
+
+        assert _parse_bind(":8081") == ("0.0.0.0", 8081)
+        assert _parse_bind("127.0.0.1:9090") == ("127.0.0.1", 9090)
+        assert _parse_bind("8443") == ("0.0.0.0", 8443)

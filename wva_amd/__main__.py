@@ -64,7 +64,7 @@ def main(argv=None) -> int:
         source = SimMetricsSource(sim)
         log.info("running in emulated mode (in-memory cluster)")
 
-    app = build_app(cluster, config, source=source)
+    app = build_app(cluster, config, source=source, serve_http=True)
     app.configmap_reconciler.bootstrap_initial_configmaps()
     app.start()
     log.info("manager started (leader_elect=%s)", config.infra.enable_leader_election)

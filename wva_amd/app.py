@@ -75,11 +75,21 @@ class App:
     emitter: MetricsEmitter
     capacity_store: CapacityKnowledgeStore
     va_index: VAIndex
+    probe_server: Optional["ProbeServer"] = None
+    metrics_server: Optional["ProbeServer"] = None
 
     def start(self) -> None:
         self.manager.start()
+        if self.probe_server is not None:
+            self.probe_server.start()
+        if self.metrics_server is not None:
+            self.metrics_server.start()
 
     def stop(self) -> None:
+        if self.probe_server is not None:
+            self.probe_server.stop()
+        if self.metrics_server is not None:
+            self.metrics_server.stop()
         self.manager.stop()
 
 
@@ -92,6 +102,7 @@ def build_app(
     sfz_interval_seconds: float = 0.1,
     scrape_fetch=None,
     start_engines: bool = True,
+    serve_http: bool = False,
 ) -> App:
     # Metrics source: explicit (sim) or Prometheus from config
     source_registry = SourceRegistry()
@@ -203,6 +214,30 @@ def build_app(
     # ConfigMap bootstrap before runnables (cmd/main.go:322-336)
     configmap_reconciler.bootstrap_initial_configmaps()
 
+    # HTTP surfaces (cmd/main.go:266-287,482-498): probe address serves
+    # /healthz + /readyz (readyz gated on ConfigMap bootstrap); metrics
+    # address serves /metrics from the emitter's registry.
+    probe_server = metrics_server = None
+    if serve_http:
+        from .runtime.http import ProbeServer
+
+        probe_addr = config.infra.health_probe_bind_address
+        metrics_addr = config.infra.metrics_bind_address
+        if probe_addr and probe_addr != "0":
+            probe_server = ProbeServer(
+                probe_addr,
+                healthz=manager.healthz,
+                readyz=manager.readyz,
+                serve_metrics=False,
+            )
+        if metrics_addr and metrics_addr != "0":
+            metrics_server = ProbeServer(
+                metrics_addr,
+                healthz=manager.healthz,
+                readyz=manager.readyz,
+                registry=emitter.registry,
+            )
+
     return App(
         cluster=cluster,
         config=config,
@@ -220,6 +255,8 @@ def build_app(
         emitter=emitter,
         capacity_store=capacity_store,
         va_index=VAIndex(cluster),
+        probe_server=probe_server,
+        metrics_server=metrics_server,
     )
 
 

@@ -1,0 +1,99 @@
+"""Probe + metrics HTTP endpoints.
+
+The reference binds two listeners from cmd/main.go: a metrics address
+(controller-runtime /metrics, reference cmd/main.go:266-287) and a health
+probe address serving /healthz + /readyz, with readyz gated on ConfigMap
+bootstrap (cmd/main.go:482-498). This module is the Python equivalent:
+one stdlib ThreadingHTTPServer per address, run as daemon threads.
+
+/metrics renders the prometheus_client registry the MetricsEmitter
+registered its wva_* series into — the HPA/prometheus-adapter contract
+surface.
+"""
+from __future__ import annotations
+
+import threading
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from typing import Callable, Optional, Tuple
+
+from prometheus_client import CollectorRegistry, REGISTRY, generate_latest
+from prometheus_client.exposition import CONTENT_TYPE_LATEST
+
+
+def _parse_bind(addr: str) -> Tuple[str, int]:
+    """':8081' / '0.0.0.0:8081' / '8081' → (host, port)."""
+    addr = addr.strip()
+    if ":" in addr:
+        host, _, port = addr.rpartition(":")
+        return host or "0.0.0.0", int(port)
+    return "0.0.0.0", int(addr)
+
+
+class ProbeServer:
+    """Serves /healthz, /readyz and (optionally) /metrics on one address."""
+
+    def __init__(
+        self,
+        bind_address: str,
+        healthz: Callable[[], bool],
+        readyz: Callable[[], bool],
+        registry: Optional[CollectorRegistry] = None,
+        serve_metrics: bool = True,
+    ):
+        self._healthz = healthz
+        self._readyz = readyz
+        self._registry = registry if registry is not None else REGISTRY
+        self._serve_metrics = serve_metrics
+        host, port = _parse_bind(bind_address)
+        outer = self
+
+        class Handler(BaseHTTPRequestHandler):
+            def do_GET(self):  # noqa: N802 — stdlib handler API
+                if self.path.startswith("/healthz"):
+                    self._probe(outer._healthz)
+                elif self.path.startswith("/readyz"):
+                    self._probe(outer._readyz)
+                elif self.path.startswith("/metrics") and outer._serve_metrics:
+                    body = generate_latest(outer._registry)
+                    self.send_response(200)
+                    self.send_header("Content-Type", CONTENT_TYPE_LATEST)
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                else:
+                    self.send_response(404)
+                    self.end_headers()
+
+            def _probe(self, check: Callable[[], bool]) -> None:
+                try:
+                    ok = check()
+                except Exception:  # noqa: BLE001 — probe must answer
+                    ok = False
+                body = b"ok" if ok else b"unhealthy"
+                self.send_response(200 if ok else 503)
+                self.send_header("Content-Type", "text/plain")
+                self.send_header("Content-Length", str(len(body)))
+                self.end_headers()
+                self.wfile.write(body)
+
+            def log_message(self, *args):  # silence per-request lines
+                pass
+
+        self._server = ThreadingHTTPServer((host, port), Handler)
+        self._thread: Optional[threading.Thread] = None
+
+    @property
+    def port(self) -> int:
+        return self._server.server_address[1]
+
+    def start(self) -> None:
+        self._thread = threading.Thread(
+            target=self._server.serve_forever, daemon=True, name="probe-http"
+        )
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._server.shutdown()
+        self._server.server_close()
+        if self._thread:
+            self._thread.join(timeout=5)
