@@ -185,7 +185,9 @@ class TestProbeServer:
 
         registry = CollectorRegistry()
         emitter = MetricsEmitter(registry=registry)
-        emitter.emit("va1", "ns1", "MI355X", current=2, desired=3)
+        emitter.emit_replica_metrics(
+            "va1", "ns1", current=2, desired=3, accelerator_type="MI355X"
+        )
         ready = {"ok": False}
         srv = ProbeServer(
             "127.0.0.1:0",
