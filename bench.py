@@ -252,6 +252,9 @@ def main() -> None:
     if has_gpu:
         from wva_amd.calibration.itl_benchmark import calibrate_service_profile
         from wva_amd.calibration.model import LLAMA_3_8B
+        from wva_amd.ops import enable_tuned_gemms
+
+        enable_tuned_gemms()  # committed MI355X GEMM solution table
 
         # hipGraph-captured stepping (calibration/graph.py): the serving
         # configuration a production engine runs — measured α drops ~4%

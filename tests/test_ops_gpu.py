@@ -258,3 +258,13 @@ class TestGraphedDecoder:
         model.reset(2, 14)
         with pytest.raises(ValueError):
             GraphedDecoder(model, batch=2, warmup_steps=3)
+
+
+class TestTunedGemms:
+    def test_tuned_table_loads(self, dev):
+        from wva_amd.ops import enable_tuned_gemms
+
+        assert enable_tuned_gemms() is True
+        import torch as _t
+        assert _t.cuda.tunable.is_enabled()
+        assert not _t.cuda.tunable.tuning_is_enabled()

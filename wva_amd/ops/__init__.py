@@ -228,3 +228,28 @@ def gqa_decode_attn(
     return gqa_decode_attn_ref(q, k_cache, v_cache, context_lens, scale).to(
         q.dtype
     )
+
+
+def enable_tuned_gemms() -> bool:
+    """Load the committed TunableOp GEMM solution table for gfx950.
+
+    `wva_amd/ops/tunableop_gfx950.csv` holds hipBLASLt/rocBLAS solution
+    choices autotuned on MI355X for the decode GEMM shapes (qkv/o/
+    gate_up/down/lm_head at batches 1-64) — measured α drops ~10% vs the
+    default heuristics (profiles/tunableop_ab.json). Solutions are keyed
+    to PyTorch/hipBLASLt versions via the CSV's Validator rows; on a
+    mismatched stack TunableOp discards them and falls back to defaults.
+    Returns True if the table was loaded.
+    """
+    import os as _os
+
+    import torch as _torch
+
+    path = _os.path.join(_os.path.dirname(__file__), "tunableop_gfx950.csv")
+    if not _torch.cuda.is_available() or not _os.path.exists(path):
+        return False
+    tun = _torch.cuda.tunable
+    tun.enable(True)
+    tun.tuning_enable(False)  # read-only: never autotune in production
+    tun.read_file(path)
+    return True
