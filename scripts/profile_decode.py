@@ -66,19 +66,25 @@ def attn_bench():
         k = torch.randn(batch, hk, S, 128, device="cuda", dtype=torch.bfloat16)
         v = torch.randn(batch, hk, S, 128, device="cuda", dtype=torch.bfloat16)
         lens = torch.full((batch,), ctx, device="cuda", dtype=torch.int32)
-        for _ in range(3):
-            ops.gqa_decode_attn(q, k, v, lens)
-        torch.cuda.synchronize()
+        from wva_amd.ops import _require_ext
+        ext = _require_ext()
+        scale = 128 ** -0.5
         import time as _t
-        n = 20
-        t0 = _t.perf_counter()
-        for _ in range(n):
-            ops.gqa_decode_attn(q, k, v, lens)
-        torch.cuda.synchronize()
-        us = (_t.perf_counter() - t0) / n * 1e6
-        kv_bytes = 2 * batch * ctx * hk * 128 * 2
-        print(f"attn b={batch} hq={hq} hk={hk} ctx={ctx}: {us:.1f} us, "
-              f"KV {kv_bytes/1e6:.1f} MB, {kv_bytes/us/1e3:.2f} TB/s")
+        for name, fn in [("v3", ext.gqa_decode_attn),
+                         ("v4", ext.gqa_decode_attn_v4)]:
+            for _ in range(3):
+                fn(q, k, v, lens, scale)
+            torch.cuda.synchronize()
+            n = 20
+            t0 = _t.perf_counter()
+            for _ in range(n):
+                fn(q, k, v, lens, scale)
+            torch.cuda.synchronize()
+            us = (_t.perf_counter() - t0) / n * 1e6
+            kv_bytes = 2 * batch * ctx * hk * 128 * 2
+            print(f"attn[{name}] b={batch} hq={hq} hk={hk} ctx={ctx}: "
+                  f"{us:.1f} us, KV {kv_bytes/1e6:.1f} MB, "
+                  f"{kv_bytes/us/1e3:.2f} TB/s")
 
 
 def calibrate_70b():

@@ -268,3 +268,43 @@ class TestTunedGemms:
         import torch as _t
         assert _t.cuda.tunable.is_enabled()
         assert not _t.cuda.tunable.tuning_is_enabled()
+
+
+class TestAttnV4:
+    @pytest.mark.parametrize("batch,hq,hk,ctx", [
+        (2, 8, 2, 33), (4, 32, 8, 512), (2, 64, 8, 300), (3, 4, 4, 100),
+        (2, 8, 1, 2000),  # split path (small grid)
+    ])
+    def test_v4_matches_reference(self, dev, batch, hq, hk, ctx):
+        from wva_amd.ops import _require_ext, gqa_decode_attn_ref
+
+        ext = _require_ext()
+        torch.manual_seed(11)
+        S = max(ctx + 8, 64)
+        q = torch.randn(batch, hq, 128, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(batch, hk, S, 128, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(batch, hk, S, 128, device=dev, dtype=torch.bfloat16)
+        lens = torch.randint(1, ctx + 1, (batch,), device=dev,
+                             dtype=torch.int32)
+        lens[0] = ctx
+        scale = 128 ** -0.5
+        out = ext.gqa_decode_attn_v4(q, k, v, lens, scale)
+        ref = gqa_decode_attn_ref(
+            q.float().cpu(), k.float().cpu(), v.float().cpu(),
+            lens.cpu(), scale,
+        )
+        torch.testing.assert_close(out.float().cpu(), ref, atol=2e-2,
+                                   rtol=2e-2)
+
+    def test_v4_matches_v3(self, dev):
+        from wva_amd.ops import _require_ext
+
+        ext = _require_ext()
+        torch.manual_seed(5)
+        q = torch.randn(8, 32, 128, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(8, 8, 640, 128, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(8, 8, 640, 128, device=dev, dtype=torch.bfloat16)
+        lens = torch.full((8,), 640, device=dev, dtype=torch.int32)
+        a = ext.gqa_decode_attn(q, k, v, lens, 128 ** -0.5)
+        b = ext.gqa_decode_attn_v4(q, k, v, lens, 128 ** -0.5)
+        torch.testing.assert_close(a.float(), b.float(), atol=5e-3, rtol=5e-3)

@@ -215,6 +215,195 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v4: MFMA scores. Differences from v3:
+//   * K is NOT staged through LDS at all: each wave loads its 16 tile
+//     positions straight from HBM into mfma_f32_16x16x32_bf16 A
+//     fragments (dwordx4 per lane per k-subtile) and computes the
+//     64-position × 16-head score tile in 4 MFMAs on the matrix cores.
+//     Scores land in LDS (s_smem) for the cross-wave softmax; only V
+//     still goes through LDS (its position-major PV sweep needs the
+//     transpose LDS provides). LDS drops ~17 KiB → higher residency.
+//   * A/B fragments use the same lane→k bijection for K and Q, which is
+//     sufficient for correctness (the k-sum is permutation-invariant; the
+//     C/D mapping col=lane&15, row=(lane>>4)·4+reg was probe-verified on
+//     MI355X — scripts/mfma_probe.hip).
+//   * Softmax state, p staging and the vector PV sweep are v3 verbatim.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8_frag;
+typedef __attribute__((ext_vector_type(4))) float f32x4_frag;
+
+#define S_ROW 65  // s_smem row stride (floats): 64 + 1 pad
+
+__global__ __launch_bounds__(256, 2) void gqa_decode_attn_v4_kernel(
+    bf16* __restrict__ out,
+    float* __restrict__ workspace,
+    const bf16* __restrict__ q,
+    const bf16* __restrict__ k_cache,
+    const bf16* __restrict__ v_cache,
+    const int* __restrict__ context_lens,
+    const int num_q_heads,
+    const int num_kv_heads,
+    const int max_seq,
+    const float scale) {
+  const int b = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int split = blockIdx.z;
+  const int num_splits = gridDim.z;
+  const int G = num_q_heads / num_kv_heads;
+  const int ctx = context_lens[b];
+
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+
+  __shared__ float s_smem[MAX_G][S_ROW];      // score tile [head][pos]
+  __shared__ unsigned int v_smem[TILE * ROW_DW];
+  __shared__ float p_smem[MAX_G][TILE];
+
+  // --- Q fragments (persistent): B operand, col = head = lane%16,
+  // k = 32·kk + 8·(lane/16) + i. Heads ≥ G are zero. ---
+  bf16x8_frag q_frag[4];
+  {
+    const int g = lane % 16;
+    const int k0 = 8 * (lane / 16);
+    if (g < G) {
+      const bf16* qrow = q + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM;
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        const short* src = reinterpret_cast<const short*>(qrow + 32 * kk + k0);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) q_frag[kk][i] = src[i];
+      }
+    } else {
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+        for (int i = 0; i < 8; ++i) q_frag[kk][i] = 0;
+    }
+  }
+
+  const int heads_mine = (G > wave) ? (G - wave + NUM_WAVES - 1) / NUM_WAVES : 0;
+  float m[2], s[2], acc[2][2];
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    m[j] = -INFINITY;
+    s[j] = 0.0f;
+    acc[j][0] = acc[j][1] = 0.0f;
+  }
+
+  const long kv_row_dw = HEAD_DIM / 2;
+  const bf16* k_slab =
+      k_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM;
+  const unsigned int* v_base = reinterpret_cast<const unsigned int*>(
+      v_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM);
+
+  for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
+    const int tn = min(TILE, ctx - t0);
+
+    // --- stage V only (v3 staging loop, half the traffic) ---
+    {
+      typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
+      const int tile_u2 = tn * (HEAD_DIM / 4);
+      const unsigned int* v_src = &v_base[(long)t0 * kv_row_dw];
+      for (int idx = wave * WAVE_SIZE + lane; idx < tile_u2;
+           idx += NUM_WAVES * WAVE_SIZE) {
+        const int row = idx >> 5;
+        const int d2 = (idx & 31) * 2;
+        const uint2_t val = *reinterpret_cast<const uint2_t*>(&v_src[idx * 2]);
+        *reinterpret_cast<uint2_t*>(&v_smem[row * ROW_DW + d2]) = val;
+      }
+    }
+
+    // --- MFMA scores: wave w owns tile positions 16w..16w+15 ---
+    {
+      const int prow = 16 * wave + (lane % 16);       // A row = position
+      const int pos = t0 + prow;
+      const int pos_c = pos < max_seq ? pos : max_seq - 1;  // clamped load
+      const short* krow =
+          reinterpret_cast<const short*>(k_slab + (long)pos_c * HEAD_DIM);
+      const int k0 = 8 * (lane / 16);
+      f32x4_frag sf = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8_frag a_frag;
+        const short* src = krow + 32 * kk + k0;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) a_frag[i] = src[i];
+        sf = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, q_frag[kk], sf,
+                                                     0, 0, 0);
+      }
+      // C/D: col = lane&15 (head), row = (lane>>4)·4 + i (position)
+      const int g = lane & 15;
+      if (g < G) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int t = 16 * wave + (lane >> 4) * 4 + i;
+          s_smem[g][t] = (t < tn) ? sf[i] * scale : -INFINITY;
+        }
+      }
+    }
+    __syncthreads();  // V + scores visible
+
+    // --- softmax per owned head (v3, scores from s_smem) ---
+    const bool live = lane < tn;
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      const float score = live ? s_smem[g][lane] : -INFINITY;
+      const float tile_max = wave_reduce_max(score);
+      const float m_new = fmaxf(m[j], tile_max);
+      const float p = live ? __expf(score - m_new) : 0.0f;
+      const float tile_sum = wave_reduce_sum(p);
+      const float corr = (m[j] == -INFINITY) ? 0.0f : __expf(m[j] - m_new);
+      s[j] = s[j] * corr + tile_sum;
+      acc[j][0] *= corr;
+      acc[j][1] *= corr;
+      m[j] = m_new;
+      p_smem[g][lane] = p;
+    }
+
+    // --- PV (v3 verbatim) ---
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      const float* pg = p_smem[g];
+      float a0 = acc[j][0], a1 = acc[j][1];
+      for (int t = 0; t < tn; ++t) {
+        const unsigned int vw = v_smem[t * ROW_DW + lane];
+        const bf16x2 vv = *reinterpret_cast<const bf16x2*>(&vw);
+        const float p = pg[t];
+        a0 = fmaf(p, bf2f(vv.x), a0);
+        a1 = fmaf(p, bf2f(vv.y), a1);
+      }
+      acc[j][0] = a0;
+      acc[j][1] = a1;
+    }
+    __syncthreads();
+  }
+
+  if (num_splits == 1) {
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      const float inv = s[j] > 0.0f ? 1.0f / s[j] : 0.0f;
+      bf16x2* orow = reinterpret_cast<bf16x2*>(
+          out + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM);
+      orow[lane] = bf16x2{f2bf(acc[j][0] * inv), f2bf(acc[j][1] * inv)};
+    }
+  } else {
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      float* wsp = workspace +
+          ((((long)b * num_kv_heads + kvh) * G + g) * num_splits + split) *
+              (2 + HEAD_DIM);
+      if (lane == 0) {
+        wsp[0] = m[j];
+        wsp[1] = s[j];
+      }
+      wsp[2 + 2 * lane] = acc[j][0];
+      wsp[2 + 2 * lane + 1] = acc[j][1];
+    }
+  }
+}
+
 // Merge split-KV partials: one 64-thread wave per (b, q_head).
 __global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
     bf16* __restrict__ out,          // [B, Hq, 128]
@@ -274,6 +463,25 @@ extern "C" void launch_gqa_decode_attn(
   dim3 grid(batch, num_kv_heads, num_splits);
   dim3 block(256);
   hipLaunchKernelGGL(gqa_decode_attn_kernel, grid, block, 0, stream,
+                     (bf16*)out, (float*)workspace, (const bf16*)q,
+                     (const bf16*)k_cache, (const bf16*)v_cache, context_lens,
+                     num_q_heads, num_kv_heads, max_seq, scale);
+  if (num_splits > 1) {
+    dim3 mgrid(batch, num_q_heads);
+    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64), 0,
+                       stream, (bf16*)out, (const float*)workspace,
+                       num_q_heads, num_kv_heads, num_splits);
+  }
+}
+
+extern "C" void launch_gqa_decode_attn_v4(
+    void* out, void* workspace, const void* q, const void* k_cache,
+    const void* v_cache, const int* context_lens, int batch, int num_q_heads,
+    int num_kv_heads, int max_seq, int num_splits, float scale,
+    hipStream_t stream) {
+  dim3 grid(batch, num_kv_heads, num_splits);
+  dim3 block(256);
+  hipLaunchKernelGGL(gqa_decode_attn_v4_kernel, grid, block, 0, stream,
                      (bf16*)out, (float*)workspace, (const bf16*)q,
                      (const bf16*)k_cache, (const bf16*)v_cache, context_lens,
                      num_q_heads, num_kv_heads, max_seq, scale);

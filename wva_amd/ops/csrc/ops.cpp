@@ -34,6 +34,13 @@ extern "C" void launch_gqa_decode_attn(void* out, void* workspace,
                                        int num_q_heads, int num_kv_heads,
                                        int max_seq, int num_splits,
                                        float scale, hipStream_t stream);
+extern "C" void launch_gqa_decode_attn_v4(void* out, void* workspace,
+                                          const void* q, const void* k_cache,
+                                          const void* v_cache,
+                                          const int* context_lens, int batch,
+                                          int num_q_heads, int num_kv_heads,
+                                          int max_seq, int num_splits,
+                                          float scale, hipStream_t stream);
 
 namespace {
 
@@ -140,9 +147,10 @@ torch::Tensor silu_mul_fused(torch::Tensor gate_up) {
   return out;
 }
 
-torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
-                              torch::Tensor v_cache,
-                              torch::Tensor context_lens, double scale) {
+torch::Tensor gqa_decode_attn_impl(torch::Tensor q, torch::Tensor k_cache,
+                                   torch::Tensor v_cache,
+                                   torch::Tensor context_lens, double scale,
+                                   bool use_v4) {
   check_bf16_contig(q, "q");
   check_bf16_contig(k_cache, "k_cache");
   check_bf16_contig(v_cache, "v_cache");
@@ -173,12 +181,29 @@ torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
         q.options().dtype(torch::kFloat32));
     ws_ptr = workspace.data_ptr();
   }
-  launch_gqa_decode_attn(out.data_ptr(), ws_ptr, q.data_ptr(),
-                         k_cache.data_ptr(), v_cache.data_ptr(),
-                         context_lens.data_ptr<int>(), batch, num_q_heads,
-                         num_kv_heads, max_seq, num_splits, (float)scale,
-                         current_stream());
+  auto launch = use_v4 ? launch_gqa_decode_attn_v4 : launch_gqa_decode_attn;
+  launch(out.data_ptr(), ws_ptr, q.data_ptr(),
+         k_cache.data_ptr(), v_cache.data_ptr(),
+         context_lens.data_ptr<int>(), batch, num_q_heads,
+         num_kv_heads, max_seq, num_splits, (float)scale,
+         current_stream());
   return out;
+}
+
+torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
+                              torch::Tensor v_cache,
+                              torch::Tensor context_lens, double scale) {
+  return gqa_decode_attn_impl(q, k_cache, v_cache, context_lens, scale,
+                              /*use_v4=*/false);
+}
+
+// MFMA-scores variant (attention.hip v4) exposed separately for A/B
+// benchmarking; flips to the default once measured faster.
+torch::Tensor gqa_decode_attn_v4(torch::Tensor q, torch::Tensor k_cache,
+                                 torch::Tensor v_cache,
+                                 torch::Tensor context_lens, double scale) {
+  return gqa_decode_attn_impl(q, k_cache, v_cache, context_lens, scale,
+                              /*use_v4=*/true);
 }
 
 }  // namespace
@@ -202,6 +227,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("theta") = 500000.0);
   m.def("gqa_decode_attn", &gqa_decode_attn,
         "GQA decode attention over contiguous KV cache",
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("context_lens"), py::arg("scale"));
+  m.def("gqa_decode_attn_v4", &gqa_decode_attn_v4,
+        "GQA decode attention, MFMA-scores variant",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("context_lens"), py::arg("scale"));
 }
