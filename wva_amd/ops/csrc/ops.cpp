@@ -190,20 +190,21 @@ torch::Tensor gqa_decode_attn_impl(torch::Tensor q, torch::Tensor k_cache,
   return out;
 }
 
+// Default = v4 (MFMA scores): measured 25-33% faster than v3 across all
+// decode shapes on MI355X (profiles/attn_v4_ab.txt). v3 stays addressable
+// for regression comparison.
 torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
                               torch::Tensor v_cache,
                               torch::Tensor context_lens, double scale) {
   return gqa_decode_attn_impl(q, k_cache, v_cache, context_lens, scale,
-                              /*use_v4=*/false);
+                              /*use_v4=*/true);
 }
 
-// MFMA-scores variant (attention.hip v4) exposed separately for A/B
-// benchmarking; flips to the default once measured faster.
-torch::Tensor gqa_decode_attn_v4(torch::Tensor q, torch::Tensor k_cache,
+torch::Tensor gqa_decode_attn_v3(torch::Tensor q, torch::Tensor k_cache,
                                  torch::Tensor v_cache,
                                  torch::Tensor context_lens, double scale) {
   return gqa_decode_attn_impl(q, k_cache, v_cache, context_lens, scale,
-                              /*use_v4=*/true);
+                              /*use_v4=*/false);
 }
 
 }  // namespace
@@ -229,8 +230,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "GQA decode attention over contiguous KV cache",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("context_lens"), py::arg("scale"));
-  m.def("gqa_decode_attn_v4", &gqa_decode_attn_v4,
-        "GQA decode attention, MFMA-scores variant",
+  m.def("gqa_decode_attn_v4", &gqa_decode_attn,
+        "GQA decode attention, MFMA-scores variant (= default)",
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("context_lens"), py::arg("scale"));
+  m.def("gqa_decode_attn_v3", &gqa_decode_attn_v3,
+        "GQA decode attention, pre-MFMA shared-tile variant",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("context_lens"), py::arg("scale"));
 }
