@@ -443,12 +443,12 @@ __global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
 
 extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
                                           int max_ctx_hint) {
-  // Fill 4 resident workgroups per CU (the v3 kernel's admissible
-  // residency: 39 KiB LDS, 109 VGPRs) = 1024 WGs; cap so each split
-  // still gets >= 2 tiles of work.
+  // Fill the v4 kernel's resident workgroups: 70 VGPRs / ~21 KiB LDS
+  // admit 7 WGs/CU (compiler occupancy report) → target ~1792 WGs over
+  // 256 CUs; cap so each split still gets >= 2 tiles of work.
   const int base = batch * num_kv_heads;
-  if (base >= 1024) return 1;
-  int splits = 1024 / base;
+  if (base >= 1792) return 1;
+  int splits = 1792 / base;
   if (splits > 16) splits = 16;
   const int max_useful = max_ctx_hint > 0 ? (max_ctx_hint + 2 * TILE - 1) / (2 * TILE) : splits;
   if (splits > max_useful && max_useful >= 1) splits = max_useful;
