@@ -218,3 +218,40 @@ def tunableop_ab():
     print("tunableop A/B:", json.dumps(rec, indent=2))
     with open("gpurun_out/tunableop_ab.json", "w") as f:
         json.dump(rec, f, indent=2)
+
+
+def skinny_gemm_bench():
+    """skinny_linear vs hipBLASLt (torch.matmul) on the decode shapes."""
+    import time as _t
+
+    import torch
+
+    from wva_amd.ops import _require_ext
+    ext = _require_ext()
+
+    shapes = [  # (label, M, N, K)
+        ("qkv", 64, 6144, 4096), ("o", 64, 4096, 4096),
+        ("gate_up", 64, 28672, 4096), ("down", 64, 4096, 14336),
+        ("lm_head", 64, 128256, 4096),
+        ("qkv_b1", 1, 6144, 4096), ("gate_up_b1", 1, 28672, 4096),
+        ("down_b1", 1, 4096, 14336), ("lm_head_b1", 1, 128256, 4096),
+    ]
+    for label, M, N, K in shapes:
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+        wt = w.t()
+        for name, fn in [("blaslt", lambda: x @ wt),
+                         ("skinny", lambda: ext.skinny_linear(x, w))]:
+            for _ in range(5):
+                fn()
+            torch.cuda.synchronize()
+            n = 30
+            t0 = _t.perf_counter()
+            for _ in range(n):
+                fn()
+            torch.cuda.synchronize()
+            us = (_t.perf_counter() - t0) / n * 1e6
+            wbytes = 2 * N * K
+            print(f"{label:12s} [{name}] M={M:3d} N={N:6d} K={K:5d}: "
+                  f"{us:7.1f} us  W {wbytes/1e6:6.1f} MB  "
+                  f"{wbytes/us/1e3:5.2f} TB/s")

@@ -308,3 +308,35 @@ class TestAttnV4:
         a = ext.gqa_decode_attn(q, k, v, lens, 128 ** -0.5)
         b = ext.gqa_decode_attn_v4(q, k, v, lens, 128 ** -0.5)
         torch.testing.assert_close(a.float(), b.float(), atol=5e-3, rtol=5e-3)
+
+
+class TestSkinnyLinear:
+    @pytest.mark.parametrize("M,N,K", [
+        (1, 6144, 4096), (8, 4096, 4096), (16, 28672, 4096),
+        (17, 4096, 14336), (33, 128256, 4096), (64, 4096, 4096),
+        (64, 28672, 4096), (5, 100, 512), (64, 14336, 512),
+    ])
+    def test_matches_matmul(self, dev, M, N, K):
+        from wva_amd.ops import _require_ext
+
+        ext = _require_ext()
+        torch.manual_seed(M * 1000 + N)
+        x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.05
+        y = ext.skinny_linear(x, w)
+        ref = (x.float() @ w.float().t())
+        torch.testing.assert_close(y.float(), ref, atol=0.5, rtol=2e-2)
+
+    def test_linear_dispatch(self, dev):
+        from wva_amd import ops
+
+        x = torch.randn(4, 512, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(256, 512, device=dev, dtype=torch.bfloat16)
+        y = ops.linear(x, w)
+        torch.testing.assert_close(
+            y.float(), (x.float() @ w.float().t()), atol=0.5, rtol=2e-2
+        )
+        # big M falls back to hipBLASLt
+        xb = torch.randn(128, 512, device=dev, dtype=torch.bfloat16)
+        yb = ops.linear(xb, w)
+        assert yb.shape == (128, 256)

@@ -183,7 +183,7 @@ class LlamaDecodeModel:
             else:
                 h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
 
-            qkv = h @ layer.wqkv.t()  # hipBLASLt GEMM
+            qkv = ops.linear(h, layer.wqkv)  # skinny MFMA GEMM (M<=64)
             # fused: RoPE on the strided qkv row + KV-cache append
             q = ops.rope_append_kv(
                 qkv, self.k_cache[li][:B], self.v_cache[li][:B],
@@ -194,14 +194,14 @@ class LlamaDecodeModel:
             attn = ops.gqa_decode_attn(
                 q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
             )
-            x = attn.reshape(B, cfg.q_size) @ layer.wo.t()
+            x = ops.linear(attn.reshape(B, cfg.q_size), layer.wo)
 
             h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
-            gate_up = h2 @ layer.w_gate_up.t()
+            gate_up = ops.linear(h2, layer.w_gate_up)
             act = ops.silu_mul_fused(gate_up)
-            x = act @ layer.w_down.t()
+            x = ops.linear(act, layer.w_down)
 
         final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)
-        logits = final @ self.lm_head.t()
+        logits = ops.linear(final, self.lm_head)
         self.context_lens[:B] += 1
         return logits

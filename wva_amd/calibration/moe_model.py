@@ -172,9 +172,9 @@ class MixtralDecodeModel:
             if token_idx.numel() == 0:
                 continue
             x_e = h2.index_select(0, token_idx)
-            gate_up = x_e @ layer.w_gate_up[e].t()
+            gate_up = ops.linear(x_e, layer.w_gate_up[e])
             act = ops.silu_mul_fused(gate_up)
-            y = act @ layer.w_down[e].t()
+            y = ops.linear(act, layer.w_down[e])
             w_e = weights[token_idx, k_idx].unsqueeze(-1)
             out.index_add_(0, token_idx, (y * w_e).to(out.dtype))
         return out
@@ -195,7 +195,7 @@ class MixtralDecodeModel:
             else:
                 h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
 
-            qkv = h @ layer.wqkv.t()
+            qkv = ops.linear(h, layer.wqkv)
             q = ops.rope_append_kv(
                 qkv, self.k_cache[li][:B], self.v_cache[li][:B],
                 positions, cfg.num_q_heads, cfg.num_kv_heads, cfg.rope_theta,
@@ -204,12 +204,12 @@ class MixtralDecodeModel:
             attn = ops.gqa_decode_attn(
                 q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
             )
-            x = attn.reshape(B, cfg.q_size) @ layer.wo.t()
+            x = ops.linear(attn.reshape(B, cfg.q_size), layer.wo)
 
             h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
             x = self._moe_mlp(layer, h2)
 
         final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)
-        logits = final @ self.lm_head.t()
+        logits = ops.linear(final, self.lm_head)
         self.context_lens[:B] += 1
         return logits

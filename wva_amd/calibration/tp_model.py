@@ -154,7 +154,7 @@ class TPLlamaDecodeModel:
             else:
                 h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
 
-            qkv = h @ layer.wqkv.t()  # local heads only
+            qkv = ops.linear(h, layer.wqkv)  # local heads only
             q = ops.rope_append_kv(
                 qkv, self.k_cache[li][:B], self.v_cache[li][:B],
                 positions, shard.num_q_heads, shard.num_kv_heads,
@@ -165,14 +165,14 @@ class TPLlamaDecodeModel:
                 q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
             )
             # row-parallel o-proj: partials all-reduced over RCCL/xGMI
-            x = self._all_reduce(attn.reshape(B, shard.q_size) @ layer.wo.t())
+            x = self._all_reduce(ops.linear(attn.reshape(B, shard.q_size), layer.wo))
 
             h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
-            gate_up = h2 @ layer.w_gate_up.t()
+            gate_up = ops.linear(h2, layer.w_gate_up)
             act = ops.silu_mul_fused(gate_up)
-            x = self._all_reduce(act @ layer.w_down.t())
+            x = self._all_reduce(ops.linear(act, layer.w_down))
 
         final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)
-        logits = final @ self.lm_head.t()
+        logits = ops.linear(final, self.lm_head)
         self.context_lens[:B] += 1
         return logits

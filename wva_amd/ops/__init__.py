@@ -253,3 +253,17 @@ def enable_tuned_gemms() -> bool:
     tun.tuning_enable(False)  # read-only: never autotune in production
     tun.read_file(path)
     return True
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Decode linear y = x @ w.T ([M,K] @ [N,K]^T).
+
+    On GPU with M <= 64 and K % 128 == 0 this runs the weight-streaming
+    MFMA kernel (csrc/skinny_gemm.hip) — measured well above hipBLASLt's
+    tile kernels on decode shapes (profiles/). Larger shapes (prefill,
+    big batches) stay on hipBLASLt via torch.matmul, per the guide rule
+    that plain library GEMMs belong to the library.
+    """
+    if x.is_cuda and x.shape[0] <= 64 and x.shape[1] % 128 == 0:
+        return _require_ext().skinny_linear(x, w)
+    return x @ w.t()

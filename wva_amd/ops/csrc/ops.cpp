@@ -27,6 +27,11 @@ extern "C" void launch_rope_append_kv(const void* qkv, void* q_out,
                                       hipStream_t stream);
 extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
                                           int max_ctx_hint);
+extern "C" int skinny_gemm_num_splits(int N, int K, int nt);
+extern "C" int skinny_gemm_tile_n(int M);
+extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
+                                   const void* w, int M, int N, int K,
+                                   int num_splits, hipStream_t stream);
 extern "C" void launch_gqa_decode_attn(void* out, void* workspace,
                                        const void* q, const void* k_cache,
                                        const void* v_cache,
@@ -207,6 +212,34 @@ torch::Tensor gqa_decode_attn_v3(torch::Tensor q, torch::Tensor k_cache,
                               /*use_v4=*/false);
 }
 
+// Decode linear: y[M,N] = x[M,K] @ w[N,K]^T on the weight-streaming
+// MFMA kernel (csrc/skinny_gemm.hip). Caller guarantees M <= 64 and
+// K % 128 == 0; larger M belongs to hipBLASLt.
+torch::Tensor skinny_linear(torch::Tensor x, torch::Tensor w) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2, "x [M,K], w [N,K]");
+  const int M = x.size(0);
+  const int K = x.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  TORCH_CHECK(M <= 64, "skinny_linear requires M <= 64");
+  TORCH_CHECK(K % 128 == 0, "K must be a multiple of 128");
+  auto y = torch::empty({M, N}, x.options());
+  const int nt = skinny_gemm_tile_n(M);
+  const int sk = skinny_gemm_num_splits(N, K, nt);
+  torch::Tensor ws;
+  void* ws_ptr = nullptr;
+  if (sk > 1) {
+    ws = torch::empty({(long)sk * M * N},
+                      x.options().dtype(torch::kFloat32));
+    ws_ptr = ws.data_ptr();
+  }
+  launch_skinny_gemm(y.data_ptr(), ws_ptr, x.data_ptr(), w.data_ptr(), M, N,
+                     K, sk, current_stream());
+  return y;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -234,6 +267,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "GQA decode attention, MFMA-scores variant (= default)",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("context_lens"), py::arg("scale"));
+  m.def("skinny_linear", &skinny_linear,
+        "Weight-streaming decode GEMM: x[M,K] @ w[N,K]^T, M <= 64",
+        py::arg("x"), py::arg("w"));
   m.def("gqa_decode_attn_v3", &gqa_decode_attn_v3,
         "GQA decode attention, pre-MFMA shared-tile variant",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),

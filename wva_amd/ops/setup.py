@@ -26,6 +26,7 @@ setup(
                 os.path.join(this_dir, "csrc", "rope.hip"),
                 os.path.join(this_dir, "csrc", "silu_mul.hip"),
                 os.path.join(this_dir, "csrc", "attention.hip"),
+                os.path.join(this_dir, "csrc", "skinny_gemm.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
