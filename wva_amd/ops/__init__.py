@@ -258,12 +258,18 @@ def enable_tuned_gemms() -> bool:
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Decode linear y = x @ w.T ([M,K] @ [N,K]^T).
 
-    On GPU with M <= 64 and K % 128 == 0 this runs the weight-streaming
-    MFMA kernel (csrc/skinny_gemm.hip) — measured well above hipBLASLt's
-    tile kernels on decode shapes (profiles/). Larger shapes (prefill,
-    big batches) stay on hipBLASLt via torch.matmul, per the guide rule
-    that plain library GEMMs belong to the library.
+    Dispatch is measurement-driven (profiles/skinny_gemm_ab.txt):
+    hipBLASLt is near-roofline on big-N decode shapes (gate_up, lm_head)
+    but under-fills the 256-CU chip on small-N ones (its tile grid is
+    N/MT workgroups); the in-tree weight-streaming MFMA kernel
+    (csrc/skinny_gemm.hip) wins there via split-K fill. Everything else
+    (prefill, big batches) stays on hipBLASLt via torch.matmul.
     """
-    if x.is_cuda and x.shape[0] <= 64 and x.shape[1] % 128 == 0:
+    if (
+        x.is_cuda
+        and x.shape[0] <= 64
+        and x.shape[1] % 128 == 0
+        and w.shape[0] <= 8192  # small-N region (blaslt wins above)
+    ):
         return _require_ext().skinny_linear(x, w)
     return x @ w.t()

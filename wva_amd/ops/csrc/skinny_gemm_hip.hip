@@ -104,42 +104,41 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
     if (ch + 1 < chunks_per_split) stage(cur ^ 1, k0 + KT);
 
     // --- compute on the current chunk ---
+    // All of a wave's pairs share the same n-tile (pn = (wave + 4j) % nt
+    // is constant: nt | 4), so B loads hoist out of the pair loop; A
+    // fragments are single aligned b128 global loads (L2-resident), B
+    // fragments two b64 LDS reads (odd 66-dword rows are 8 B-aligned).
     const int a_col0 = 8 * (lane / 16);  // k offset of this lane's frag
-    for (int p = wave, slot = 0; p < num_pairs && slot < 4;
-         p += NUM_WAVES, ++slot) {
-      const int pm = p / nt;             // m-tile
-      const int pn = p % nt;             // n-tile within the WG
-      f32x4_frag d = acc[slot];
+    const int pn_w = wave % nt;
+    const int b_row = pn_w * 16 + (lane % 16);
+    const unsigned int* b_base = &w_smem[cur][b_row * W_ROW_DW];
 #pragma unroll
-      for (int kk = 0; kk < KT / 32; ++kk) {
-        // A fragment from global (L2-resident): row = m, k contiguous 8
-        bf16x8_frag a_frag;
-        {
-          const int m = pm * 16 + (lane % 16);
-          if (m < M) {
-            const short* src = reinterpret_cast<const short*>(
-                a + (long)m * K + k0 + kk * 32 + a_col0);
-#pragma unroll
-            for (int i = 0; i < 8; ++i) a_frag[i] = src[i];
-          } else {
-#pragma unroll
-            for (int i = 0; i < 8; ++i) a_frag[i] = 0;
-          }
-        }
-        // W (B operand) fragment from LDS: col = n = lane%16's row,
-        // same k window — aligned 16 B read
-        bf16x8_frag b_frag;
-        {
-          const int row = pn * 16 + (lane % 16);
-          const short* src = reinterpret_cast<const short*>(
-              &w_smem[cur][row * W_ROW_DW + (kk * 32 + a_col0) / 2]);
-#pragma unroll
-          for (int i = 0; i < 8; ++i) b_frag[i] = src[i];
-        }
-        d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, d, 0, 0,
-                                                    0);
+    for (int kk = 0; kk < KT / 32; ++kk) {
+      bf16x8_frag b_frag;
+      {
+        typedef __attribute__((ext_vector_type(2))) unsigned int u2;
+        const u2* src = reinterpret_cast<const u2*>(
+            &b_base[kk * 16 + a_col0 / 2]);
+        const u2 lo = src[0];
+        const u2 hi = src[1];
+        unsigned int words[4] = {lo[0], lo[1], hi[0], hi[1]};
+        b_frag = *reinterpret_cast<const bf16x8_frag*>(words);
       }
-      acc[slot] = d;
+      for (int p = wave, slot = 0; p < num_pairs && slot < 4;
+           p += NUM_WAVES, ++slot) {
+        const int pm = p / nt;
+        bf16x8_frag a_frag;
+        const int m = pm * 16 + (lane % 16);
+        if (m < M) {
+          a_frag = *reinterpret_cast<const bf16x8_frag*>(
+              a + (long)m * K + k0 + kk * 32 + a_col0);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) a_frag[i] = 0;
+        }
+        acc[slot] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag, acc[slot], 0, 0, 0);
+      }
     }
     __syncthreads();  // everyone done with `cur` before it is re-staged
   }
