@@ -36,6 +36,10 @@
 typedef __attribute__((ext_vector_type(8))) short bf16x8_frag;
 typedef __attribute__((ext_vector_type(4))) float f32x4_frag;
 
+// NSUB = n-subtiles (16-row groups) per wave: 1 for M <= 16, 2 for
+// larger M — the second subtile reuses the same A fragments, halving
+// the L2 A-read : HBM W-read ratio that bounds the M=64 case.
+template <int NSUB>
 __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
     bf16* __restrict__ c,          // [M, N] (null when split-K)
     float* __restrict__ ws,        // [SK, M, N] fp32 partials (or null)
@@ -49,10 +53,17 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
   const int sk = blockIdx.y;
   const int SK = gridDim.y;
 
-  // this wave's 16 W rows; a dead row folds into a zero fragment
-  const int n_row = blockIdx.x * NT + 16 * wave + (lane % 16);
-  const bool n_live = n_row < N;
-  const bf16* w_row = w + (long)(n_live ? n_row : 0) * K;
+  // this wave's NSUB×16 W rows; dead rows fold into zero fragments
+  const int wg_rows = NT * NSUB;
+  const int n_row0 = blockIdx.x * wg_rows + 16 * NSUB * wave + (lane % 16);
+  const bf16* w_rows[NSUB];
+  bool n_live[NSUB];
+#pragma unroll
+  for (int ns = 0; ns < NSUB; ++ns) {
+    const int n_row = n_row0 + 16 * ns;
+    n_live[ns] = n_row < N;
+    w_rows[ns] = w + (long)(n_live[ns] ? n_row : 0) * K;
+  }
 
   const int mt = (M + 15) / 16;          // ≤ 4 m-tiles
   const int m_row = lane % 16;           // within an m-tile
@@ -63,20 +74,26 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
   const int k_begin = sk * chunks_per_split * KT;
   const int k_end = k_begin + chunks_per_split * KT;
 
-  f32x4_frag acc[4];
+  f32x4_frag acc[4][NSUB];
 #pragma unroll
-  for (int p = 0; p < 4; ++p) acc[p] = f32x4_frag{0.f, 0.f, 0.f, 0.f};
+  for (int p = 0; p < 4; ++p)
+#pragma unroll
+    for (int ns = 0; ns < NSUB; ++ns)
+      acc[p][ns] = f32x4_frag{0.f, 0.f, 0.f, 0.f};
 
   for (int k0 = k_begin; k0 < k_end; k0 += KT) {
 #pragma unroll
     for (int kk = 0; kk < KT / 32; ++kk) {
       const int k = k0 + kk * 32 + col0;
-      bf16x8_frag b_frag;
-      if (n_live) {
-        b_frag = *reinterpret_cast<const bf16x8_frag*>(w_row + k);
-      } else {
+      bf16x8_frag b_frag[NSUB];
 #pragma unroll
-        for (int i = 0; i < 8; ++i) b_frag[i] = 0;
+      for (int ns = 0; ns < NSUB; ++ns) {
+        if (n_live[ns]) {
+          b_frag[ns] = *reinterpret_cast<const bf16x8_frag*>(w_rows[ns] + k);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) b_frag[ns][i] = 0;
+        }
       }
 #pragma unroll
       for (int pm = 0; pm < 4; ++pm) {
@@ -89,26 +106,33 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
 #pragma unroll
           for (int i = 0; i < 8; ++i) a_frag[i] = 0;
         }
-        acc[pm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
-                                                          acc[pm], 0, 0, 0);
+#pragma unroll
+        for (int ns = 0; ns < NSUB; ++ns) {
+          acc[pm][ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag[ns], acc[pm][ns], 0, 0, 0);
+        }
       }
     }
   }
 
-  // --- store: lane holds (m = (lane>>4)*4 + i, n = 16*wave + lane&15) ---
-  const int n_out = blockIdx.x * NT + 16 * wave + (lane & 15);
-  if (n_out >= N) return;
+  // --- store: lane holds (m = (lane>>4)*4 + i, n = ... + lane&15) ---
 #pragma unroll
-  for (int pm = 0; pm < 4; ++pm) {
-    if (pm >= mt) break;
+  for (int ns = 0; ns < NSUB; ++ns) {
+    const int n_out =
+        blockIdx.x * wg_rows + 16 * NSUB * wave + 16 * ns + (lane & 15);
+    if (n_out >= N) continue;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int m = pm * 16 + (lane >> 4) * 4 + i;
-      if (m >= M) continue;
-      if (ws != nullptr) {
-        ws[((long)sk * M + m) * N + n_out] = acc[pm][i];
-      } else {
-        c[(long)m * N + n_out] = f2bf(acc[pm][i]);
+    for (int pm = 0; pm < 4; ++pm) {
+      if (pm >= mt) break;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int m = pm * 16 + (lane >> 4) * 4 + i;
+        if (m >= M) continue;
+        if (ws != nullptr) {
+          ws[((long)sk * M + m) * N + n_out] = acc[pm][ns][i];
+        } else {
+          c[(long)m * N + n_out] = f2bf(acc[pm][ns][i]);
+        }
       }
     }
   }
@@ -126,10 +150,9 @@ __global__ __launch_bounds__(256) void skinny_gemm_merge_kernel(
 }
 
 // SK heuristic: fill ≈1792 workgroups; SK must divide K/KT, power of two
-// ≤ 16.
+// ≤ 16. `nt` is the workgroup N-tile width (64 or 128 by M).
 extern "C" int skinny_gemm_num_splits(int N, int K, int nt) {
-  (void)nt;
-  const int n_wgs = (N + NT - 1) / NT;
+  const int n_wgs = (N + nt - 1) / nt;
   const int k_chunks = K / KT;
   int sk = 1;
   while (sk < 16 && n_wgs * sk * 2 <= 1792 && (k_chunks % (sk * 2)) == 0) {
@@ -141,12 +164,18 @@ extern "C" int skinny_gemm_num_splits(int N, int K, int nt) {
 extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
                                    const void* w, int M, int N, int K,
                                    int num_splits, hipStream_t stream) {
-  dim3 grid((N + NT - 1) / NT, num_splits);
+  const int nt = (M > 16) ? 2 * NT : NT;
+  dim3 grid((N + nt - 1) / nt, num_splits);
   dim3 block(256);
   float* ws_ptr = num_splits > 1 ? (float*)ws : nullptr;
   bf16* c_ptr = num_splits > 1 ? nullptr : (bf16*)c;
-  hipLaunchKernelGGL(skinny_gemm_kernel, grid, block, 0, stream, c_ptr,
-                     ws_ptr, (const bf16*)a, (const bf16*)w, M, N, K);
+  if (nt == NT) {
+    hipLaunchKernelGGL(skinny_gemm_kernel<1>, grid, block, 0, stream, c_ptr,
+                       ws_ptr, (const bf16*)a, (const bf16*)w, M, N, K);
+  } else {
+    hipLaunchKernelGGL(skinny_gemm_kernel<2>, grid, block, 0, stream, c_ptr,
+                       ws_ptr, (const bf16*)a, (const bf16*)w, M, N, K);
+  }
   if (num_splits > 1) {
     const long MN = (long)M * N;
     dim3 mgrid((MN + 255) / 256);
@@ -155,7 +184,4 @@ extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
   }
 }
 
-extern "C" int skinny_gemm_tile_n(int M) {
-  (void)M;
-  return NT;
-}
+extern "C" int skinny_gemm_tile_n(int M) { return (M > 16) ? 2 * NT : NT; }
