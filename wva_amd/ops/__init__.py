@@ -267,9 +267,10 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """
     if (
         x.is_cuda
-        and x.shape[0] <= 64
+        and x.shape[0] <= 16
         and x.shape[1] % 128 == 0
-        and w.shape[0] <= 8192  # small-N region (blaslt wins above)
+        and x.shape[1] <= 8192      # long-K loses to blaslt (down-proj)
+        and w.shape[0] <= 8192      # big-N loses to blaslt (gate_up/head)
     ):
         return _require_ext().skinny_linear(x, w)
     return x @ w.t()
