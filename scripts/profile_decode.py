@@ -70,8 +70,9 @@ def attn_bench():
         ext = _require_ext()
         scale = 128 ** -0.5
         import time as _t
-        for name, fn in [("v3", ext.gqa_decode_attn),
-                         ("v4", ext.gqa_decode_attn_v4)]:
+        for name, fn in [("v3", ext.gqa_decode_attn_v3),
+                         ("v4", ext.gqa_decode_attn_v4),
+                         ("v5", ext.gqa_decode_attn_v5)]:
             for _ in range(3):
                 fn(q, k, v, lens, scale)
             torch.cuda.synchronize()

@@ -340,3 +340,32 @@ class TestSkinnyLinear:
         xb = torch.randn(128, 512, device=dev, dtype=torch.bfloat16)
         yb = ops.linear(xb, w)
         assert yb.shape == (128, 256)
+
+
+class TestAttnV5:
+    @pytest.mark.parametrize("batch,hq,hk,ctx", [
+        (2, 8, 2, 33), (4, 32, 8, 512), (2, 64, 8, 300), (3, 4, 4, 100),
+        (2, 8, 1, 2000),  # split path
+    ])
+    def test_v5_matches_reference(self, dev, batch, hq, hk, ctx):
+        from wva_amd.ops import _require_ext, gqa_decode_attn_ref
+
+        ext = _require_ext()
+        torch.manual_seed(17)
+        S = max(ctx + 8, 64)
+        q = torch.randn(batch, hq, 128, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(batch, hk, S, 128, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(batch, hk, S, 128, device=dev, dtype=torch.bfloat16)
+        lens = torch.randint(1, ctx + 1, (batch,), device=dev,
+                             dtype=torch.int32)
+        lens[0] = ctx
+        scale = 128 ** -0.5
+        out = ext.gqa_decode_attn_v5(q, k, v, lens, scale)
+        ref = gqa_decode_attn_ref(
+            q.float().cpu(), k.float().cpu(), v.float().cpu(),
+            lens.cpu(), scale,
+        )
+        # P is cast to bf16 for the MFMA PV (FA2-standard) — slightly
+        # looser tolerance than v4's fp32 PV accumulation
+        torch.testing.assert_close(out.float().cpu(), ref, atol=4e-2,
+                                   rtol=4e-2)

@@ -234,6 +234,12 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_kernel(
 typedef __attribute__((ext_vector_type(8))) short bf16x8_frag;
 typedef __attribute__((ext_vector_type(4))) float f32x4_frag;
 
+// fp32 -> bf16 raw bits (round-to-nearest-even via hardware convert)
+__device__ __forceinline__ short f2bf_bits(float v) {
+  const bf16 b = f2bf(v);
+  return *reinterpret_cast<const short*>(&b);
+}
+
 #define S_ROW 65  // s_smem row stride (floats): 64 + 1 pad
 
 __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v4_kernel(
@@ -404,6 +410,247 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v4_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v5: v4 + MFMA PV. PMC counters showed v4 issue-bound on VALU (4.07 G
+// VALU vs 30 M MFMA instructions — the vector PV sweep, 64 positions ×
+// 2 fma × heads per wave per tile). v5 computes O += P·V on the matrix
+// cores too:
+//   * P [16 heads × 64 pos] (zero-padded heads) as A fragments from
+//     p_smem, cast fp32→bf16 at fragment build (FA2-standard precision);
+//   * V as B fragments straight from v_smem u16 reads (no conversion —
+//     V is bf16 in LDS);
+//   * wave w owns head_dim n-subtiles {w, w+4} → 2 persistent f32x4
+//     accumulators; online-softmax rescale multiplies each accumulator
+//     register by corr[head(reg)] from LDS before the tile's 4 MFMAs.
+// Needs one extra barrier per tile (every wave now reads every head's
+// p row) and LDS broadcast of corr/s; softmax state stays wave-owned.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
+    bf16* __restrict__ out,
+    float* __restrict__ workspace,
+    const bf16* __restrict__ q,
+    const bf16* __restrict__ k_cache,
+    const bf16* __restrict__ v_cache,
+    const int* __restrict__ context_lens,
+    const int num_q_heads,
+    const int num_kv_heads,
+    const int max_seq,
+    const float scale) {
+  const int b = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int split = blockIdx.z;
+  const int num_splits = gridDim.z;
+  const int G = num_q_heads / num_kv_heads;
+  const int ctx = context_lens[b];
+
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+
+  __shared__ float s_smem[MAX_G][S_ROW];
+  __shared__ unsigned int v_smem[TILE * ROW_DW];
+  __shared__ float p_smem[MAX_G][TILE];
+  __shared__ float corr_smem[MAX_G];
+  __shared__ float sum_smem[MAX_G];
+  __shared__ float m_smem[MAX_G];
+
+  // Q fragments (same as v4)
+  bf16x8_frag q_frag[4];
+  {
+    const int g = lane % 16;
+    const int k0 = 8 * (lane / 16);
+    if (g < G) {
+      const bf16* qrow = q + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM;
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        const short* src = reinterpret_cast<const short*>(qrow + 32 * kk + k0);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) q_frag[kk][i] = src[i];
+      }
+    } else {
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+        for (int i = 0; i < 8; ++i) q_frag[kk][i] = 0;
+    }
+  }
+
+  const int heads_mine = (G > wave) ? (G - wave + NUM_WAVES - 1) / NUM_WAVES : 0;
+  float m[2], s[2];
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    m[j] = -INFINITY;
+    s[j] = 0.0f;
+  }
+  // PV accumulators: wave owns n-subtiles wave and wave+4 of head_dim;
+  // lane holds (head = (lane>>4)*4 + reg, dim = 16*nsub + lane&15).
+  f32x4_frag acc_pv[2] = {f32x4_frag{0.f, 0.f, 0.f, 0.f},
+                          f32x4_frag{0.f, 0.f, 0.f, 0.f}};
+
+  const long kv_row_dw = HEAD_DIM / 2;
+  const bf16* k_slab =
+      k_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM;
+  const unsigned int* v_base = reinterpret_cast<const unsigned int*>(
+      v_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM);
+
+  for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
+    const int tn = min(TILE, ctx - t0);
+
+    // --- stage V (v4 loop) ---
+    {
+      typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
+      const int tile_u2 = tn * (HEAD_DIM / 4);
+      const unsigned int* v_src = &v_base[(long)t0 * kv_row_dw];
+      for (int idx = wave * WAVE_SIZE + lane; idx < tile_u2;
+           idx += NUM_WAVES * WAVE_SIZE) {
+        const int row = idx >> 5;
+        const int d2 = (idx & 31) * 2;
+        const uint2_t val = *reinterpret_cast<const uint2_t*>(&v_src[idx * 2]);
+        *reinterpret_cast<uint2_t*>(&v_smem[row * ROW_DW + d2]) = val;
+      }
+    }
+
+    // --- MFMA scores (v4) ---
+    {
+      const int prow = 16 * wave + (lane % 16);
+      const int pos = t0 + prow;
+      const int pos_c = pos < max_seq ? pos : max_seq - 1;
+      const short* krow =
+          reinterpret_cast<const short*>(k_slab + (long)pos_c * HEAD_DIM);
+      const int k0 = 8 * (lane / 16);
+      f32x4_frag sf = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8_frag a_frag;
+        const short* src = krow + 32 * kk + k0;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) a_frag[i] = src[i];
+        sf = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, q_frag[kk], sf,
+                                                     0, 0, 0);
+      }
+      const int g = lane & 15;
+      if (g < G) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int t = 16 * wave + (lane >> 4) * 4 + i;
+          s_smem[g][t] = (t < tn) ? sf[i] * scale : -INFINITY;
+        }
+      }
+    }
+    __syncthreads();  // V + scores visible
+
+    // --- softmax per owned head (v4) + publish p, corr ---
+    const bool live = lane < tn;
+    for (int j = 0; j < heads_mine; ++j) {
+      const int g = wave + j * NUM_WAVES;
+      const float score = live ? s_smem[g][lane] : -INFINITY;
+      const float tile_max = wave_reduce_max(score);
+      const float m_new = fmaxf(m[j], tile_max);
+      const float p = live ? __expf(score - m_new) : 0.0f;
+      const float tile_sum = wave_reduce_sum(p);
+      const float corr = (m[j] == -INFINITY) ? 0.0f : __expf(m[j] - m_new);
+      s[j] = s[j] * corr + tile_sum;
+      m[j] = m_new;
+      p_smem[g][lane] = p;
+      if (lane == 0) corr_smem[g] = corr;
+    }
+    // dead heads (g >= G) keep corr undefined; PV zero-pads their rows
+    __syncthreads();  // p + corr visible to all waves
+
+    // --- MFMA PV: O[16 heads × dims] += P[16×64] · V[64×dims] ---
+    {
+      // A fragments (P): shared by this wave's two n-subtiles
+      bf16x8_frag p_frag[2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int g = lane % 16;
+        const int pos0 = 32 * kk + 8 * (lane / 16);
+        if (g < G) {
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            p_frag[kk][i] = f2bf_bits(p_smem[g][pos0 + i]);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) p_frag[kk][i] = 0;
+        }
+      }
+      const unsigned short* v_u16 =
+          reinterpret_cast<const unsigned short*>(v_smem);
+#pragma unroll
+      for (int nsub = 0; nsub < 2; ++nsub) {
+        const int dim = 16 * (wave + 4 * nsub) + (lane % 16);
+        // rescale accumulator by corr[head(reg)]
+        f32x4_frag d = acc_pv[nsub];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int h = (lane >> 4) * 4 + i;
+          d[i] *= (h < G) ? corr_smem[h] : 0.0f;
+        }
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          bf16x8_frag v_frag;
+          const int pos0 = 32 * kk + 8 * (lane / 16);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            const int t = pos0 + i;
+            v_frag[i] = (short)v_u16[(t * ROW_DW + (dim >> 1)) * 2 +
+                                     (dim & 1)];
+          }
+          d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[kk], v_frag, d,
+                                                      0, 0, 0);
+        }
+        acc_pv[nsub] = d;
+      }
+    }
+    __syncthreads();  // PV consumed V/p before next tile overwrites
+  }
+
+  // publish per-head softmax sums (and maxima for the split path)
+  for (int j = 0; j < heads_mine; ++j) {
+    const int g = wave + j * NUM_WAVES;
+    if (lane == 0) {
+      sum_smem[g] = s[j];
+      m_smem[g] = m[j];
+    }
+  }
+  __syncthreads();
+
+  if (num_splits == 1) {
+    // lane holds (head = (lane>>4)*4+reg, dim = 16*(wave+4*nsub)+lane&15)
+#pragma unroll
+    for (int nsub = 0; nsub < 2; ++nsub) {
+      const int dim = 16 * (wave + 4 * nsub) + (lane & 15);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int h = (lane >> 4) * 4 + i;
+        if (h >= G) continue;
+        const float sv = sum_smem[h];
+        const float inv = sv > 0.0f ? 1.0f / sv : 0.0f;
+        out[((long)b * num_q_heads + kvh * G + h) * HEAD_DIM + dim] =
+            f2bf(acc_pv[nsub][i] * inv);
+      }
+    }
+  } else {
+#pragma unroll
+    for (int nsub = 0; nsub < 2; ++nsub) {
+      const int dim = 16 * (wave + 4 * nsub) + (lane & 15);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int h = (lane >> 4) * 4 + i;
+        if (h >= G) continue;
+        float* wsp = workspace +
+            ((((long)b * num_kv_heads + kvh) * G + h) * num_splits + split) *
+                (2 + HEAD_DIM);
+        if (wave == 0 && nsub == 0 && (lane & 15) == 0) {
+          wsp[0] = m_smem[h];
+          wsp[1] = sum_smem[h];
+        }
+        wsp[2 + dim] = acc_pv[nsub][i];
+      }
+    }
+  }
+}
+
 // Merge split-KV partials: one 64-thread wave per (b, q_head).
 __global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
     bf16* __restrict__ out,          // [B, Hq, 128]
@@ -463,6 +710,25 @@ extern "C" void launch_gqa_decode_attn(
   dim3 grid(batch, num_kv_heads, num_splits);
   dim3 block(256);
   hipLaunchKernelGGL(gqa_decode_attn_kernel, grid, block, 0, stream,
+                     (bf16*)out, (float*)workspace, (const bf16*)q,
+                     (const bf16*)k_cache, (const bf16*)v_cache, context_lens,
+                     num_q_heads, num_kv_heads, max_seq, scale);
+  if (num_splits > 1) {
+    dim3 mgrid(batch, num_q_heads);
+    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64), 0,
+                       stream, (bf16*)out, (const float*)workspace,
+                       num_q_heads, num_kv_heads, num_splits);
+  }
+}
+
+extern "C" void launch_gqa_decode_attn_v5(
+    void* out, void* workspace, const void* q, const void* k_cache,
+    const void* v_cache, const int* context_lens, int batch, int num_q_heads,
+    int num_kv_heads, int max_seq, int num_splits, float scale,
+    hipStream_t stream) {
+  dim3 grid(batch, num_kv_heads, num_splits);
+  dim3 block(256);
+  hipLaunchKernelGGL(gqa_decode_attn_v5_kernel, grid, block, 0, stream,
                      (bf16*)out, (float*)workspace, (const bf16*)q,
                      (const bf16*)k_cache, (const bf16*)v_cache, context_lens,
                      num_q_heads, num_kv_heads, max_seq, scale);

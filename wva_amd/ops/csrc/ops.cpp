@@ -39,6 +39,13 @@ extern "C" void launch_gqa_decode_attn(void* out, void* workspace,
                                        int num_q_heads, int num_kv_heads,
                                        int max_seq, int num_splits,
                                        float scale, hipStream_t stream);
+extern "C" void launch_gqa_decode_attn_v5(void* out, void* workspace,
+                                          const void* q, const void* k_cache,
+                                          const void* v_cache,
+                                          const int* context_lens, int batch,
+                                          int num_q_heads, int num_kv_heads,
+                                          int max_seq, int num_splits,
+                                          float scale, hipStream_t stream);
 extern "C" void launch_gqa_decode_attn_v4(void* out, void* workspace,
                                           const void* q, const void* k_cache,
                                           const void* v_cache,
@@ -155,7 +162,7 @@ torch::Tensor silu_mul_fused(torch::Tensor gate_up) {
 torch::Tensor gqa_decode_attn_impl(torch::Tensor q, torch::Tensor k_cache,
                                    torch::Tensor v_cache,
                                    torch::Tensor context_lens, double scale,
-                                   bool use_v4) {
+                                   int use_v4) {
   check_bf16_contig(q, "q");
   check_bf16_contig(k_cache, "k_cache");
   check_bf16_contig(v_cache, "v_cache");
@@ -186,7 +193,9 @@ torch::Tensor gqa_decode_attn_impl(torch::Tensor q, torch::Tensor k_cache,
         q.options().dtype(torch::kFloat32));
     ws_ptr = workspace.data_ptr();
   }
-  auto launch = use_v4 ? launch_gqa_decode_attn_v4 : launch_gqa_decode_attn;
+  auto launch = use_v4 == 2 ? launch_gqa_decode_attn_v5
+                : use_v4 == 1 ? launch_gqa_decode_attn_v4
+                              : launch_gqa_decode_attn;
   launch(out.data_ptr(), ws_ptr, q.data_ptr(),
          k_cache.data_ptr(), v_cache.data_ptr(),
          context_lens.data_ptr<int>(), batch, num_q_heads,
@@ -202,14 +211,22 @@ torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
                               torch::Tensor v_cache,
                               torch::Tensor context_lens, double scale) {
   return gqa_decode_attn_impl(q, k_cache, v_cache, context_lens, scale,
-                              /*use_v4=*/true);
+                              /*use_v4=*/1);
+}
+
+// v5 (MFMA scores + MFMA PV) — A/B variant until measured faster.
+torch::Tensor gqa_decode_attn_v5(torch::Tensor q, torch::Tensor k_cache,
+                                 torch::Tensor v_cache,
+                                 torch::Tensor context_lens, double scale) {
+  return gqa_decode_attn_impl(q, k_cache, v_cache, context_lens, scale,
+                              /*use_v4=*/2);
 }
 
 torch::Tensor gqa_decode_attn_v3(torch::Tensor q, torch::Tensor k_cache,
                                  torch::Tensor v_cache,
                                  torch::Tensor context_lens, double scale) {
   return gqa_decode_attn_impl(q, k_cache, v_cache, context_lens, scale,
-                              /*use_v4=*/false);
+                              /*use_v4=*/0);
 }
 
 // Decode linear: y[M,N] = x[M,K] @ w[N,K]^T on the weight-streaming
@@ -270,6 +287,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_linear", &skinny_linear,
         "Weight-streaming decode GEMM: x[M,K] @ w[N,K]^T, M <= 64",
         py::arg("x"), py::arg("w"));
+  m.def("gqa_decode_attn_v5", &gqa_decode_attn_v5,
+        "GQA decode attention, MFMA scores + MFMA PV variant",
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("context_lens"), py::arg("scale"));
   m.def("gqa_decode_attn_v3", &gqa_decode_attn_v3,
         "GQA decode attention, pre-MFMA shared-tile variant",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
