@@ -594,8 +594,11 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
 #pragma unroll
           for (int i = 0; i < 8; ++i) {
             const int t = pos0 + i;
-            v_frag[i] = (short)v_u16[(t * ROW_DW + (dim >> 1)) * 2 +
-                                     (dim & 1)];
+            // t >= tn rows are unstaged (stale LDS can decode to NaN,
+            // and 0·NaN = NaN even though p[t] = 0) — zero them
+            v_frag[i] = (t < tn)
+                ? (short)v_u16[(t * ROW_DW + (dim >> 1)) * 2 + (dim & 1)]
+                : (short)0;
           }
           d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag[kk], v_frag, d,
                                                       0, 0, 0);
