@@ -204,14 +204,20 @@ torch::Tensor gqa_decode_attn_impl(torch::Tensor q, torch::Tensor k_cache,
   return out;
 }
 
-// Default = v4 (MFMA scores): measured 25-33% faster than v3 across all
-// decode shapes on MI355X (profiles/attn_v4_ab.txt). v3 stays addressable
-// for regression comparison.
+// Default dispatch, measured per shape class (profiles/attn_v4_ab.txt):
+// v5 (MFMA scores + MFMA PV) wins when the PV matrix is well-utilized
+// (G = 8) or the grid is unsplit (large batch); v4 (MFMA scores, vector
+// PV) wins for G < 8 with split-KV where v5's padded 16-row PV and its
+// third per-tile barrier cost more than the VALU sweep it replaces.
 torch::Tensor gqa_decode_attn(torch::Tensor q, torch::Tensor k_cache,
                               torch::Tensor v_cache,
                               torch::Tensor context_lens, double scale) {
+  const int G = (int)(q.size(1) / k_cache.size(1));
+  const int splits = gqa_decode_attn_num_splits(
+      (int)q.size(0), (int)k_cache.size(1), (int)k_cache.size(2));
+  const int variant = (G >= 8 || splits == 1) ? 2 : 1;
   return gqa_decode_attn_impl(q, k_cache, v_cache, context_lens, scale,
-                              /*use_v4=*/1);
+                              variant);
 }
 
 // v5 (MFMA scores + MFMA PV) — A/B variant until measured faster.
