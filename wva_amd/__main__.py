@@ -11,6 +11,7 @@ FakeCluster implements the client surface a REST-backed client would).
 from __future__ import annotations
 
 import argparse
+import os
 import signal
 import sys
 import time
@@ -38,6 +39,15 @@ def parse_flags(argv=None):
         action="store_true",
         help="run against the in-memory emulated cluster",
     )
+    p.add_argument(
+        "--kube-api-url",
+        default=None,
+        help="Kubernetes API server URL (REST mode); in-cluster config is "
+             "auto-detected via KUBERNETES_SERVICE_HOST when omitted",
+    )
+    p.add_argument("--kube-token", default=None)
+    p.add_argument("--kube-ca-cert", default=None)
+    p.add_argument("--kube-insecure-skip-verify", action="store_true")
     return p.parse_args(argv)
 
 
@@ -57,12 +67,32 @@ def main(argv=None) -> int:
     )
     setup_logging(config.infra.logger_verbosity or 2)
 
-    cluster = FakeCluster()
     source = None
     if args.emulated:
+        cluster = FakeCluster()
         sim = ClusterSim(cluster)
         source = SimMetricsSource(sim)
         log.info("running in emulated mode (in-memory cluster)")
+    elif args.kube_api_url or os.environ.get("KUBERNETES_SERVICE_HOST"):
+        from .kube.rest import RestCluster
+
+        if args.kube_api_url:
+            cluster = RestCluster(
+                args.kube_api_url,
+                token=args.kube_token,
+                ca_cert_path=args.kube_ca_cert,
+                insecure_skip_verify=args.kube_insecure_skip_verify,
+            )
+            log.info("REST mode against %s", args.kube_api_url)
+        else:
+            cluster = RestCluster.in_cluster()
+            log.info("in-cluster REST mode")
+    else:
+        cluster = FakeCluster()
+        log.warning(
+            "no --emulated, --kube-api-url or in-cluster environment: "
+            "running against an empty in-memory cluster"
+        )
 
     app = build_app(cluster, config, source=source, serve_http=True)
     app.configmap_reconciler.bootstrap_initial_configmaps()

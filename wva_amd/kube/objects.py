@@ -231,6 +231,21 @@ class ServiceMonitor:
 
 
 @dataclass
+class Namespace:
+    """core/v1 Namespace — read for the exclusion annotation
+    `wva.llmd.ai/exclude` (reference predicates.go:184-243)."""
+
+    metadata: ObjectMeta = field(default_factory=ObjectMeta)
+
+    kind: str = "Namespace"
+    api_version: str = "v1"
+
+    @property
+    def name(self) -> str:
+        return self.metadata.name
+
+
+@dataclass
 class Lease:
     """coordination.k8s.io/v1 Lease used for leader election."""
 

@@ -1,0 +1,349 @@
+"""REST-backed Kubernetes cluster client.
+
+Implements the same surface as FakeCluster (create/get/try_get/list/
+update/update_status/delete/scale/record_event/watch/stop_watch) against
+a real Kubernetes API server, so the whole controller stack —
+`build_app(cluster=RestCluster(...), ...)` — runs unchanged in-cluster
+or against a kubeconfig'd cluster. The reference achieves the same with
+controller-runtime's client + cache (cmd/main.go:266-297); this client
+is deliberately cache-less: reads go to the API server, and the Manager
+layer's watch dispatch provides the event-driven path.
+
+Auth: bearer token (in-cluster serviceaccount token file or explicit),
+CA bundle or insecure-skip-verify. Watches use the API server's
+`?watch=true` chunked-JSON stream, one pump thread per watch() call,
+fanned into the same queue.Queue[WatchEvent] contract FakeCluster
+serves. BOOKMARKs are consumed for resourceVersion continuity; streams
+auto-reconnect from the last seen resourceVersion (client-go informer
+semantics, simplified).
+
+No network exists in the build environment, so tests run this client
+against an in-process minimal API server backed by a FakeCluster
+(tests/test_rest_cluster.py) — list/get/create/put/patch/delete/watch
+round-trip through real HTTP and the serde layer.
+"""
+from __future__ import annotations
+
+import json
+import ssl
+import threading
+import queue
+import urllib.error
+import urllib.request
+from typing import Any, Dict, List, Optional
+
+from ..utils.logging import get_logger
+from . import serde
+from .fake import ADDED, DELETED, MODIFIED, ConflictError, NotFoundError, WatchEvent
+
+log = get_logger("kube.rest")
+
+SERVICEACCOUNT_TOKEN = "/var/run/secrets/kubernetes.io/serviceaccount/token"
+SERVICEACCOUNT_CA = "/var/run/secrets/kubernetes.io/serviceaccount/ca.crt"
+
+
+class ApiError(RuntimeError):
+    def __init__(self, status: int, body: str):
+        super().__init__(f"API error {status}: {body[:200]}")
+        self.status = status
+        self.body = body
+
+
+class RestCluster:
+    """FakeCluster-compatible client over the Kubernetes REST API."""
+
+    def __init__(
+        self,
+        base_url: str,
+        token: Optional[str] = None,
+        token_path: Optional[str] = None,
+        ca_cert_path: Optional[str] = None,
+        insecure_skip_verify: bool = False,
+        timeout_seconds: float = 10.0,
+        watch_timeout_seconds: float = 300.0,
+    ):
+        self.base_url = base_url.rstrip("/")
+        self._token = token
+        self._token_path = token_path
+        self.timeout = timeout_seconds
+        self.watch_timeout = watch_timeout_seconds
+        if self.base_url.startswith("https"):
+            if insecure_skip_verify:
+                self._ssl = ssl._create_unverified_context()
+            else:
+                self._ssl = ssl.create_default_context(
+                    cafile=ca_cert_path or None
+                )
+        else:
+            self._ssl = None
+        self._watches: List["_WatchPump"] = []
+        self._lock = threading.Lock()
+
+    @classmethod
+    def in_cluster(cls, **kwargs) -> "RestCluster":
+        """In-cluster config: KUBERNETES_SERVICE_{HOST,PORT} + SA token."""
+        import os
+
+        host = os.environ["KUBERNETES_SERVICE_HOST"]
+        port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+        return cls(
+            f"https://{host}:{port}",
+            token_path=SERVICEACCOUNT_TOKEN,
+            ca_cert_path=SERVICEACCOUNT_CA,
+            **kwargs,
+        )
+
+    # --- HTTP plumbing ---
+
+    def _headers(self, content_type: Optional[str] = None) -> Dict[str, str]:
+        h = {"Accept": "application/json"}
+        token = self._token
+        if token is None and self._token_path:
+            try:
+                with open(self._token_path) as f:
+                    token = f.read().strip()
+            except OSError:
+                token = None
+        if token:
+            h["Authorization"] = f"Bearer {token}"
+        if content_type:
+            h["Content-Type"] = content_type
+        return h
+
+    def _request(
+        self,
+        method: str,
+        path: str,
+        body: Optional[Dict[str, Any]] = None,
+        content_type: str = "application/json",
+        timeout: Optional[float] = None,
+    ) -> Dict[str, Any]:
+        url = self.base_url + path
+        data = json.dumps(body).encode() if body is not None else None
+        req = urllib.request.Request(
+            url, data=data, method=method,
+            headers=self._headers(content_type if data else None),
+        )
+        try:
+            with urllib.request.urlopen(
+                req, timeout=timeout or self.timeout, context=self._ssl
+            ) as resp:
+                payload = resp.read()
+        except urllib.error.HTTPError as e:
+            body_text = e.read().decode(errors="replace")
+            if e.code == 404:
+                raise NotFoundError("?", "?", path) from e
+            if e.code == 409:
+                raise ConflictError(body_text) from e
+            raise ApiError(e.code, body_text) from e
+        return json.loads(payload) if payload else {}
+
+    # --- FakeCluster surface ---
+
+    def create(self, obj: Any) -> Any:
+        kind = obj.kind
+        path = serde.resource_path(kind, obj.metadata.namespace)
+        out = self._request("POST", path, serde.encode(obj))
+        return serde.decode(kind, out)
+
+    def get(self, kind: str, namespace: str, name: str) -> Any:
+        path = serde.resource_path(kind, namespace, name)
+        try:
+            out = self._request("GET", path)
+        except NotFoundError:
+            raise NotFoundError(kind, namespace, name) from None
+        return serde.decode(kind, out)
+
+    def try_get(self, kind: str, namespace: str, name: str) -> Optional[Any]:
+        try:
+            return self.get(kind, namespace, name)
+        except NotFoundError:
+            return None
+
+    def list(
+        self,
+        kind: str,
+        namespace: Optional[str] = None,
+        label_selector: Optional[Dict[str, str]] = None,
+    ) -> List[Any]:
+        path = serde.resource_path(kind, namespace)
+        if label_selector:
+            sel = ",".join(f"{k}={v}" for k, v in sorted(label_selector.items()))
+            path += "?labelSelector=" + urllib.request.quote(sel)
+        out = self._request("GET", path)
+        return [serde.decode(kind, item) for item in out.get("items", [])]
+
+    def update(self, obj: Any, bump_generation: bool = False) -> Any:
+        # bump_generation is API-server behavior on real clusters (spec
+        # changes bump it); accepted for signature compatibility.
+        kind = obj.kind
+        path = serde.resource_path(kind, obj.metadata.namespace, obj.metadata.name)
+        out = self._request("PUT", path, serde.encode(obj))
+        return serde.decode(kind, out)
+
+    def update_status(self, obj: Any) -> Any:
+        """Status subresource write — PATCH (merge) with the full nested
+        desired object, matching the reference's full-object patch-base
+        workaround for CRD partial-patch validation (#731,
+        variantautoscaling_controller.go:244-252)."""
+        kind = obj.kind
+        path = serde.resource_path(
+            kind, obj.metadata.namespace, obj.metadata.name
+        ) + "/status"
+        body = serde.encode(obj)
+        out = self._request(
+            "PATCH", path, body, content_type="application/merge-patch+json"
+        )
+        return serde.decode(kind, out)
+
+    def delete(self, kind: str, namespace: str, name: str) -> None:
+        path = serde.resource_path(kind, namespace, name)
+        try:
+            self._request("DELETE", path)
+        except NotFoundError:
+            raise NotFoundError(kind, namespace, name) from None
+
+    def scale(self, kind: str, namespace: str, name: str, replicas: int) -> Any:
+        """Scale subresource write (DirectActuator path,
+        direct_actuator.go:78-104)."""
+        path = serde.resource_path(kind, namespace, name) + "/scale"
+        body = {
+            "apiVersion": "autoscaling/v1",
+            "kind": "Scale",
+            "metadata": {"name": name, "namespace": namespace},
+            "spec": {"replicas": int(replicas)},
+        }
+        self._request(
+            "PATCH", path, body, content_type="application/merge-patch+json"
+        )
+        return self.get(kind, namespace, name)
+
+    def record_event(
+        self, obj: Any, event_type: str, reason: str, message: str
+    ) -> None:
+        from ..api.types import utcnow, rfc3339
+
+        meta = obj.metadata
+        body = {
+            "apiVersion": "v1",
+            "kind": "Event",
+            "metadata": {
+                "generateName": f"{meta.name}.",
+                "namespace": meta.namespace or "default",
+            },
+            "involvedObject": {
+                "kind": obj.kind,
+                "name": meta.name,
+                "namespace": meta.namespace,
+                "uid": meta.uid,
+            },
+            "type": event_type,
+            "reason": reason,
+            "message": message,
+            "firstTimestamp": rfc3339(utcnow()),
+            "lastTimestamp": rfc3339(utcnow()),
+            "count": 1,
+        }
+        ns = meta.namespace or "default"
+        try:
+            self._request("POST", f"/api/v1/namespaces/{ns}/events", body)
+        except (ApiError, NotFoundError) as e:  # events are best-effort
+            log.debug("event record failed: %s", e)
+
+    # --- watches ---
+
+    def watch(self, kinds: Optional[List[str]] = None) -> "queue.Queue[WatchEvent]":
+        q: "queue.Queue[WatchEvent]" = queue.Queue()
+        kinds = kinds or list(serde.SERDE.keys())
+        pumps = [_WatchPump(self, kind, q) for kind in kinds]
+        with self._lock:
+            self._watches.extend(pumps)
+        for p in pumps:
+            p.start()
+        q._wva_pumps = pumps  # type: ignore[attr-defined]
+        return q
+
+    def stop_watch(self, q: "queue.Queue[WatchEvent]") -> None:
+        pumps = getattr(q, "_wva_pumps", [])
+        for p in pumps:
+            p.stop()
+        with self._lock:
+            self._watches = [p for p in self._watches if p not in pumps]
+
+    def close(self) -> None:
+        with self._lock:
+            pumps, self._watches = self._watches, []
+        for p in pumps:
+            p.stop()
+
+
+class _WatchPump(threading.Thread):
+    """One kind's watch stream → WatchEvent queue, with reconnect."""
+
+    def __init__(self, cluster: RestCluster, kind: str, out: "queue.Queue[WatchEvent]"):
+        super().__init__(daemon=True, name=f"watch-{kind.lower()}")
+        self.cluster = cluster
+        self.kind = kind
+        self.out = out
+        self._stop = threading.Event()
+        self._resource_version: Optional[str] = None
+
+    def stop(self) -> None:
+        self._stop.set()
+
+    def run(self) -> None:
+        # initial LIST to seed resourceVersion and emit ADDED for existing
+        # objects (informer semantics: downstream reconcilers are
+        # level-triggered, so synthetic ADDEDs are correct)
+        try:
+            path = serde.resource_path(self.kind, None)
+            out = self.cluster._request("GET", path)
+            self._resource_version = (out.get("metadata") or {}).get(
+                "resourceVersion"
+            )
+            for item in out.get("items", []):
+                self.out.put(
+                    WatchEvent(ADDED, self.kind, serde.decode(self.kind, item))
+                )
+        except Exception as e:  # noqa: BLE001 — keep pumping
+            log.warning("watch list %s failed: %s", self.kind, e)
+
+        while not self._stop.is_set():
+            try:
+                self._stream_once()
+            except Exception as e:  # noqa: BLE001
+                if self._stop.is_set():
+                    return
+                log.debug("watch %s reconnect after: %s", self.kind, e)
+                self._stop.wait(1.0)
+
+    def _stream_once(self) -> None:
+        path = serde.resource_path(self.kind, None) + "?watch=true"
+        if self._resource_version:
+            path += f"&resourceVersion={self._resource_version}"
+        path += "&allowWatchBookmarks=true"
+        url = self.cluster.base_url + path
+        req = urllib.request.Request(url, headers=self.cluster._headers())
+        with urllib.request.urlopen(
+            req, timeout=self.cluster.watch_timeout, context=self.cluster._ssl
+        ) as resp:
+            while not self._stop.is_set():
+                line = resp.readline()
+                if not line:
+                    return  # server closed; reconnect
+                line = line.strip()
+                if not line:
+                    continue
+                evt = json.loads(line)
+                etype = evt.get("type", "")
+                obj_d = evt.get("object") or {}
+                rv = (obj_d.get("metadata") or {}).get("resourceVersion")
+                if rv:
+                    self._resource_version = rv
+                if etype == "BOOKMARK":
+                    continue
+                if etype not in (ADDED, MODIFIED, DELETED):
+                    continue
+                self.out.put(
+                    WatchEvent(etype, self.kind, serde.decode(self.kind, obj_d))
+                )
