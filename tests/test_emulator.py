@@ -218,3 +218,21 @@ class TestITLTableProfile:
         p = ServiceProfile.from_itl_table([32, 1, 8], [47.15, 23.86, 42.56])
         assert p.itl_ms(1) == pytest.approx(23.86)
         assert p.itl_ms(32) == pytest.approx(47.15)
+
+
+class TestMoEDensePath:
+    def test_dense_matches_sparse(self):
+        import torch
+        from wva_amd.calibration.moe_model import TINY_MOE, MixtralDecodeModel
+
+        m = MixtralDecodeModel(TINY_MOE, max_batch=8, max_seq=32, device="cpu")
+        h2 = torch.randn(8, TINY_MOE.hidden_size, dtype=torch.bfloat16)
+        layer = m.layers[0]
+        logits = h2.float() @ layer.w_router.t().float()
+        w, sel = torch.topk(logits, TINY_MOE.top_k, dim=-1)
+        w = torch.softmax(w, dim=-1).to(h2.dtype)
+        dense = m._moe_mlp_dense(layer, h2, w, sel)
+        sparse = m._moe_mlp_sparse(layer, h2, w, sel)
+        torch.testing.assert_close(
+            dense.float(), sparse.float(), atol=3e-2, rtol=3e-2
+        )
