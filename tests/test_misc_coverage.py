@@ -225,3 +225,118 @@ class TestProbeServer:
         assert _parse_bind(":8081") == ("0.0.0.0", 8081)
         assert _parse_bind("127.0.0.1:9090") == ("127.0.0.1", 9090)
         assert _parse_bind("8443") == ("0.0.0.0", 8443)
+
+
+class TestSerdeRoundTrip:
+    """encode→decode must be lossless for every kind the controller
+    touches (kube/serde.py)."""
+
+    def _roundtrip(self, obj):
+        from wva_amd.kube import serde
+
+        kind = obj.kind
+        d = serde.encode(obj)
+        back = serde.decode(kind, d)
+        assert serde.encode(back) == d
+        return back
+
+    def test_deployment(self):
+        from wva_amd.api.types import ObjectMeta
+        from wva_amd.kube.objects import (
+            Container, Deployment, EnvVar, PodTemplateSpec,
+        )
+
+        d = Deployment(
+            metadata=ObjectMeta(name="d", namespace="ns",
+                                labels={"a": "b"}, generation=4),
+            replicas=3,
+            selector={"app": "d"},
+            template=PodTemplateSpec(
+                labels={"app": "d"},
+                annotations={"x": "y"},
+                containers=[Container(
+                    name="vllm", image="img", command=["sh", "-c"],
+                    args=["--tensor-parallel-size", "8"],
+                    env=[EnvVar("VLLM_USE_V1", "1")],
+                    requests={"amd.com/gpu": "8"},
+                    limits={"amd.com/gpu": "8"},
+                )],
+                init_containers=[Container(name="init")],
+                node_selector={"amd.com/gpu.product": "MI355X"},
+            ),
+        )
+        d.status.replicas = 3
+        d.status.ready_replicas = 2
+        back = self._roundtrip(d)
+        assert back.template.containers[0].env[0].name == "VLLM_USE_V1"
+        assert back.status.ready_replicas == 2
+
+    def test_pod_node_configmap_service(self):
+        from wva_amd.api.types import ObjectMeta
+        from wva_amd.kube.objects import (
+            ConfigMap, Container, Node, Pod, PodStatus, Service, ServicePort,
+        )
+
+        self._roundtrip(Pod(
+            metadata=ObjectMeta(name="p", namespace="ns",
+                                owner_references=[{"kind": "ReplicaSet",
+                                                   "name": "rs-1"}]),
+            containers=[Container()],
+            node_name="n0",
+            status=PodStatus(phase="Pending", ready=False, pod_ip="10.1.2.3"),
+        ))
+        self._roundtrip(Node(
+            metadata=ObjectMeta(name="n0", labels={"k": "v"}),
+            allocatable={"amd.com/gpu": "8"}, capacity={"amd.com/gpu": "8"},
+        ))
+        self._roundtrip(ConfigMap(
+            metadata=ObjectMeta(name="c", namespace="ns"), data={"k": "v"},
+        ))
+        self._roundtrip(Service(
+            metadata=ObjectMeta(name="s", namespace="ns"),
+            selector={"app": "x"},
+            ports=[ServicePort(name="metrics", port=9090, target_port=9090)],
+        ))
+
+    def test_inferencepool_v1alpha2_bare_selector(self):
+        from wva_amd.kube import serde
+
+        # v1alpha2 used a bare label map instead of matchLabels
+        pool = serde.decode("InferencePool", {
+            "apiVersion": "inference.networking.x-k8s.io/v1alpha2",
+            "kind": "InferencePool",
+            "metadata": {"name": "p", "namespace": "ns"},
+            "spec": {
+                "selector": {"app": "vllm"},
+                "targetPortNumber": 8000,
+                "extensionRef": {"name": "epp"},
+            },
+        })
+        assert pool.selector == {"app": "vllm"}
+        assert pool.epp_service_name == "epp"
+
+    def test_lease_and_va(self):
+        from wva_amd.api.types import (
+            CrossVersionObjectReference, ObjectMeta, VariantAutoscaling,
+            VariantAutoscalingSpec, utcnow,
+        )
+        from wva_amd.kube.objects import Lease
+
+        lease = Lease(
+            metadata=ObjectMeta(name="wva-lock", namespace="wva-system"),
+            holder_identity="pod-a",
+            lease_duration_seconds=60,
+            acquire_time=utcnow(), renew_time=utcnow(),
+        )
+        back = self._roundtrip(lease)
+        assert back.holder_identity == "pod-a"
+
+        va = VariantAutoscaling(
+            metadata=ObjectMeta(name="v", namespace="ns"),
+            spec=VariantAutoscalingSpec(
+                scale_target_ref=CrossVersionObjectReference(name="v"),
+                model_id="m", variant_cost="25.5",
+            ),
+        )
+        back = self._roundtrip(va)
+        assert back.spec.variant_cost == "25.5"
