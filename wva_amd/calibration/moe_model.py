@@ -107,6 +107,15 @@ class _MoELayer:
         )
         self.w_gate_up = list(self.w_gate_up_stacked)
         self.w_down = list(self.w_down_stacked)
+        # dense-path operands (no transposed VIEWS at run time — hipBLASLt
+        # takes a catastrophically slow route for strided-batched GEMMs on
+        # transposed views of these shapes; measured 11 ms/layer):
+        #  * gate_up runs as ONE flat GEMM against [E·2I, H] (a free view)
+        #  * down runs as bmm against a pre-transposed CONTIGUOUS [E, I, H]
+        self.w_gate_up_flat = self.w_gate_up_stacked.reshape(
+            cfg.num_experts * 2 * cfg.intermediate_size, h
+        )
+        self.w_down_t = self.w_down_stacked.transpose(1, 2).contiguous()
 
 
 class MixtralDecodeModel:
@@ -194,19 +203,16 @@ class MixtralDecodeModel:
     def _moe_mlp_dense(self, layer, h2, weights, selected):
         cfg = self.cfg
         B = h2.shape[0]
-        w_full = torch.zeros(
-            B, cfg.num_experts, device=h2.device, dtype=h2.dtype
-        )
+        E, I = cfg.num_experts, cfg.intermediate_size
+        w_full = torch.zeros(B, E, device=h2.device, dtype=h2.dtype)
         w_full.scatter_(1, selected, weights)
-        E = cfg.num_experts
-        h_rep = h2.unsqueeze(0).expand(E, B, cfg.hidden_size)
-        gate_up = torch.bmm(
-            h_rep, layer.w_gate_up_stacked.transpose(1, 2)
-        )  # [E, B, 2I]
+        # all experts' gate_up as one plain GEMM: [B, H] @ [H, E*2I]
+        gate_up = ops.linear(h2, layer.w_gate_up_flat)  # [B, E*2I]
         act = ops.silu_mul_fused(
-            gate_up.reshape(E * B, 2 * cfg.intermediate_size)
-        ).reshape(E, B, cfg.intermediate_size)
-        y = torch.bmm(act, layer.w_down_stacked.transpose(1, 2))
+            gate_up.reshape(B * E, 2 * I)
+        ).reshape(B, E, I)
+        # down per expert: bmm over contiguous pre-transposed weights
+        y = torch.bmm(act.transpose(0, 1).contiguous(), layer.w_down_t)
         return torch.einsum("ebh,be->bh", y, w_full)
 
     def _moe_mlp_sparse(self, layer, h2, weights, selected):
