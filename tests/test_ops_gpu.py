@@ -193,8 +193,9 @@ class TestMoEDecodeModel:
             for attr in ("input_norm", "post_attn_norm", "wqkv", "wo",
                          "w_router"):
                 setattr(lc, attr, getattr(lg, attr).cpu())
-            lc.w_gate_up = [w.cpu() for w in lg.w_gate_up]
-            lc.w_down = [w.cpu() for w in lg.w_down]
+            lc.set_expert_weights(
+                lg.w_gate_up_stacked.cpu(), lg.w_down_stacked.cpu()
+            )
         gpu.context_lens[:2] = 0
         cpu.context_lens[:2] = 0
         tokens = torch.randint(0, TINY_MOE.vocab_size, (2,))

@@ -117,6 +117,20 @@ class _MoELayer:
         )
         self.w_down_t = self.w_down_stacked.transpose(1, 2).contiguous()
 
+    def set_expert_weights(self, gate_up_stacked, down_stacked) -> None:
+        """Replace expert weights, recomputing every derived operand
+        (tests copy weights between devices through this)."""
+        self.w_gate_up_stacked = gate_up_stacked
+        self.w_down_stacked = down_stacked
+        self.w_gate_up = list(gate_up_stacked)
+        self.w_down = list(down_stacked)
+        e2i, h = (
+            gate_up_stacked.shape[0] * gate_up_stacked.shape[1],
+            gate_up_stacked.shape[2],
+        )
+        self.w_gate_up_flat = gate_up_stacked.reshape(e2i, h)
+        self.w_down_t = down_stacked.transpose(1, 2).contiguous()
+
 
 class MixtralDecodeModel:
     """Decode-only MoE engine; same KV-cache layout as LlamaDecodeModel."""
