@@ -256,3 +256,53 @@ def skinny_gemm_bench():
             print(f"{label:12s} [{name}] M={M:3d} N={N:6d} K={K:5d}: "
                   f"{us:7.1f} us  W {wbytes/1e6:6.1f} MB  "
                   f"{wbytes/us/1e3:5.2f} TB/s")
+
+
+def moe_bench():
+    """Per-op timing of the dense MoE layer path (Mixtral shapes)."""
+    import time as _t
+
+    import torch
+
+    from wva_amd import ops
+
+    E, I, H, B = 8, 14336, 4096, 64
+    dt = torch.bfloat16
+    dev = "cuda"
+    h2 = torch.randn(B, H, device=dev, dtype=dt)
+    w_flat = torch.randn(E * 2 * I, H, device=dev, dtype=dt)
+    w_down_t = torch.randn(E, I, H, device=dev, dtype=dt)
+    act_e = torch.randn(E, B, I, device=dev, dtype=dt)
+    w_full = torch.rand(B, E, device=dev, dtype=dt)
+    y = torch.randn(E, B, H, device=dev, dtype=dt)
+    gate_up = torch.randn(B, E * 2 * I, device=dev, dtype=dt)
+
+    def bench(name, fn, n=10):
+        for _ in range(3):
+            fn()
+        torch.cuda.synchronize()
+        t0 = _t.perf_counter()
+        for _ in range(n):
+            fn()
+        torch.cuda.synchronize()
+        print(f"{name:28s} {(_t.perf_counter()-t0)/n*1e3:9.3f} ms")
+
+    bench("gate_up flat linear", lambda: h2 @ w_flat.t())
+    bench("silu_mul_fused", lambda: ops.silu_mul_fused(
+        gate_up.reshape(B * E, 2 * I)))
+    bench("down bmm", lambda: torch.bmm(act_e, w_down_t))
+    bench("combine einsum", lambda: torch.einsum("ebh,be->bh", y, w_full))
+    bench("act transpose+contig", lambda: act_e.transpose(0, 1).reshape(
+        B, E, I).transpose(0, 1).contiguous())
+    # the full layer path
+    from wva_amd.calibration.moe_model import MIXTRAL_8X7B, _MoELayer
+    gen = torch.Generator(device=dev); gen.manual_seed(0)
+    layer = _MoELayer(MIXTRAL_8X7B, torch.device(dev), dt, gen)
+    from wva_amd.calibration.moe_model import MixtralDecodeModel
+    m = MixtralDecodeModel.__new__(MixtralDecodeModel)
+    m.cfg = MIXTRAL_8X7B
+    logits = h2.float() @ layer.w_router.t().float()
+    w, sel = torch.topk(logits, 2, dim=-1)
+    w = torch.softmax(w, dim=-1).to(dt)
+    bench("full _moe_mlp_dense", lambda: m._moe_mlp_dense(layer, h2, w, sel))
+    bench("full _moe_mlp_sparse", lambda: m._moe_mlp_sparse(layer, h2, w, sel), n=3)

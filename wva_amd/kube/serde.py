@@ -346,6 +346,24 @@ def servicemonitor_from_dict(d: Dict[str, Any]) -> ServiceMonitor:
 
 
 # --- Lease ---
+# The LeaderElector works in epoch-float seconds (lease arithmetic);
+# the wire format is RFC3339 MicroTime. Encode accepts either; decode
+# returns epoch floats.
+
+def _lease_time_to(v: Any) -> Optional[str]:
+    if v is None:
+        return None
+    if isinstance(v, (int, float)):
+        from datetime import datetime, timezone
+
+        return rfc3339(datetime.fromtimestamp(float(v), timezone.utc))
+    return rfc3339(v)
+
+
+def _lease_time_from(s: Optional[str]) -> Optional[float]:
+    dt = parse_rfc3339(s)
+    return dt.timestamp() if dt is not None else None
+
 
 def lease_to_dict(o: Lease) -> Dict[str, Any]:
     return {
@@ -355,8 +373,8 @@ def lease_to_dict(o: Lease) -> Dict[str, Any]:
         "spec": {
             "holderIdentity": o.holder_identity,
             "leaseDurationSeconds": int(o.lease_duration_seconds),
-            "acquireTime": rfc3339(o.acquire_time),
-            "renewTime": rfc3339(o.renew_time),
+            "acquireTime": _lease_time_to(o.acquire_time),
+            "renewTime": _lease_time_to(o.renew_time),
         },
     }
 
@@ -367,8 +385,8 @@ def lease_from_dict(d: Dict[str, Any]) -> Lease:
         metadata=_meta_from(d.get("metadata") or {}),
         holder_identity=spec.get("holderIdentity", ""),
         lease_duration_seconds=float(spec.get("leaseDurationSeconds", 60) or 60),
-        acquire_time=parse_rfc3339(spec.get("acquireTime")),
-        renew_time=parse_rfc3339(spec.get("renewTime")),
+        acquire_time=_lease_time_from(spec.get("acquireTime")),
+        renew_time=_lease_time_from(spec.get("renewTime")),
     )
 
 
