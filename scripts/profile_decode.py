@@ -292,6 +292,7 @@ def moe_bench():
         gate_up.reshape(B * E, 2 * I)))
     bench("down bmm", lambda: torch.bmm(act_e, w_down_t))
     bench("combine einsum", lambda: torch.einsum("ebh,be->bh", y, w_full))
+    bench("combine mul+sum", lambda: (y * w_full.t().unsqueeze(-1)).sum(dim=0))
     bench("act transpose+contig", lambda: act_e.transpose(0, 1).reshape(
         B, E, I).transpose(0, 1).contiguous())
     # the full layer path
