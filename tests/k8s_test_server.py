@@ -81,22 +81,31 @@ class K8sTestServer:
                         return self._send(404, {"message": "not found"})
                     return self._send(200, serde.encode(obj))
                 if qs.get("watch", ["false"])[0] == "true":
-                    return self._watch(kind)
+                    rv = int(qs.get("resourceVersion", ["-1"])[0] or -1)
+                    return self._watch(kind, rv)
                 sel = None
                 if "labelSelector" in qs:
                     sel = dict(
                         kv.split("=", 1) for kv in qs["labelSelector"][0].split(",")
                     )
-                objs = outer.cluster.list(kind, namespace=ns, label_selector=sel)
+                objs, snap_rv = outer.cluster.snapshot(
+                    kind, namespace=ns, label_selector=sel
+                )
                 return self._send(200, {
                     "apiVersion": "v1",
                     "kind": f"{kind}List",
-                    "metadata": {"resourceVersion": "1"},
+                    "metadata": {"resourceVersion": str(snap_rv)},
                     "items": [serde.encode(o) for o in objs],
                 })
 
-            def _watch(self, kind: str):
-                q = outer.cluster.watch([kind])
+            def _watch(self, kind: str, rv: int = -1):
+                # rv >= 0: replay from the event log (atomic with the
+                # subscription — the race a real API server closes with
+                # its watch cache); rv < 0: live-only
+                if rv >= 0:
+                    q = outer.cluster.watch_since([kind], rv)
+                else:
+                    q = outer.cluster.watch([kind])
                 self.send_response(200)
                 self.send_header("Content-Type", "application/json")
                 self.send_header("Transfer-Encoding", "chunked")

@@ -313,3 +313,27 @@ class TestLeaderElectionOverRest:
         finally:
             a.close()
             b.close()
+
+
+class TestWatchSince:
+    def test_no_event_lost_between_list_and_watch(self, server):
+        """The LIST→WATCH gap: mutations between a snapshot and the watch
+        subscription must be replayed via resourceVersion (FakeCluster
+        watch_since; the API-server watch-cache contract). Without it,
+        events racing the stream establishment were silently lost."""
+        from wva_amd.kube.fake import ADDED
+
+        backing = server.cluster
+        backing.create(make_deployment("a"))
+        objs, rv = backing.snapshot("Deployment")
+        assert len(objs) == 1
+        # mutation AFTER snapshot, BEFORE subscription
+        backing.create(make_deployment("raced"))
+        q = backing.watch_since(["Deployment"], rv)
+        evt = q.get(timeout=2)
+        assert evt.type == ADDED and evt.obj.name == "raced"
+        # live events still flow
+        backing.create(make_deployment("live"))
+        evt = q.get(timeout=2)
+        assert evt.obj.name == "live"
+        backing.stop_watch(q)

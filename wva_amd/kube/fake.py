@@ -85,6 +85,11 @@ class FakeCluster:
         self._rv = 0
         self._watchers: List[Tuple[Optional[set], "queue.Queue[WatchEvent]"]] = []
         self.events: List[Event] = []
+        # bounded event log for resourceVersion-resumable watches (the
+        # API-server watch-cache analog): (rv, event), rv = the global
+        # monotonic resourceVersion at mutation time
+        self._event_log: List[Tuple[int, WatchEvent]] = []
+        self._event_log_cap = 4096
 
     # --- internals ---
 
@@ -94,9 +99,13 @@ class FakeCluster:
 
     def _notify(self, event_type: str, obj: Any) -> None:
         kind = _kind_of(obj)
+        evt = WatchEvent(type=event_type, kind=kind, obj=_clone(obj))
+        self._event_log.append((self._rv, evt))
+        if len(self._event_log) > self._event_log_cap:
+            del self._event_log[: len(self._event_log) // 2]
         for kinds, q in list(self._watchers):
             if kinds is None or kind in kinds:
-                q.put(WatchEvent(type=event_type, kind=kind, obj=_clone(obj)))
+                q.put(evt)
 
     # --- CRUD ---
 
@@ -230,6 +239,37 @@ class FakeCluster:
         with self._lock:
             self._watchers.append((set(kinds) if kinds else None, q))
         return q
+
+    def watch_since(
+        self, kinds: Optional[List[str]], resource_version: int
+    ) -> "queue.Queue[WatchEvent]":
+        """Watch resuming from a resourceVersion: events with rv >
+        resource_version are replayed from the log, then live — replay +
+        subscription are atomic, so nothing is lost in between (the
+        API-server watch-cache contract the REST watch path needs)."""
+        q: "queue.Queue[WatchEvent]" = queue.Queue()
+        ks = set(kinds) if kinds else None
+        with self._lock:
+            for rv, evt in self._event_log:
+                if rv > resource_version and (ks is None or evt.kind in ks):
+                    q.put(evt)
+            self._watchers.append((ks, q))
+        return q
+
+    def snapshot(
+        self,
+        kind: str,
+        namespace: Optional[str] = None,
+        label_selector: Optional[Dict[str, str]] = None,
+    ) -> Tuple[List[Any], int]:
+        """list() + the resourceVersion of the snapshot, atomically —
+        a LIST+WATCH(resourceVersion=rv) pair over these two calls
+        observes every mutation exactly once."""
+        with self._lock:
+            return (
+                self.list(kind, namespace=namespace, label_selector=label_selector),
+                self._rv,
+            )
 
     def stop_watch(self, q: "queue.Queue[WatchEvent]") -> None:
         with self._lock:
