@@ -109,3 +109,50 @@ class TestMoEReference:
         # Mixtral-8x7B ≈ 47B params ≈ 87-94 GiB bf16: resident on 288 GB
         # MI355X, NOT on 192 GB MI300X together with a useful KV budget
         assert 80 < gb < 100
+
+
+class TestPrefill:
+    def test_prefill_matches_stepwise_decode(self):
+        """prefill(S tokens) must equal decoding the same tokens one by
+        one from an empty cache — same math, different schedule."""
+        import torch
+
+        from wva_amd.calibration.model import TINY, LlamaDecodeModel
+
+        torch.manual_seed(3)
+        m1 = LlamaDecodeModel(TINY, max_batch=2, max_seq=32, device="cpu",
+                              seed=9)
+        m2 = LlamaDecodeModel(TINY, max_batch=2, max_seq=32, device="cpu",
+                              seed=9)
+        S = 7
+        tokens = torch.randint(0, TINY.vocab_size, (2, S))
+
+        logits_pre = m1.prefill(tokens)
+
+        m2.context_lens.zero_()
+        for s in range(S):
+            logits_step = m2.decode_step(tokens[:, s])
+
+        torch.testing.assert_close(
+            logits_pre.float(), logits_step.float(), atol=5e-2, rtol=5e-2
+        )
+        assert int(m1.context_lens[0]) == S
+        # KV caches agree too
+        torch.testing.assert_close(
+            m1.k_cache[0][:2, :, :S].float(),
+            m2.k_cache[0][:2, :, :S].float(), atol=5e-2, rtol=5e-2,
+        )
+
+    def test_prefill_then_decode_continues(self):
+        import torch
+
+        from wva_amd.calibration.model import TINY, LlamaDecodeModel
+
+        m = LlamaDecodeModel(TINY, max_batch=1, max_seq=32, device="cpu")
+        tokens = torch.randint(0, TINY.vocab_size, (1, 5))
+        m.prefill(tokens)
+        nxt = torch.randint(0, TINY.vocab_size, (1,))
+        logits = m.decode_step(nxt)
+        assert logits.shape == (1, TINY.vocab_size)
+        assert int(m.context_lens[0]) == 6
+        assert torch.isfinite(logits.float()).all()

@@ -37,6 +37,7 @@ class CalibrationResult:
     num_gpu_blocks: int
     block_size: int
     decode_tokens_per_s_peak: float
+    prefill_tokens_per_s: float = 0.0
 
     def to_json(self) -> str:
         return json.dumps(asdict(self), indent=2)
@@ -100,6 +101,29 @@ def measure_itl(
     return times[len(times) // 2]
 
 
+def measure_prefill_tps(
+    model: LlamaDecodeModel,
+    batch: int = 2,
+    seq: int = 1024,
+    iters: int = 3,
+) -> float:
+    """Prefill throughput (prompt tokens/s) — the TTFT model's input
+    (ServiceProfile.prefill_tokens_per_s), measured instead of assumed."""
+    seq = min(seq, model.max_seq)
+    batch = min(batch, model.max_batch)
+    tokens = torch.randint(
+        0, model.cfg.vocab_size, (batch, seq), device=model.device
+    )
+    model.prefill(tokens)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        model.prefill(tokens)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    return batch * seq * iters / dt
+
+
 def derive_kv_capacity(
     cfg: LlamaConfig,
     gpu_memory_utilization: float = 0.9,
@@ -132,6 +156,7 @@ def calibrate_service_profile(
         batch_sizes = [1, 2, 4, 8, 16, 32, 64]
     max_batch = max(batch_sizes)
     model = LlamaDecodeModel(cfg, max_batch=max_batch, max_seq=max_seq)
+    prefill_tps = measure_prefill_tps(model)
     itl: List[float] = []
     for b in batch_sizes:
         itl.append(
@@ -155,6 +180,7 @@ def calibrate_service_profile(
         num_gpu_blocks=blocks,
         block_size=16,
         decode_tokens_per_s_peak=peak_tps,
+        prefill_tokens_per_s=prefill_tps,
     )
     profile = ServiceProfile(
         alpha_ms=max(alpha, 0.1),
@@ -162,6 +188,7 @@ def calibrate_service_profile(
         max_num_seqs=256,
         num_gpu_blocks=blocks,
         block_size=16,
+        prefill_tokens_per_s=prefill_tps,
     )
     # free the model before returning (bench reuses the GPU)
     del model

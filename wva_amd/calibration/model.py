@@ -205,3 +205,98 @@ class LlamaDecodeModel:
         logits = ops.linear(final, self.lm_head)
         self.context_lens[:B] += 1
         return logits
+
+    @torch.no_grad()
+    def prefill(self, token_ids: torch.Tensor) -> torch.Tensor:
+        """Prefill `token_ids [B, S]`: computes all S positions in one
+        pass (causal attention), fills the KV cache, sets context_lens=S
+        and returns last-position logits [B, vocab].
+
+        Grounds the ServiceProfile's prefill_tokens_per_s (the TTFT
+        model's input) with a measured value instead of a constant.
+        Attention here is explicit matmul attention on hipBLASLt —
+        prefill is GEMM-shaped compute (S×S scores), exactly what the
+        library is for; the hand-written kernels cover the memory-bound
+        decode path. RMSNorm/RoPE/SwiGLU reuse the HIP kernels (they are
+        row-shaped and batch-size-agnostic).
+        """
+        cfg = self.cfg
+        B, S = token_ids.shape
+        T = B * S
+        if S > self.max_seq:
+            raise ValueError(f"S={S} exceeds max_seq={self.max_seq}")
+        positions = (
+            torch.arange(S, device=self.device, dtype=torch.int32)
+            .repeat(B)
+        )  # [T] per-token position
+
+        x = self.embed.index_select(0, token_ids.reshape(-1))  # [T, H]
+        residual: Optional[torch.Tensor] = None
+
+        causal = torch.full(
+            (S, S), float("-inf"), device=self.device, dtype=torch.float32
+        ).triu(1)
+
+        for li, layer in enumerate(self.layers):
+            if residual is None:
+                residual = x.clone()
+                h = ops.rmsnorm(x, layer.input_norm, None, cfg.rms_eps)
+            else:
+                h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
+
+            qkv = h @ layer.wqkv.t()  # [T, (Hq+2Hk)·D] — prefill M is large
+            q, k, v = qkv.split(
+                [cfg.q_size, cfg.kv_size, cfg.kv_size], dim=-1
+            )
+            q = q.reshape(T, cfg.num_q_heads, cfg.head_dim).contiguous()
+            k = k.reshape(T, cfg.num_kv_heads, cfg.head_dim).contiguous()
+            ops.rope(q, k, positions, cfg.rope_theta)  # in-place HIP kernel
+            v = v.reshape(T, cfg.num_kv_heads, cfg.head_dim)
+
+            # append to the head-major cache [B, Hk, S_max, D]
+            k_b = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
+            v_b = v.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
+            self.k_cache[li][:B, :, :S].copy_(k_b.permute(0, 2, 1, 3))
+            self.v_cache[li][:B, :, :S].copy_(v_b.permute(0, 2, 1, 3))
+
+            # causal GQA attention (explicit matmul form)
+            G = cfg.num_q_heads // cfg.num_kv_heads
+            qh = (
+                q.reshape(B, S, cfg.num_kv_heads, G, cfg.head_dim)
+                .permute(0, 2, 3, 1, 4)
+                .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+            )
+            kh = (
+                self.k_cache[li][:B, :, :S]
+                .unsqueeze(2).expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
+                .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+            )
+            vh = (
+                self.v_cache[li][:B, :, :S]
+                .unsqueeze(2).expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
+                .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+            )
+            scores = (
+                torch.bmm(qh, kh.transpose(1, 2)).float() * self.scale
+                + causal
+            )
+            p = torch.softmax(scores, dim=-1).to(self.dtype)
+            attn = torch.bmm(p, vh)  # [B*Hq, S, D]
+            attn = (
+                attn.reshape(B, cfg.num_q_heads, S, cfg.head_dim)
+                .permute(0, 2, 1, 3)
+                .reshape(T, cfg.q_size)
+            ).contiguous()
+            x = attn @ layer.wo.t()
+
+            h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
+            gate_up = h2 @ layer.w_gate_up.t()
+            act = ops.silu_mul_fused(gate_up)
+            x = act @ layer.w_down.t()
+
+        final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)
+        last = final.reshape(B, S, cfg.hidden_size)[:, -1].contiguous()
+        logits = last @ self.lm_head.t()
+        self.context_lens.zero_()
+        self.context_lens[:B] = S
+        return logits
