@@ -370,3 +370,28 @@ class TestAttnV5:
         # looser tolerance than v4's fp32 PV accumulation
         torch.testing.assert_close(out.float().cpu(), ref, atol=4e-2,
                                    rtol=4e-2)
+
+
+class TestPrefillGpu:
+    def test_prefill_matches_stepwise_on_gpu(self, dev):
+        from wva_amd.calibration.model import TINY, LlamaDecodeModel
+
+        m1 = LlamaDecodeModel(TINY, max_batch=2, max_seq=32, seed=21)
+        m2 = LlamaDecodeModel(TINY, max_batch=2, max_seq=32, seed=21)
+        S = 9
+        tokens = torch.randint(0, TINY.vocab_size, (2, S), device=dev)
+        logits_pre = m1.prefill(tokens)
+        m2.context_lens.zero_()
+        for s in range(S):
+            logits_step = m2.decode_step(tokens[:, s])
+        torch.testing.assert_close(
+            logits_pre.float(), logits_step.float(), atol=8e-2, rtol=8e-2
+        )
+
+    def test_prefill_tps_measured(self, dev):
+        from wva_amd.calibration.itl_benchmark import measure_prefill_tps
+        from wva_amd.calibration.model import TINY, LlamaDecodeModel
+
+        m = LlamaDecodeModel(TINY, max_batch=2, max_seq=64)
+        tps = measure_prefill_tps(m, batch=2, seq=64, iters=2)
+        assert tps > 0
