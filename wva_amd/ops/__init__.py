@@ -271,6 +271,7 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         and x.shape[1] % 128 == 0
         and x.shape[1] <= 8192      # long-K loses to blaslt (down-proj)
         and w.shape[0] <= 8192      # big-N loses to blaslt (gate_up/head)
+        and not os.environ.get("WVA_DISABLE_SKINNY")  # A/B kill-switch
     ):
         return _require_ext().skinny_linear(x, w)
     return x @ w.t()
