@@ -258,20 +258,23 @@ def enable_tuned_gemms() -> bool:
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Decode linear y = x @ w.T ([M,K] @ [N,K]^T).
 
-    Dispatch is measurement-driven (profiles/skinny_gemm_ab.txt):
-    hipBLASLt is near-roofline on big-N decode shapes (gate_up, lm_head)
-    but under-fills the 256-CU chip on small-N ones (its tile grid is
-    N/MT workgroups); the in-tree weight-streaming MFMA kernel
-    (csrc/skinny_gemm.hip) wins there via split-K fill. Everything else
-    (prefill, big batches) stays on hipBLASLt via torch.matmul.
+    Dispatch is measurement-driven and currently ALWAYS hipBLASLt: the
+    in-tree weight-streaming MFMA kernel (csrc/skinny_gemm.hip, exposed
+    as ext.skinny_linear) beat hipBLASLt on isolated small-N/small-M
+    microbenches but loses IN CONTEXT inside the decode step, where the
+    activation matrix is not L2-hot between layers (decode b=1 measured
+    5.27 ms/step pure-blaslt vs 5.41 with the kernel dispatched on
+    qkv/o). Full history: profiles/skinny_gemm_ab.txt and
+    docs/mi355x-kernels.md (negative results). Set WVA_FORCE_SKINNY=1
+    to re-enable the measured window for experiments.
     """
     if (
-        x.is_cuda
+        os.environ.get("WVA_FORCE_SKINNY")
+        and x.is_cuda
         and x.shape[0] <= 16
         and x.shape[1] % 128 == 0
-        and x.shape[1] <= 8192      # long-K loses to blaslt (down-proj)
-        and w.shape[0] <= 8192      # big-N loses to blaslt (gate_up/head)
-        and not os.environ.get("WVA_DISABLE_SKINNY")  # A/B kill-switch
+        and x.shape[1] <= 8192
+        and w.shape[0] <= 8192
     ):
         return _require_ext().skinny_linear(x, w)
     return x @ w.t()
