@@ -395,3 +395,36 @@ class TestPrefillGpu:
         m = LlamaDecodeModel(TINY, max_batch=2, max_seq=64)
         tps = measure_prefill_tps(m, batch=2, seq=64, iters=2)
         assert tps > 0
+
+
+class TestPrefillAttnKernel:
+    @pytest.mark.parametrize("B,hq,hk,S", [
+        (2, 8, 2, 64), (1, 32, 8, 512), (2, 4, 4, 100), (1, 64, 8, 333),
+        (3, 8, 1, 17),
+    ])
+    def test_matches_naive_causal(self, dev, B, hq, hk, S):
+        from wva_amd import ops
+
+        torch.manual_seed(B * 100 + S)
+        S_max = max(S + 8, 64)
+        T = B * S
+        q = torch.randn(T, hq, 128, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(B, hk, S_max, 128, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(B, hk, S_max, 128, device=dev, dtype=torch.bfloat16)
+        scale = 128 ** -0.5
+        out = ops.prefill_attn(q, k, v, B, S, scale)
+
+        # naive fp32 causal reference
+        G = hq // hk
+        qf = q.float().reshape(B, S, hq, 128)
+        ref = torch.empty_like(qf)
+        causal = torch.full((S, S), float("-inf"), device=dev).triu(1)
+        for h in range(hq):
+            kh = k[:, h // G, :S].float()   # [B, S, 128]
+            vh = v[:, h // G, :S].float()
+            sc = torch.bmm(qf[:, :, h], kh.transpose(1, 2)) * scale + causal
+            p = torch.softmax(sc, dim=-1)
+            ref[:, :, h] = torch.bmm(p, vh)
+        torch.testing.assert_close(
+            out.float().reshape(B, S, hq, 128), ref, atol=4e-2, rtol=4e-2
+        )

@@ -259,34 +259,42 @@ class LlamaDecodeModel:
             self.k_cache[li][:B, :, :S].copy_(k_b.permute(0, 2, 1, 3))
             self.v_cache[li][:B, :, :S].copy_(v_b.permute(0, 2, 1, 3))
 
-            # causal GQA attention (explicit matmul form)
-            G = cfg.num_q_heads // cfg.num_kv_heads
-            qh = (
-                q.reshape(B, S, cfg.num_kv_heads, G, cfg.head_dim)
-                .permute(0, 2, 3, 1, 4)
-                .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
-            )
-            kh = (
-                self.k_cache[li][:B, :, :S]
-                .unsqueeze(2).expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
-                .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
-            )
-            vh = (
-                self.v_cache[li][:B, :, :S]
-                .unsqueeze(2).expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
-                .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
-            )
-            scores = (
-                torch.bmm(qh, kh.transpose(1, 2)).float() * self.scale
-                + causal
-            )
-            p = torch.softmax(scores, dim=-1).to(self.dtype)
-            attn = torch.bmm(p, vh)  # [B*Hq, S, D]
-            attn = (
-                attn.reshape(B, cfg.num_q_heads, S, cfg.head_dim)
-                .permute(0, 2, 1, 3)
-                .reshape(T, cfg.q_size)
-            ).contiguous()
+            if q.is_cuda:
+                # hand-written MFMA flash-prefill kernel (gfx950)
+                attn = ops.prefill_attn(
+                    q, self.k_cache[li], self.v_cache[li], B, S, self.scale
+                ).reshape(T, cfg.q_size)
+            else:
+                # CPU numerics reference: explicit causal matmul attention
+                G = cfg.num_q_heads // cfg.num_kv_heads
+                qh = (
+                    q.reshape(B, S, cfg.num_kv_heads, G, cfg.head_dim)
+                    .permute(0, 2, 3, 1, 4)
+                    .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+                )
+                kh = (
+                    self.k_cache[li][:B, :, :S]
+                    .unsqueeze(2)
+                    .expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
+                    .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+                )
+                vh = (
+                    self.v_cache[li][:B, :, :S]
+                    .unsqueeze(2)
+                    .expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
+                    .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+                )
+                scores = (
+                    torch.bmm(qh, kh.transpose(1, 2)).float() * self.scale
+                    + causal
+                )
+                p = torch.softmax(scores, dim=-1).to(self.dtype)
+                attn = torch.bmm(p, vh)  # [B*Hq, S, D]
+                attn = (
+                    attn.reshape(B, cfg.num_q_heads, S, cfg.head_dim)
+                    .permute(0, 2, 1, 3)
+                    .reshape(T, cfg.q_size)
+                ).contiguous()
             x = attn @ layer.wo.t()
 
             h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)

@@ -278,3 +278,19 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     ):
         return _require_ext().skinny_linear(x, w)
     return x @ w.t()
+
+
+def prefill_attn(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    batch: int,
+    seq: int,
+    scale: float,
+) -> torch.Tensor:
+    """Causal GQA flash-prefill over the (populated) head-major caches.
+
+    GPU-only (fail-loud): the CPU prefill reference lives in
+    calibration/model.py's matmul branch.
+    """
+    return _require_ext().prefill_attn(q, k_cache, v_cache, batch, seq, scale)

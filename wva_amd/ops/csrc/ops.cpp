@@ -28,6 +28,11 @@ extern "C" void launch_rope_append_kv(const void* qkv, void* q_out,
 extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
                                           int max_ctx_hint);
 extern "C" int skinny_gemm_num_splits(int N, int K, int nt);
+extern "C" void launch_prefill_attn(void* out, const void* q,
+                                    const void* k_cache, const void* v_cache,
+                                    int batch, int seq, int num_q_heads,
+                                    int num_kv_heads, int max_seq,
+                                    float scale, hipStream_t stream);
 extern "C" int skinny_gemm_tile_n(int M);
 extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
                                    const void* w, int M, int N, int K,
@@ -263,6 +268,30 @@ torch::Tensor skinny_linear(torch::Tensor x, torch::Tensor w) {
   return y;
 }
 
+// Causal flash-attention prefill over the (already-populated) head-major
+// KV caches (csrc/prefill_attention.hip). q/out: [B*S, Hq, 128].
+torch::Tensor prefill_attn(torch::Tensor q, torch::Tensor k_cache,
+                           torch::Tensor v_cache, int64_t batch, int64_t seq,
+                           double scale) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(q.dim() == 3 && q.size(2) == 128, "q must be [T, Hq, 128]");
+  TORCH_CHECK(q.size(0) == batch * seq, "T must equal B*S");
+  TORCH_CHECK(k_cache.dim() == 4, "k_cache must be [B, Hk, S_max, D]");
+  const int num_q_heads = q.size(1);
+  const int num_kv_heads = k_cache.size(1);
+  const int max_seq = k_cache.size(2);
+  TORCH_CHECK(seq <= max_seq, "seq exceeds cache size");
+  TORCH_CHECK(num_q_heads % num_kv_heads == 0, "Hq must divide by Hk");
+  TORCH_CHECK(num_q_heads / num_kv_heads <= 8, "GQA group size must be <= 8");
+  auto out = torch::empty_like(q);
+  launch_prefill_attn(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                      v_cache.data_ptr(), (int)batch, (int)seq, num_q_heads,
+                      num_kv_heads, max_seq, (float)scale, current_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -290,6 +319,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "GQA decode attention, MFMA-scores variant (= default)",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("context_lens"), py::arg("scale"));
+  m.def("prefill_attn", &prefill_attn,
+        "Causal flash-attention prefill over head-major KV caches",
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("batch"), py::arg("seq"), py::arg("scale"));
   m.def("skinny_linear", &skinny_linear,
         "Weight-streaming decode GEMM: x[M,K] @ w[N,K]^T, M <= 64",
         py::arg("x"), py::arg("w"));

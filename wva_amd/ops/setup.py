@@ -27,6 +27,7 @@ setup(
                 os.path.join(this_dir, "csrc", "silu_mul.hip"),
                 os.path.join(this_dir, "csrc", "attention.hip"),
                 os.path.join(this_dir, "csrc", "skinny_gemm.hip"),
+                os.path.join(this_dir, "csrc", "prefill_attention.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
