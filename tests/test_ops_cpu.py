@@ -156,3 +156,25 @@ class TestPrefill:
         assert logits.shape == (1, TINY.vocab_size)
         assert int(m.context_lens[0]) == 6
         assert torch.isfinite(logits.float()).all()
+
+
+class TestMoEPrefill:
+    def test_moe_prefill_matches_stepwise(self):
+        import torch
+
+        from wva_amd.calibration.moe_model import TINY_MOE, MixtralDecodeModel
+
+        torch.manual_seed(4)
+        m1 = MixtralDecodeModel(TINY_MOE, max_batch=2, max_seq=32,
+                                device="cpu", seed=5)
+        m2 = MixtralDecodeModel(TINY_MOE, max_batch=2, max_seq=32,
+                                device="cpu", seed=5)
+        S = 6
+        tokens = torch.randint(0, TINY_MOE.vocab_size, (2, S))
+        logits_pre = m1.prefill(tokens)
+        m2.context_lens.zero_()
+        for s in range(S):
+            logits_step = m2.decode_step(tokens[:, s])
+        torch.testing.assert_close(
+            logits_pre.float(), logits_step.float(), atol=6e-2, rtol=6e-2
+        )

@@ -283,3 +283,97 @@ class MixtralDecodeModel:
         logits = ops.linear(final, self.lm_head)
         self.context_lens[:B] += 1
         return logits
+
+    @torch.no_grad()
+    def prefill(self, token_ids: torch.Tensor) -> torch.Tensor:
+        """Prefill [B, S]: one causal pass (MFMA flash-prefill kernel on
+        GPU), fills the KV caches, returns last-position logits.
+
+        The expert MLP takes the SPARSE gather path here: prefill is
+        compute-bound (each expert sees ~T·k/E of T = B·S tokens, and
+        T is large), so dense-batching would genuinely 4× the MLP FLOPs —
+        the opposite trade from decode (see _moe_mlp). The per-expert
+        `nonzero()` syncs amortize over one large pass.
+        """
+        cfg = self.cfg
+        B, S = token_ids.shape
+        T = B * S
+        if S > self.max_seq:
+            raise ValueError(f"S={S} exceeds max_seq={self.max_seq}")
+        positions = (
+            torch.arange(S, device=self.device, dtype=torch.int32).repeat(B)
+        )
+
+        x = self.embed.index_select(0, token_ids.reshape(-1))
+        residual = None
+        causal = None
+        if not x.is_cuda:
+            causal = torch.full(
+                (S, S), float("-inf"), device=self.device, dtype=torch.float32
+            ).triu(1)
+
+        for li, layer in enumerate(self.layers):
+            if residual is None:
+                residual = x.clone()
+                h = ops.rmsnorm(x, layer.input_norm, None, cfg.rms_eps)
+            else:
+                h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
+
+            qkv = h @ layer.wqkv.t()
+            q, k, v = qkv.split([cfg.q_size, cfg.kv_size, cfg.kv_size], dim=-1)
+            q = q.reshape(T, cfg.num_q_heads, cfg.head_dim).contiguous()
+            k = k.reshape(T, cfg.num_kv_heads, cfg.head_dim).contiguous()
+            ops.rope(q, k, positions, cfg.rope_theta)
+            k_b = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
+            v_b = v.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
+            self.k_cache[li][:B, :, :S].copy_(k_b.permute(0, 2, 1, 3))
+            self.v_cache[li][:B, :, :S].copy_(v_b.permute(0, 2, 1, 3))
+
+            if q.is_cuda:
+                attn = ops.prefill_attn(
+                    q, self.k_cache[li], self.v_cache[li], B, S, self.scale
+                ).reshape(T, cfg.q_size)
+            else:
+                G = cfg.num_q_heads // cfg.num_kv_heads
+                qh = (
+                    q.reshape(B, S, cfg.num_kv_heads, G, cfg.head_dim)
+                    .permute(0, 2, 3, 1, 4)
+                    .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+                )
+                kh = (
+                    self.k_cache[li][:B, :, :S]
+                    .unsqueeze(2)
+                    .expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
+                    .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+                )
+                vh = (
+                    self.v_cache[li][:B, :, :S]
+                    .unsqueeze(2)
+                    .expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
+                    .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
+                )
+                scores = (
+                    torch.bmm(qh, kh.transpose(1, 2)).float() * self.scale
+                    + causal
+                )
+                p = torch.softmax(scores, dim=-1).to(self.dtype)
+                attn = torch.bmm(p, vh)
+                attn = (
+                    attn.reshape(B, cfg.num_q_heads, S, cfg.head_dim)
+                    .permute(0, 2, 1, 3)
+                    .reshape(T, cfg.q_size)
+                ).contiguous()
+            x = attn @ layer.wo.t()
+
+            h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
+            router_logits = h2.float() @ layer.w_router.t().float()
+            weights, selected = torch.topk(router_logits, cfg.top_k, dim=-1)
+            weights = torch.softmax(weights, dim=-1).to(h2.dtype)
+            x = self._moe_mlp_sparse(layer, h2, weights, selected)
+
+        final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)
+        last = final.reshape(B, S, cfg.hidden_size)[:, -1].contiguous()
+        logits = last @ self.lm_head.t()
+        self.context_lens.zero_()
+        self.context_lens[:B] = S
+        return logits
