@@ -124,9 +124,10 @@ class TestMeasuredMultiModel:
         for tick in range(10):
             for _ in range(60):
                 # Mixtral serves far fewer req/s per replica (measured
-                # ~2.7k tok/s vs 10k): load it proportionally
-                sim.generate_arrivals(m_llama, lambda t: 400.0, 0.25, 100, 50)
-                sim.generate_arrivals(m_mix, lambda t: 60.0, 0.25, 100, 50)
+                # ~2.7k vs 9.8k tok/s): load it proportionally; both
+                # loads exceed one replica's measured capacity
+                sim.generate_arrivals(m_llama, lambda t: 900.0, 0.25, 100, 50)
+                sim.generate_arrivals(m_mix, lambda t: 300.0, 0.25, 100, 50)
                 sim.advance(0.25)
             app.saturation_engine.optimize()
             app.va_reconciler.reconcile("default", "vllm-llama")
