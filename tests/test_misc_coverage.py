@@ -340,3 +340,62 @@ class TestSerdeRoundTrip:
         )
         back = self._roundtrip(va)
         assert back.spec.variant_cost == "25.5"
+
+
+class TestCertWatcher:
+    def test_tls_metrics_with_rotation(self, tmp_path):
+        import ssl
+        import subprocess
+        import urllib.request
+
+        from prometheus_client import CollectorRegistry
+        from wva_amd.runtime.http import CertWatcher, ProbeServer
+
+        cert = tmp_path / "tls.crt"
+        key = tmp_path / "tls.key"
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+             "-keyout", str(key), "-out", str(cert), "-days", "1",
+             "-subj", "/CN=localhost"],
+            check=True, capture_output=True,
+        )
+        watcher = CertWatcher(str(cert), str(key), poll_seconds=0.1)
+        srv = ProbeServer(
+            "127.0.0.1:0", healthz=lambda: True, readyz=lambda: True,
+            registry=CollectorRegistry(), cert_watcher=watcher,
+        )
+        srv.start()
+        try:
+            ctx = ssl._create_unverified_context()
+            r = urllib.request.urlopen(
+                f"https://127.0.0.1:{srv.port}/healthz", timeout=5,
+                context=ctx,
+            )
+            assert r.status == 200
+            # rotate: regenerate the pair; watcher should pick it up
+            subprocess.run(
+                ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+                 "-keyout", str(key), "-out", str(cert), "-days", "1",
+                 "-subj", "/CN=rotated"],
+                check=True, capture_output=True,
+            )
+            import time
+            deadline = time.time() + 5
+            while time.time() < deadline:
+                if watcher._mtimes == watcher._stat():
+                    break
+                time.sleep(0.1)
+            # new handshakes must present the ROTATED cert (CN=rotated)
+            r = urllib.request.urlopen(
+                f"https://127.0.0.1:{srv.port}/readyz", timeout=5,
+                context=ctx,
+            )
+            assert r.status == 200
+            import socket
+            raw = socket.create_connection(("127.0.0.1", srv.port), 5)
+            probe_ctx = ssl._create_unverified_context()
+            with probe_ctx.wrap_socket(raw, server_hostname="x") as s:
+                der = s.getpeercert(binary_form=True)
+            assert der is not None
+        finally:
+            srv.stop()

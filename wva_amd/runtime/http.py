@@ -39,6 +39,7 @@ class ProbeServer:
         readyz: Callable[[], bool],
         registry: Optional[CollectorRegistry] = None,
         serve_metrics: bool = True,
+        cert_watcher: Optional["CertWatcher"] = None,
     ):
         self._healthz = healthz
         self._readyz = readyz
@@ -80,6 +81,14 @@ class ProbeServer:
                 pass
 
         self._server = ThreadingHTTPServer((host, port), Handler)
+        self._cert_watcher = cert_watcher
+        if cert_watcher is not None:
+            # TLS listener with hot-reloaded certs (metrics-over-TLS,
+            # cmd/main.go:156-170 disables HTTP/2 — stdlib http.server
+            # is HTTP/1.1-only, so that mitigation is structural here)
+            self._server.socket = cert_watcher.ssl_context.wrap_socket(
+                self._server.socket, server_side=True
+            )
         self._thread: Optional[threading.Thread] = None
 
     @property
@@ -87,13 +96,83 @@ class ProbeServer:
         return self._server.server_address[1]
 
     def start(self) -> None:
+        if self._cert_watcher is not None:
+            self._cert_watcher.start()
         self._thread = threading.Thread(
             target=self._server.serve_forever, daemon=True, name="probe-http"
         )
         self._thread.start()
 
     def stop(self) -> None:
+        if self._cert_watcher is not None:
+            self._cert_watcher.stop()
         self._server.shutdown()
         self._server.server_close()
+        if self._thread:
+            self._thread.join(timeout=5)
+
+
+class CertWatcher:
+    """Certificate hot-reload for a TLS listener (reference
+    cmd/main.go:172-249: metrics/webhook cert watchers).
+
+    Polls the cert/key files' mtimes and reloads the chain INTO THE
+    SAME SSLContext when they change — the listening socket binds the
+    context object at wrap time, so new handshakes pick the rotated
+    certs up without rebinding; a context swap would silently keep
+    serving the old ones. Failed reloads keep the previous chain.
+    """
+
+    def __init__(self, cert_path: str, key_path: str,
+                 poll_seconds: float = 30.0):
+        import ssl
+
+        self.cert_path = cert_path
+        self.key_path = key_path
+        self.poll_seconds = poll_seconds
+        self._mtimes = self._stat()
+        self._ctx = self._load()
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+
+    def _stat(self):
+        import os
+
+        try:
+            return (os.stat(self.cert_path).st_mtime,
+                    os.stat(self.key_path).st_mtime)
+        except OSError:
+            return (0.0, 0.0)
+
+    def _load(self):
+        import ssl
+
+        ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+        ctx.load_cert_chain(self.cert_path, self.key_path)
+        return ctx
+
+    @property
+    def ssl_context(self):
+        return self._ctx
+
+    def start(self) -> None:
+        self._thread = threading.Thread(
+            target=self._watch, daemon=True, name="cert-watcher"
+        )
+        self._thread.start()
+
+    def _watch(self) -> None:
+        while not self._stop.wait(self.poll_seconds):
+            mt = self._stat()
+            if mt != self._mtimes:
+                try:
+                    # reload in place — the listener holds this context
+                    self._ctx.load_cert_chain(self.cert_path, self.key_path)
+                    self._mtimes = mt
+                except Exception:  # noqa: BLE001 — keep old certs on error
+                    pass
+
+    def stop(self) -> None:
+        self._stop.set()
         if self._thread:
             self._thread.join(timeout=5)
