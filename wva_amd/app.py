@@ -231,11 +231,20 @@ def build_app(
                 serve_metrics=False,
             )
         if metrics_addr and metrics_addr != "0":
+            watcher = None
+            infra = config.infra
+            if infra.metrics_cert_path and infra.metrics_key_path:
+                from .runtime.http import CertWatcher
+
+                watcher = CertWatcher(
+                    infra.metrics_cert_path, infra.metrics_key_path
+                )
             metrics_server = ProbeServer(
                 metrics_addr,
                 healthz=manager.healthz,
                 readyz=manager.readyz,
                 registry=emitter.registry,
+                cert_watcher=watcher,
             )
 
     return App(

@@ -71,6 +71,8 @@ class InfrastructureConfig:
     rest_client_timeout_seconds: float = 60.0
     secure_metrics: bool = True
     enable_http2: bool = False
+    metrics_cert_path: str = ""   # TLS cert for the metrics listener
+    metrics_key_path: str = ""    # (cert-manager mounts; hot-reloaded)
     watch_namespace: str = ""
     logger_verbosity: int = 0
     optimization_interval_seconds: float = 60.0
