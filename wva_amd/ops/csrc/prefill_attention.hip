@@ -83,7 +83,13 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
   const int lane = threadIdx.x & (WAVE_SIZE - 1);
   const int wave = threadIdx.x / WAVE_SIZE;
 
-  __shared__ unsigned int v_smem[KT_POS * ROW_DW];
+  // V staged TRANSPOSED: [dim][pos] u16 rows (stride 68 = 64 + 4 pad:
+  // 2·row mod 32 is distinct for the 16 rows of a fragment read, so the
+  // b64 pairs are bank-conflict-free). A B-fragment (k = 8 consecutive
+  // positions at fixed dim) is then one contiguous 16 B row segment
+  // instead of 8 scattered u16 reads.
+#define VT_ROW 68
+  __shared__ unsigned short v_smem_t[HEAD_DIM * VT_ROW];
   __shared__ float p_smem[NUM_WAVES][16][P_ROW];
 
   const bf16* k_slab =
@@ -134,17 +140,20 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
   for (int kt0 = 0; kt0 < kv_end; kt0 += KT_POS) {
     const int kn = min(KT_POS, S - kt0);
 
-    // --- stage V tile (cooperative, all waves) ---
+    // --- stage V tile transposed (cooperative, all waves): each
+    // thread reads one dword (2 dims at one position, coalesced along
+    // the HBM row) and scatters two u16s into dim-major rows ---
     {
-      typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
-      const int tile_u2 = kn * (HEAD_DIM / 4);
       const unsigned int* v_src = &v_base[(long)kt0 * (HEAD_DIM / 2)];
-      for (int idx = wave * WAVE_SIZE + lane; idx < tile_u2;
+      const int total = kn * (HEAD_DIM / 2);  // dwords in the tile
+      for (int idx = wave * WAVE_SIZE + lane; idx < total;
            idx += NUM_WAVES * WAVE_SIZE) {
-        const int row = idx >> 5;
-        const int d2 = (idx & 31) * 2;
-        const uint2_t val = *reinterpret_cast<const uint2_t*>(&v_src[idx * 2]);
-        *reinterpret_cast<uint2_t*>(&v_smem[row * ROW_DW + d2]) = val;
+        const int pos = idx >> 6;          // 64 dwords per position row
+        const int d2 = idx & 63;           // dim pair
+        const unsigned int val = v_src[idx];
+        v_smem_t[(2 * d2) * VT_ROW + pos] = (unsigned short)(val & 0xffffu);
+        v_smem_t[(2 * d2 + 1) * VT_ROW + pos] =
+            (unsigned short)(val >> 16);
       }
     }
 
@@ -233,10 +242,10 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
 
     __syncthreads();  // V staged + P visible (own pane only, but V needs it)
 
-    // --- O += P·V: A = P (row = q-row, k = positions), B = V from LDS ---
+    // --- O += P·V: A = P (row = q-row, k = positions), B = V from the
+    // transposed LDS tile (one contiguous row segment per fragment) ---
     {
-      const unsigned short* v_u16 =
-          reinterpret_cast<const unsigned short*>(v_smem);
+      const bool full = (kn == KT_POS);
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {  // 2 position subtiles of 32
         bf16x8_frag p_frag;
@@ -250,14 +259,16 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
 #pragma unroll
         for (int n = 0; n < 8; ++n) {
           const int dim = 16 * n + (lane % 16);
-          bf16x8_frag v_frag;
           const int p0 = 32 * kk + 8 * (lane / 16);
+          bf16x8_frag v_frag;
+          const unsigned short* vrow = &v_smem_t[dim * VT_ROW + p0];
+          if (full) {
 #pragma unroll
-          for (int i = 0; i < 8; ++i) {
-            const int t = p0 + i;
-            v_frag[i] = (t < kn)
-                ? (short)v_u16[(t * ROW_DW + (dim >> 1)) * 2 + (dim & 1)]
-                : (short)0;
+            for (int i = 0; i < 8; ++i) v_frag[i] = (short)vrow[i];
+          } else {
+#pragma unroll
+            for (int i = 0; i < 8; ++i)
+              v_frag[i] = (p0 + i < kn) ? (short)vrow[i] : (short)0;
           }
           o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               p_frag, v_frag, o_acc[n], 0, 0, 0);
