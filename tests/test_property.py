@@ -89,3 +89,51 @@ def test_parse_duration_never_hangs(s):
         assert isinstance(v, float)
     except ValueError:
         pass
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    st.text(alphabet=st.characters(whitelist_categories=("Ll", "Nd"),
+                                   max_codepoint=122), min_size=1, max_size=20),
+    st.dictionaries(
+        st.text(alphabet="abcdefxyz.-/", min_size=1, max_size=15),
+        st.text(alphabet="abcdefxyz0123456789.-", max_size=15),
+        max_size=4,
+    ),
+    st.integers(min_value=0, max_value=50),
+    st.integers(min_value=0, max_value=10 ** 9),
+)
+def test_serde_deployment_roundtrip_property(name, labels, replicas, rv):
+    from wva_amd.api.types import ObjectMeta
+    from wva_amd.kube import serde
+    from wva_amd.kube.objects import Container, Deployment, PodTemplateSpec
+
+    d = Deployment(
+        metadata=ObjectMeta(name=name, namespace="ns", labels=labels,
+                            resource_version=rv),
+        replicas=replicas,
+        selector=dict(labels),
+        template=PodTemplateSpec(labels=dict(labels),
+                                 containers=[Container(args=["--x"])]),
+    )
+    wire = serde.encode(d)
+    back = serde.decode("Deployment", wire)
+    assert back.metadata.name == name
+    assert back.metadata.labels == labels
+    assert back.replicas == replicas
+    assert back.metadata.resource_version == rv
+    assert serde.encode(back) == wire
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    st.floats(min_value=0.1, max_value=100.0, allow_nan=False),
+    st.floats(min_value=0.0, max_value=5.0, allow_nan=False),
+    st.integers(min_value=1, max_value=512),
+)
+def test_service_profile_itl_monotone(alpha, beta, batch):
+    from wva_amd.emulator.vllm_sim import ServiceProfile
+
+    p = ServiceProfile(alpha_ms=alpha, beta_ms=beta)
+    assert p.itl_ms(batch) >= p.itl_ms(max(batch - 1, 1)) - 1e-9
+    assert p.itl_ms(batch) >= alpha - 1e-9
