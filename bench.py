@@ -123,12 +123,14 @@ def build_scenario(profile, seed: int):
     # whole optimization interval (improvement over reference parity 1.0)
     service_time_s = OUTPUT_TOKENS * profile.itl_ms(profile.max_num_seqs) / 1000.0
     drain = min(1.0, service_time_s / ENGINE_INTERVAL_S)
-    config.update_saturation_config(
-        SaturationScalingConfig.from_dict({
-            "analyzerName": "saturation",
-            "schedulerQueueDrainFactor": drain,
-        })
-    )
+    sat = {
+        "analyzerName": "saturation",
+        "schedulerQueueDrainFactor": drain,
+    }
+    if os.environ.get("WVA_BENCH_LEAD"):
+        # experiment knob: predictive scale-up lead (seconds)
+        sat["scaleUpLeadSeconds"] = float(os.environ["WVA_BENCH_LEAD"])
+    config.update_saturation_config(SaturationScalingConfig.from_dict(sat))
     config.mark_bootstrap_complete()
     app = build_app(
         cluster, config, source=source,

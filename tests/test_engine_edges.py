@@ -180,3 +180,51 @@ class TestLimitedMode:
 
         app.saturation_engine.limiter.inventory = BoomInventory()
         app.saturation_engine.optimize()  # logged + returned, no raise
+
+
+class TestScaleUpLead:
+    def test_lead_inflates_required_on_rising_demand(self):
+        import time as _time
+
+        from wva_amd.analyzers.interfaces import AnalyzerResult
+        from wva_amd.config.saturation import SaturationScalingConfig
+        from wva_amd.engines.saturation import SaturationEngine
+
+        eng = SaturationEngine.__new__(SaturationEngine)
+        eng._demand_history = {}
+        cfg = SaturationScalingConfig.from_dict(
+            {"analyzerName": "saturation", "scaleUpLeadSeconds": 30.0}
+        )
+
+        r1 = AnalyzerResult(total_demand=1000.0, required_capacity=0.0)
+        out1 = eng._apply_scale_up_lead("m", "ns", cfg, r1)
+        assert out1.required_capacity == 0.0  # no history yet
+
+        eng._demand_history["m|ns"] = (_time.time() - 10.0, 1000.0)
+        r2 = AnalyzerResult(total_demand=1500.0, required_capacity=100.0)
+        out2 = eng._apply_scale_up_lead("m", "ns", cfg, r2)
+        assert out2.required_capacity > 100.0
+        # capped at 25% of demand / threshold
+        assert out2.required_capacity <= 100.0 + 0.25 * 1500.0 / cfg.scale_up_threshold + 1e-6
+
+    def test_lead_ignores_falling_demand_and_default_off(self):
+        import time as _time
+
+        from wva_amd.analyzers.interfaces import AnalyzerResult
+        from wva_amd.config.saturation import SaturationScalingConfig
+        from wva_amd.engines.saturation import SaturationEngine
+
+        eng = SaturationEngine.__new__(SaturationEngine)
+        eng._demand_history = {"m|ns": (_time.time() - 10.0, 2000.0)}
+        cfg = SaturationScalingConfig.from_dict(
+            {"analyzerName": "saturation", "scaleUpLeadSeconds": 30.0}
+        )
+        r = AnalyzerResult(total_demand=1500.0, required_capacity=50.0)
+        out = eng._apply_scale_up_lead("m", "ns", cfg, r)
+        assert out.required_capacity == 50.0  # falling: no inflation
+
+        cfg0 = SaturationScalingConfig.from_dict({"analyzerName": "saturation"})
+        eng._demand_history = {"m|ns": (_time.time() - 10.0, 1000.0)}
+        r2 = AnalyzerResult(total_demand=1500.0, required_capacity=50.0)
+        out2 = eng._apply_scale_up_lead("m", "ns", cfg0, r2)
+        assert out2.required_capacity == 50.0  # default off

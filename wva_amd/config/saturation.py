@@ -39,6 +39,11 @@ class SaturationScalingConfig:
     # over the reference for fast accelerators (MI355X drains hundreds of
     # requests per second per replica).
     scheduler_queue_drain_factor: float = 1.0
+    # Predictive scale-up: inflate V2 demand by its observed growth rate
+    # times this many seconds (covers pod-ready + engine-tick latency so
+    # capacity lands BEFORE the ramp reaches it). 0 disables (reference
+    # parity — the reference is purely reactive).
+    scale_up_lead_seconds: float = 0.0
     # Per-model override entries (keyed by "modelID|namespace"); populated by
     # the ConfigMap parser from the `overrides` list.
     overrides: Dict[str, "SaturationScalingConfig"] = field(default_factory=dict)
@@ -120,6 +125,7 @@ class SaturationScalingConfig:
             scheduler_queue_drain_factor=float(
                 d.get("schedulerQueueDrainFactor", 1.0)
             ),
+            scale_up_lead_seconds=float(d.get("scaleUpLeadSeconds", 0.0)),
         )
         cfg.apply_defaults()
         for entry in d.get("overrides") or []:
@@ -149,6 +155,8 @@ class SaturationScalingConfig:
             d["scaleDownBoundary"] = self.scale_down_boundary
         if self.scheduler_queue_drain_factor != 1.0:
             d["schedulerQueueDrainFactor"] = self.scheduler_queue_drain_factor
+        if self.scale_up_lead_seconds:
+            d["scaleUpLeadSeconds"] = self.scale_up_lead_seconds
         if self.overrides:
             d["overrides"] = [o.to_dict() for o in self.overrides.values()]
         return d

