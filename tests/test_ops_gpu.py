@@ -428,3 +428,29 @@ class TestPrefillAttnKernel:
         torch.testing.assert_close(
             out.float().reshape(B, S, hq, 128), ref, atol=4e-2, rtol=4e-2
         )
+
+
+class TestHBMCapacity:
+    def test_two_million_token_kv_cache_resident(self, dev):
+        """The capacity model's 288 GB claim, physically: allocate the
+        Llama-3-8B weights + a ~1.9M-token KV cache (≈248 GB total) and
+        run a decode step against the far end of it."""
+        import torch
+
+        from wva_amd.calibration.model import LLAMA_3_8B, LlamaDecodeModel
+
+        free, total = torch.cuda.mem_get_info()
+        if total < 280 * 2**30:
+            pytest.skip("not a 288 GB-class device")
+        # 8B: kv_bytes_per_token = 128 KiB -> max_batch*max_seq tokens
+        # 64 * 29696 = 1,900,544 tokens ~ 232 GiB of cache
+        model = LlamaDecodeModel(LLAMA_3_8B, max_batch=64, max_seq=29696)
+        toks = model.max_batch * model.max_seq
+        assert toks >= 1_900_000
+        model.context_lens[:64] = model.max_seq - 2  # write near the end
+        t = torch.randint(0, LLAMA_3_8B.vocab_size, (64,), device=dev)
+        logits = model.decode_step(t)
+        torch.cuda.synchronize()
+        assert torch.isfinite(logits.float()).all()
+        used = total - torch.cuda.mem_get_info()[0]
+        assert used > 200 * 2**30  # the capacity is genuinely resident
