@@ -129,13 +129,15 @@ def derive_kv_capacity(
     gpu_memory_utilization: float = 0.9,
     block_size: int = 16,
     hbm_bytes: Optional[int] = None,
+    kv_dtype_bytes: int = 2,
 ) -> Tuple[int, int]:
     """(num_gpu_blocks, kv_capacity_tokens) from free HBM after weights —
-    the vllm:cache_config_info analog, sized for 288 GB HBM3E."""
+    the vllm:cache_config_info analog, sized for 288 GB HBM3E. fp8 KV
+    (kv_dtype_bytes=1) doubles the token capacity."""
     if hbm_bytes is None:
         hbm_bytes = torch.cuda.get_device_properties(0).total_memory
     budget = int(hbm_bytes * gpu_memory_utilization) - cfg.weight_bytes()
-    per_token = cfg.kv_bytes_per_token()
+    per_token = cfg.kv_bytes_per_token(kv_dtype_bytes)
     tokens = max(budget // per_token, 0)
     blocks = tokens // block_size
     return blocks, blocks * block_size
@@ -149,21 +151,34 @@ def calibrate_service_profile(
     gpu_count: int = 1,
     iters: int = 8,
     use_graph: bool = False,
+    kv_dtype: str = "bf16",
 ) -> Tuple[ServiceProfile, CalibrationResult]:
     """Measure on the current GPU and return an emulator ServiceProfile
-    + the raw calibration record."""
+    + the raw calibration record. kv_dtype="fp8" measures the e4m3
+    KV-cache serving configuration (decode only; prefill throughput is
+    then measured on a separate bf16-cache engine)."""
     if batch_sizes is None:
         batch_sizes = [1, 2, 4, 8, 16, 32, 64]
     max_batch = max(batch_sizes)
-    model = LlamaDecodeModel(cfg, max_batch=max_batch, max_seq=max_seq)
-    prefill_tps = measure_prefill_tps(model)
+    model = LlamaDecodeModel(
+        cfg, max_batch=max_batch, max_seq=max_seq, kv_dtype=kv_dtype
+    )
+    if kv_dtype == "bf16":
+        prefill_tps = measure_prefill_tps(model)
+    else:
+        pf_model = LlamaDecodeModel(cfg, max_batch=2, max_seq=max_seq)
+        prefill_tps = measure_prefill_tps(pf_model)
+        del pf_model
+        torch.cuda.empty_cache()
     itl: List[float] = []
     for b in batch_sizes:
         itl.append(
             measure_itl(model, b, context_len, iters=iters, use_graph=use_graph)
         )
     alpha, beta, r2 = fit_itl_curve(batch_sizes, itl)
-    blocks, kv_tokens = derive_kv_capacity(cfg)
+    blocks, kv_tokens = derive_kv_capacity(
+        cfg, kv_dtype_bytes=1 if kv_dtype == "fp8" else 2
+    )
     peak_tps = max(
         b / (t / 1000.0) for b, t in zip(batch_sizes, itl)
     )

@@ -197,7 +197,7 @@ def rope_append_kv(
     k.copy_(kr.to(k.dtype))
     # head-major cache [B, Hk, S, D]
     idx = torch.arange(B)
-    k_cache[idx, :, positions.long()] = k
+    k_cache[idx, :, positions.long()] = k.to(k_cache.dtype)
     v_cache[idx, :, positions.long()] = v.to(v_cache.dtype)
     return q
 

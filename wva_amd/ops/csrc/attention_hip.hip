@@ -243,12 +243,13 @@ __device__ __forceinline__ short f2bf_bits(float v) {
 
 #define S_ROW 65  // s_smem row stride (floats): 64 + 1 pad
 
+template <bool KV_FP8>
 __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v4_kernel(
     bf16* __restrict__ out,
     float* __restrict__ workspace,
     const bf16* __restrict__ q,
-    const bf16* __restrict__ k_cache,
-    const bf16* __restrict__ v_cache,
+    const void* __restrict__ k_cache,  // bf16 or e4m3 (KV_FP8)
+    const void* __restrict__ v_cache,
     const int* __restrict__ context_lens,
     const int num_q_heads,
     const int num_kv_heads,
@@ -300,16 +301,37 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v4_kernel(
   }
 
   const long kv_row_dw = HEAD_DIM / 2;
-  const bf16* k_slab =
-      k_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM;
+  const long slab_off = ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM;
+  const bf16* k_slab = reinterpret_cast<const bf16*>(k_cache) + slab_off;
+  const fp8_t* k_slab8 = reinterpret_cast<const fp8_t*>(k_cache) + slab_off;
   const unsigned int* v_base = reinterpret_cast<const unsigned int*>(
-      v_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM);
+      reinterpret_cast<const bf16*>(v_cache) + slab_off);
+  const fp8_t* v_base8 = reinterpret_cast<const fp8_t*>(v_cache) + slab_off;
 
   for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
     const int tn = min(TILE, ctx - t0);
 
-    // --- stage V only (v3 staging loop, half the traffic) ---
-    {
+    // --- stage V only; the fp8 path up-converts during staging so the
+    // LDS layout and PV sweep are identical in both modes ---
+    if constexpr (KV_FP8) {
+      const int tile_g8 = tn * (HEAD_DIM / 8);
+      const fp8_t* v_src = v_base8 + (long)t0 * HEAD_DIM;
+      for (int idx = wave * WAVE_SIZE + lane; idx < tile_g8;
+           idx += NUM_WAVES * WAVE_SIZE) {
+        const int row = idx >> 4;          // 16 groups of 8 per row
+        const int d8 = (idx & 15) * 8;
+        const unsigned short* src =
+            reinterpret_cast<const unsigned short*>(
+                v_src + (long)row * HEAD_DIM + d8);
+        unsigned int* dst = &v_smem[row * ROW_DW + d8 / 2];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const float2_vt f = unpk2_fp8(src[j]);
+          const bf16x2 packed{f2bf(f[0]), f2bf(f[1])};
+          dst[j] = *reinterpret_cast<const unsigned int*>(&packed);
+        }
+      }
+    } else {
       typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
       const int tile_u2 = tn * (HEAD_DIM / 4);
       const unsigned int* v_src = &v_base[(long)t0 * kv_row_dw];
@@ -327,16 +349,26 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v4_kernel(
       const int prow = 16 * wave + (lane % 16);       // A row = position
       const int pos = t0 + prow;
       const int pos_c = pos < max_seq ? pos : max_seq - 1;  // clamped load
-      const short* krow =
-          reinterpret_cast<const short*>(k_slab + (long)pos_c * HEAD_DIM);
       const int k0 = 8 * (lane / 16);
       f32x4_frag sf = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         bf16x8_frag a_frag;
-        const short* src = krow + 32 * kk + k0;
+        if constexpr (KV_FP8) {
+          const unsigned short* src = reinterpret_cast<const unsigned short*>(
+              k_slab8 + (long)pos_c * HEAD_DIM + 32 * kk + k0);
 #pragma unroll
-        for (int i = 0; i < 8; ++i) a_frag[i] = src[i];
+          for (int j = 0; j < 4; ++j) {
+            const float2_vt f = unpk2_fp8(src[j]);
+            a_frag[2 * j] = f2bf_bits(f[0]);
+            a_frag[2 * j + 1] = f2bf_bits(f[1]);
+          }
+        } else {
+          const short* src = reinterpret_cast<const short*>(
+              k_slab + (long)pos_c * HEAD_DIM) + 32 * kk + k0;
+#pragma unroll
+          for (int i = 0; i < 8; ++i) a_frag[i] = src[i];
+        }
         sf = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, q_frag[kk], sf,
                                                      0, 0, 0);
       }
@@ -427,12 +459,13 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v4_kernel(
 // p row) and LDS broadcast of corr/s; softmax state stays wave-owned.
 // ---------------------------------------------------------------------------
 
+template <bool KV_FP8>
 __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
     bf16* __restrict__ out,
     float* __restrict__ workspace,
     const bf16* __restrict__ q,
-    const bf16* __restrict__ k_cache,
-    const bf16* __restrict__ v_cache,
+    const void* __restrict__ k_cache,  // bf16 or e4m3 (KV_FP8)
+    const void* __restrict__ v_cache,
     const int* __restrict__ context_lens,
     const int num_q_heads,
     const int num_kv_heads,
@@ -489,16 +522,36 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
                           f32x4_frag{0.f, 0.f, 0.f, 0.f}};
 
   const long kv_row_dw = HEAD_DIM / 2;
-  const bf16* k_slab =
-      k_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM;
+  const long slab_off = ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM;
+  const bf16* k_slab = reinterpret_cast<const bf16*>(k_cache) + slab_off;
+  const fp8_t* k_slab8 = reinterpret_cast<const fp8_t*>(k_cache) + slab_off;
   const unsigned int* v_base = reinterpret_cast<const unsigned int*>(
-      v_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM);
+      reinterpret_cast<const bf16*>(v_cache) + slab_off);
+  const fp8_t* v_base8 = reinterpret_cast<const fp8_t*>(v_cache) + slab_off;
 
   for (int t0 = split * TILE; t0 < ctx; t0 += num_splits * TILE) {
     const int tn = min(TILE, ctx - t0);
 
-    // --- stage V (v4 loop) ---
-    {
+    // --- stage V (fp8 up-converts during staging; LDS layout fixed) ---
+    if constexpr (KV_FP8) {
+      const int tile_g8 = tn * (HEAD_DIM / 8);
+      const fp8_t* v_src = v_base8 + (long)t0 * HEAD_DIM;
+      for (int idx = wave * WAVE_SIZE + lane; idx < tile_g8;
+           idx += NUM_WAVES * WAVE_SIZE) {
+        const int row = idx >> 4;
+        const int d8 = (idx & 15) * 8;
+        const unsigned short* src =
+            reinterpret_cast<const unsigned short*>(
+                v_src + (long)row * HEAD_DIM + d8);
+        unsigned int* dst = &v_smem[row * ROW_DW + d8 / 2];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const float2_vt f = unpk2_fp8(src[j]);
+          const bf16x2 packed{f2bf(f[0]), f2bf(f[1])};
+          dst[j] = *reinterpret_cast<const unsigned int*>(&packed);
+        }
+      }
+    } else {
       typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
       const int tile_u2 = tn * (HEAD_DIM / 4);
       const unsigned int* v_src = &v_base[(long)t0 * kv_row_dw];
@@ -516,16 +569,26 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
       const int prow = 16 * wave + (lane % 16);
       const int pos = t0 + prow;
       const int pos_c = pos < max_seq ? pos : max_seq - 1;
-      const short* krow =
-          reinterpret_cast<const short*>(k_slab + (long)pos_c * HEAD_DIM);
       const int k0 = 8 * (lane / 16);
       f32x4_frag sf = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         bf16x8_frag a_frag;
-        const short* src = krow + 32 * kk + k0;
+        if constexpr (KV_FP8) {
+          const unsigned short* src = reinterpret_cast<const unsigned short*>(
+              k_slab8 + (long)pos_c * HEAD_DIM + 32 * kk + k0);
 #pragma unroll
-        for (int i = 0; i < 8; ++i) a_frag[i] = src[i];
+          for (int j = 0; j < 4; ++j) {
+            const float2_vt f = unpk2_fp8(src[j]);
+            a_frag[2 * j] = f2bf_bits(f[0]);
+            a_frag[2 * j + 1] = f2bf_bits(f[1]);
+          }
+        } else {
+          const short* src = reinterpret_cast<const short*>(
+              k_slab + (long)pos_c * HEAD_DIM) + 32 * kk + k0;
+#pragma unroll
+          for (int i = 0; i < 8; ++i) a_frag[i] = src[i];
+        }
         sf = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, q_frag[kk], sf,
                                                      0, 0, 0);
       }
@@ -692,6 +755,17 @@ __global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
   orow[lane] = bf16x2{f2bf(o0 * inv), f2bf(o1 * inv)};
 }
 
+extern "C" void launch_gqa_decode_attn_v4_ex(
+    void* out, void* workspace, const void* q, const void* k_cache,
+    const void* v_cache, const int* context_lens, int batch, int num_q_heads,
+    int num_kv_heads, int max_seq, int num_splits, float scale, int kv_fp8,
+    hipStream_t stream);
+extern "C" void launch_gqa_decode_attn_v5_ex(
+    void* out, void* workspace, const void* q, const void* k_cache,
+    const void* v_cache, const int* context_lens, int batch, int num_q_heads,
+    int num_kv_heads, int max_seq, int num_splits, float scale, int kv_fp8,
+    hipStream_t stream);
+
 extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
                                           int max_ctx_hint) {
   // Fill the v4 kernel's resident workgroups: 70 VGPRs / ~21 KiB LDS
@@ -730,12 +804,29 @@ extern "C" void launch_gqa_decode_attn_v5(
     const void* v_cache, const int* context_lens, int batch, int num_q_heads,
     int num_kv_heads, int max_seq, int num_splits, float scale,
     hipStream_t stream) {
+  launch_gqa_decode_attn_v5_ex(out, workspace, q, k_cache, v_cache,
+                               context_lens, batch, num_q_heads, num_kv_heads,
+                               max_seq, num_splits, scale, 0, stream);
+}
+
+extern "C" void launch_gqa_decode_attn_v5_ex(
+    void* out, void* workspace, const void* q, const void* k_cache,
+    const void* v_cache, const int* context_lens, int batch, int num_q_heads,
+    int num_kv_heads, int max_seq, int num_splits, float scale, int kv_fp8,
+    hipStream_t stream) {
   dim3 grid(batch, num_kv_heads, num_splits);
   dim3 block(256);
-  hipLaunchKernelGGL(gqa_decode_attn_v5_kernel, grid, block, 0, stream,
-                     (bf16*)out, (float*)workspace, (const bf16*)q,
-                     (const bf16*)k_cache, (const bf16*)v_cache, context_lens,
-                     num_q_heads, num_kv_heads, max_seq, scale);
+  if (kv_fp8) {
+    hipLaunchKernelGGL(gqa_decode_attn_v5_kernel<true>, grid, block, 0,
+                       stream, (bf16*)out, (float*)workspace, (const bf16*)q,
+                       k_cache, v_cache, context_lens, num_q_heads,
+                       num_kv_heads, max_seq, scale);
+  } else {
+    hipLaunchKernelGGL(gqa_decode_attn_v5_kernel<false>, grid, block, 0,
+                       stream, (bf16*)out, (float*)workspace, (const bf16*)q,
+                       k_cache, v_cache, context_lens, num_q_heads,
+                       num_kv_heads, max_seq, scale);
+  }
   if (num_splits > 1) {
     dim3 mgrid(batch, num_q_heads);
     hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64), 0,
@@ -749,12 +840,29 @@ extern "C" void launch_gqa_decode_attn_v4(
     const void* v_cache, const int* context_lens, int batch, int num_q_heads,
     int num_kv_heads, int max_seq, int num_splits, float scale,
     hipStream_t stream) {
+  launch_gqa_decode_attn_v4_ex(out, workspace, q, k_cache, v_cache,
+                               context_lens, batch, num_q_heads, num_kv_heads,
+                               max_seq, num_splits, scale, 0, stream);
+}
+
+extern "C" void launch_gqa_decode_attn_v4_ex(
+    void* out, void* workspace, const void* q, const void* k_cache,
+    const void* v_cache, const int* context_lens, int batch, int num_q_heads,
+    int num_kv_heads, int max_seq, int num_splits, float scale, int kv_fp8,
+    hipStream_t stream) {
   dim3 grid(batch, num_kv_heads, num_splits);
   dim3 block(256);
-  hipLaunchKernelGGL(gqa_decode_attn_v4_kernel, grid, block, 0, stream,
-                     (bf16*)out, (float*)workspace, (const bf16*)q,
-                     (const bf16*)k_cache, (const bf16*)v_cache, context_lens,
-                     num_q_heads, num_kv_heads, max_seq, scale);
+  if (kv_fp8) {
+    hipLaunchKernelGGL(gqa_decode_attn_v4_kernel<true>, grid, block, 0,
+                       stream, (bf16*)out, (float*)workspace, (const bf16*)q,
+                       k_cache, v_cache, context_lens, num_q_heads,
+                       num_kv_heads, max_seq, scale);
+  } else {
+    hipLaunchKernelGGL(gqa_decode_attn_v4_kernel<false>, grid, block, 0,
+                       stream, (bf16*)out, (float*)workspace, (const bf16*)q,
+                       k_cache, v_cache, context_lens, num_q_heads,
+                       num_kv_heads, max_seq, scale);
+  }
   if (num_splits > 1) {
     dim3 mgrid(batch, num_q_heads);
     hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64), 0,

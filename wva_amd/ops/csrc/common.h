@@ -45,3 +45,30 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* lds) {
   out = wave_reduce_sum(out);  // only first num_waves lanes carry data
   return __shfl(out, 0, WAVE_SIZE);
 }
+
+// --- fp8 (OCP e4m3fn) KV-cache helpers, gfx950 hardware converts ---
+// The fp8 KV path stores K/V as 1-byte e4m3 (half the HBM traffic and
+// double the KV capacity of bf16); compute stays bf16/fp32 — fragments
+// are up-converted at build time through these.
+using fp8_t = unsigned char;
+typedef __attribute__((ext_vector_type(2))) float float2_vt;
+
+__device__ __forceinline__ fp8_t f2fp8(float v) {
+  return (fp8_t)(__builtin_amdgcn_cvt_pk_fp8_f32(v, 0.0f, 0u, false) & 0xffu);
+}
+
+// pack two floats -> two adjacent e4m3 bytes
+__device__ __forceinline__ unsigned short pk2_fp8(float a, float b) {
+  return (unsigned short)(__builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0u, false) &
+                          0xffffu);
+}
+
+// two adjacent e4m3 bytes -> two floats
+__device__ __forceinline__ float2_vt unpk2_fp8(unsigned short v) {
+  return __builtin_amdgcn_cvt_pk_f32_fp8((unsigned int)v, false);
+}
+
+__device__ __forceinline__ float fp8_to_f32(fp8_t v) {
+  const float2_vt f = __builtin_amdgcn_cvt_pk_f32_fp8((unsigned int)v, false);
+  return f[0];
+}

@@ -19,6 +19,30 @@ extern "C" void launch_silu_mul(void* out, const void* gate, const void* up,
 extern "C" void launch_silu_mul_fused(void* out, const void* gate_up,
                                       long rows, long inter,
                                       hipStream_t stream);
+extern "C" void launch_rope_append_kv_fp8(const void* qkv, void* q_out,
+                                          void* k_cache, void* v_cache,
+                                          const int* positions, int batch,
+                                          int num_q_heads, int num_kv_heads,
+                                          int head_dim, int max_seq,
+                                          float theta, hipStream_t stream);
+extern "C" void launch_gqa_decode_attn_v4_ex(void* out, void* workspace,
+                                             const void* q,
+                                             const void* k_cache,
+                                             const void* v_cache,
+                                             const int* context_lens,
+                                             int batch, int num_q_heads,
+                                             int num_kv_heads, int max_seq,
+                                             int num_splits, float scale,
+                                             int kv_fp8, hipStream_t stream);
+extern "C" void launch_gqa_decode_attn_v5_ex(void* out, void* workspace,
+                                             const void* q,
+                                             const void* k_cache,
+                                             const void* v_cache,
+                                             const int* context_lens,
+                                             int batch, int num_q_heads,
+                                             int num_kv_heads, int max_seq,
+                                             int num_splits, float scale,
+                                             int kv_fp8, hipStream_t stream);
 extern "C" void launch_rope_append_kv(const void* qkv, void* q_out,
                                       void* k_cache, void* v_cache,
                                       const int* positions, int batch,
@@ -69,6 +93,17 @@ void check_bf16_contig(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
   TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// KV caches may be bf16 or fp8 e4m3 (torch float8_e4m3fn); returns true
+// when the cache is fp8.
+bool check_cache_contig(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  if (t.scalar_type() == torch::kBFloat16) return false;
+  TORCH_CHECK(t.scalar_type() == at::kFloat8_e4m3fn,
+              name, " must be bf16 or float8_e4m3fn");
+  return true;
 }
 
 // out = rmsnorm(input [+ residual]) * weight; when residual is given it is
@@ -132,8 +167,9 @@ torch::Tensor rope_append_kv(torch::Tensor qkv, torch::Tensor k_cache,
                              int64_t num_q_heads, int64_t num_kv_heads,
                              double theta) {
   check_bf16_contig(qkv, "qkv");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool fp8 = check_cache_contig(k_cache, "k_cache");
+  TORCH_CHECK(check_cache_contig(v_cache, "v_cache") == fp8,
+              "k/v cache dtype mismatch");
   TORCH_CHECK(positions.is_cuda() && positions.scalar_type() == torch::kInt32,
               "positions must be int32 on GPU");
   TORCH_CHECK(qkv.dim() == 2, "qkv must be [B, (Hq+2Hk)*D]");
@@ -145,10 +181,11 @@ torch::Tensor rope_append_kv(torch::Tensor qkv, torch::Tensor k_cache,
   TORCH_CHECK(qkv.size(1) == (num_q_heads + 2 * num_kv_heads) * head_dim,
               "qkv width mismatch");
   auto q_out = torch::empty({batch, num_q_heads, head_dim}, qkv.options());
-  launch_rope_append_kv(qkv.data_ptr(), q_out.data_ptr(), k_cache.data_ptr(),
-                        v_cache.data_ptr(), positions.data_ptr<int>(), batch,
-                        (int)num_q_heads, (int)num_kv_heads, head_dim, max_seq,
-                        (float)theta, current_stream());
+  auto launch = fp8 ? launch_rope_append_kv_fp8 : launch_rope_append_kv;
+  launch(qkv.data_ptr(), q_out.data_ptr(), k_cache.data_ptr(),
+         v_cache.data_ptr(), positions.data_ptr<int>(), batch,
+         (int)num_q_heads, (int)num_kv_heads, head_dim, max_seq,
+         (float)theta, current_stream());
   return q_out;
 }
 
@@ -169,8 +206,11 @@ torch::Tensor gqa_decode_attn_impl(torch::Tensor q, torch::Tensor k_cache,
                                    torch::Tensor context_lens, double scale,
                                    int use_v4) {
   check_bf16_contig(q, "q");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool kv_fp8 = check_cache_contig(k_cache, "k_cache");
+  TORCH_CHECK(check_cache_contig(v_cache, "v_cache") == kv_fp8,
+              "k/v cache dtype mismatch");
+  TORCH_CHECK(!(kv_fp8 && use_v4 == 0),
+              "fp8 KV cache requires the v4/v5 MFMA kernels");
   TORCH_CHECK(context_lens.is_cuda() &&
                   context_lens.scalar_type() == torch::kInt32,
               "context_lens must be int32 on GPU");
@@ -198,14 +238,21 @@ torch::Tensor gqa_decode_attn_impl(torch::Tensor q, torch::Tensor k_cache,
         q.options().dtype(torch::kFloat32));
     ws_ptr = workspace.data_ptr();
   }
-  auto launch = use_v4 == 2 ? launch_gqa_decode_attn_v5
-                : use_v4 == 1 ? launch_gqa_decode_attn_v4
-                              : launch_gqa_decode_attn;
-  launch(out.data_ptr(), ws_ptr, q.data_ptr(),
-         k_cache.data_ptr(), v_cache.data_ptr(),
-         context_lens.data_ptr<int>(), batch, num_q_heads,
-         num_kv_heads, max_seq, num_splits, (float)scale,
-         current_stream());
+  if (use_v4 == 0) {
+    launch_gqa_decode_attn(out.data_ptr(), ws_ptr, q.data_ptr(),
+                           k_cache.data_ptr(), v_cache.data_ptr(),
+                           context_lens.data_ptr<int>(), batch, num_q_heads,
+                           num_kv_heads, max_seq, num_splits, (float)scale,
+                           current_stream());
+  } else {
+    auto launch = use_v4 == 2 ? launch_gqa_decode_attn_v5_ex
+                              : launch_gqa_decode_attn_v4_ex;
+    launch(out.data_ptr(), ws_ptr, q.data_ptr(),
+           k_cache.data_ptr(), v_cache.data_ptr(),
+           context_lens.data_ptr<int>(), batch, num_q_heads,
+           num_kv_heads, max_seq, num_splits, (float)scale,
+           kv_fp8 ? 1 : 0, current_stream());
+  }
   return out;
 }
 
