@@ -226,3 +226,24 @@ class TestFp8KvCpu:
         t = torch.randint(0, TINY.vocab_size, (1, 4))
         logits = m.prefill(t)
         assert torch.isfinite(logits.float()).all()
+
+    def test_fp8_moe_and_tp_engines(self):
+        import torch
+
+        from wva_amd.calibration.moe_model import TINY_MOE, MixtralDecodeModel
+        from wva_amd.calibration.tp_model import TPLlamaDecodeModel
+        from wva_amd.calibration.model import TINY
+
+        moe = MixtralDecodeModel(TINY_MOE, max_batch=2, max_seq=32,
+                                 device="cpu", kv_dtype="fp8")
+        assert moe.k_cache[0].dtype == torch.float8_e4m3fn
+        moe.reset(2, 8)
+        out = moe.decode_step(torch.randint(0, TINY_MOE.vocab_size, (2,)))
+        assert torch.isfinite(out.float()).all()
+
+        tp = TPLlamaDecodeModel(TINY, max_batch=2, max_seq=32, device="cpu",
+                                kv_dtype="fp8")
+        assert tp.k_cache[0].dtype == torch.float8_e4m3fn
+        tp.reset(2, 8)
+        out = tp.decode_step(torch.randint(0, TINY.vocab_size, (2,)))
+        assert torch.isfinite(out.float()).all()

@@ -142,10 +142,15 @@ class MixtralDecodeModel:
         max_seq: int = 2048,
         device: str = "cuda",
         seed: int = 0,
+        kv_dtype: str = "bf16",
     ):
         self.cfg = cfg
         self.device = torch.device(device)
         self.dtype = torch.bfloat16
+        self.kv_dtype = kv_dtype
+        self.cache_dtype = (
+            torch.float8_e4m3fn if kv_dtype == "fp8" else torch.bfloat16
+        )
         self.max_batch = max_batch
         self.max_seq = max_seq
         gen = torch.Generator(device=self.device)
@@ -168,7 +173,7 @@ class MixtralDecodeModel:
         self.k_cache = [
             torch.zeros(
                 max_batch, cfg.num_kv_heads, max_seq, cfg.head_dim,
-                device=self.device, dtype=self.dtype,
+                device=self.device, dtype=self.cache_dtype,
             )
             for _ in range(cfg.num_layers)
         ]
@@ -184,8 +189,15 @@ class MixtralDecodeModel:
         self.context_lens.zero_()
         self.context_lens[:batch] = context_len
         for layer in range(self.cfg.num_layers):
-            self.k_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
-            self.v_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+            if self.cache_dtype == torch.bfloat16:
+                self.k_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+                self.v_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+            else:  # normal_ unsupported on float8: generate + cast
+                shape = self.k_cache[layer][:batch, :, :context_len].shape
+                for cache in (self.k_cache, self.v_cache):
+                    cache[layer][:batch, :, :context_len] = torch.randn(
+                        shape, device=self.device, dtype=torch.bfloat16
+                    ).to(self.cache_dtype)
 
     def _moe_mlp(self, layer: _MoELayer, h2: torch.Tensor) -> torch.Tensor:
         """MoE MLP; dense-batched for decode batches (B ≤ 64).
@@ -296,6 +308,10 @@ class MixtralDecodeModel:
         `nonzero()` syncs amortize over one large pass.
         """
         cfg = self.cfg
+        if self.kv_dtype != "bf16" and token_ids.is_cuda:
+            raise NotImplementedError(
+                "fp8 KV prefill is not implemented (decode-only feature)"
+            )
         B, S = token_ids.shape
         T = B * S
         if S > self.max_seq:
@@ -341,13 +357,13 @@ class MixtralDecodeModel:
                     .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
                 )
                 kh = (
-                    self.k_cache[li][:B, :, :S]
+                    self.k_cache[li][:B, :, :S].to(self.dtype)
                     .unsqueeze(2)
                     .expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
                     .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)
                 )
                 vh = (
-                    self.v_cache[li][:B, :, :S]
+                    self.v_cache[li][:B, :, :S].to(self.dtype)
                     .unsqueeze(2)
                     .expand(B, cfg.num_kv_heads, G, S, cfg.head_dim)
                     .reshape(B * cfg.num_kv_heads * G, S, cfg.head_dim)

@@ -81,8 +81,13 @@ class TPLlamaDecodeModel:
         device: str = "cuda",
         seed: int = 0,
         group: Optional[dist.ProcessGroup] = None,
+        kv_dtype: str = "bf16",
     ):
         self.cfg = cfg
+        self.kv_dtype = kv_dtype
+        self.cache_dtype = (
+            torch.float8_e4m3fn if kv_dtype == "fp8" else torch.bfloat16
+        )
         self.group = group
         self.tp = dist.get_world_size(group) if dist.is_initialized() else 1
         self.rank = dist.get_rank(group) if dist.is_initialized() else 0
@@ -114,7 +119,7 @@ class TPLlamaDecodeModel:
         self.k_cache = [
             torch.zeros(
                 max_batch, self.shard.num_kv_heads, max_seq, cfg.head_dim,
-                device=self.device, dtype=self.dtype,
+                device=self.device, dtype=self.cache_dtype,
             )
             for _ in range(cfg.num_layers)
         ]
@@ -135,8 +140,15 @@ class TPLlamaDecodeModel:
         self.context_lens.zero_()
         self.context_lens[:batch] = context_len
         for layer in range(self.cfg.num_layers):
-            self.k_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
-            self.v_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+            if self.cache_dtype == torch.bfloat16:
+                self.k_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+                self.v_cache[layer][:batch, :, :context_len].normal_(0.0, 1.0)
+            else:  # normal_ unsupported on float8: generate + cast
+                shape = self.k_cache[layer][:batch, :, :context_len].shape
+                for cache in (self.k_cache, self.v_cache):
+                    cache[layer][:batch, :, :context_len] = torch.randn(
+                        shape, device=self.device, dtype=torch.bfloat16
+                    ).to(self.cache_dtype)
 
     @torch.no_grad()
     def decode_step(self, token_ids: torch.Tensor) -> torch.Tensor:
