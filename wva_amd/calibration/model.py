@@ -238,11 +238,6 @@ class LlamaDecodeModel:
         row-shaped and batch-size-agnostic).
         """
         cfg = self.cfg
-        if self.kv_dtype != "bf16" and token_ids.is_cuda:
-            raise NotImplementedError(
-                "fp8 KV prefill is not implemented (decode-only feature); "
-                "prefill with kv_dtype='bf16'"
-            )
         B, S = token_ids.shape
         T = B * S
         if S > self.max_seq:

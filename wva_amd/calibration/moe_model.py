@@ -308,10 +308,6 @@ class MixtralDecodeModel:
         `nonzero()` syncs amortize over one large pass.
         """
         cfg = self.cfg
-        if self.kv_dtype != "bf16" and token_ids.is_cuda:
-            raise NotImplementedError(
-                "fp8 KV prefill is not implemented (decode-only feature)"
-            )
         B, S = token_ids.shape
         T = B * S
         if S > self.max_seq:

@@ -52,11 +52,13 @@ extern "C" void launch_rope_append_kv(const void* qkv, void* q_out,
 extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
                                           int max_ctx_hint);
 extern "C" int skinny_gemm_num_splits(int N, int K, int nt);
-extern "C" void launch_prefill_attn(void* out, const void* q,
-                                    const void* k_cache, const void* v_cache,
-                                    int batch, int seq, int num_q_heads,
-                                    int num_kv_heads, int max_seq,
-                                    float scale, hipStream_t stream);
+extern "C" void launch_prefill_attn_ex(void* out, const void* q,
+                                       const void* k_cache,
+                                       const void* v_cache, int batch,
+                                       int seq, int num_q_heads,
+                                       int num_kv_heads, int max_seq,
+                                       float scale, int kv_fp8,
+                                       hipStream_t stream);
 extern "C" int skinny_gemm_tile_n(int M);
 extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
                                    const void* w, int M, int N, int K,
@@ -321,8 +323,9 @@ torch::Tensor prefill_attn(torch::Tensor q, torch::Tensor k_cache,
                            torch::Tensor v_cache, int64_t batch, int64_t seq,
                            double scale) {
   check_bf16_contig(q, "q");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool kv_fp8 = check_cache_contig(k_cache, "k_cache");
+  TORCH_CHECK(check_cache_contig(v_cache, "v_cache") == kv_fp8,
+              "k/v cache dtype mismatch");
   TORCH_CHECK(q.dim() == 3 && q.size(2) == 128, "q must be [T, Hq, 128]");
   TORCH_CHECK(q.size(0) == batch * seq, "T must equal B*S");
   TORCH_CHECK(k_cache.dim() == 4, "k_cache must be [B, Hk, S_max, D]");
@@ -333,9 +336,10 @@ torch::Tensor prefill_attn(torch::Tensor q, torch::Tensor k_cache,
   TORCH_CHECK(num_q_heads % num_kv_heads == 0, "Hq must divide by Hk");
   TORCH_CHECK(num_q_heads / num_kv_heads <= 8, "GQA group size must be <= 8");
   auto out = torch::empty_like(q);
-  launch_prefill_attn(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
-                      v_cache.data_ptr(), (int)batch, (int)seq, num_q_heads,
-                      num_kv_heads, max_seq, (float)scale, current_stream());
+  launch_prefill_attn_ex(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                         v_cache.data_ptr(), (int)batch, (int)seq,
+                         num_q_heads, num_kv_heads, max_seq, (float)scale,
+                         kv_fp8 ? 1 : 0, current_stream());
   return out;
 }
 

@@ -63,11 +63,12 @@ __device__ __forceinline__ float group16_sum(float v) {
   return v;
 }
 
+template <bool KV_FP8>
 __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
     bf16* __restrict__ out,            // [T, Hq, 128]
     const bf16* __restrict__ q,        // [T, Hq, 128]
-    const bf16* __restrict__ k_cache,  // [B, Hk, S_max, 128]
-    const bf16* __restrict__ v_cache,  // [B, Hk, S_max, 128]
+    const void* __restrict__ k_cache,  // [B, Hk, S_max, 128] bf16|e4m3
+    const void* __restrict__ v_cache,
     const int B,
     const int S,
     const int num_q_heads,
@@ -93,10 +94,12 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
   __shared__ unsigned short v_smem_t[HEAD_DIM * VT_ROW];
   __shared__ float p_smem[NUM_WAVES][16][P_ROW];
 
-  const bf16* k_slab =
-      k_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM;
+  const long slab_off = ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM;
+  const bf16* k_slab = reinterpret_cast<const bf16*>(k_cache) + slab_off;
+  const fp8_t* k_slab8 = reinterpret_cast<const fp8_t*>(k_cache) + slab_off;
   const unsigned int* v_base = reinterpret_cast<const unsigned int*>(
-      v_cache + ((long)b * num_kv_heads + kvh) * max_seq * HEAD_DIM);
+      reinterpret_cast<const bf16*>(v_cache) + slab_off);
+  const fp8_t* v_base8 = reinterpret_cast<const fp8_t*>(v_cache) + slab_off;
 
   // --- persistent Q fragments for this wave's 16 rows ---
   // A-frag: lane row = q-row (l%16), k = head-dim 8·(l/16)+i, 4 subtiles
@@ -142,9 +145,27 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
     const int kn = min(KT_POS, S - kt0);
 
     // --- stage V tile transposed (cooperative, all waves): each
-    // thread reads one dword (2 dims at one position, coalesced along
-    // the HBM row) and scatters two u16s into dim-major rows ---
-    {
+    // thread reads one dim-pair at one position (coalesced along the
+    // HBM row) and scatters two bf16 u16s into dim-major rows; the fp8
+    // path up-converts here so downstream fragments are unchanged ---
+    if constexpr (KV_FP8) {
+      const fp8_t* v_src = v_base8 + (long)kt0 * HEAD_DIM;
+      const int total = kn * (HEAD_DIM / 2);  // byte-pairs in the tile
+      for (int idx = wave * WAVE_SIZE + lane; idx < total;
+           idx += NUM_WAVES * WAVE_SIZE) {
+        const int pos = idx >> 6;          // 64 pairs per position row
+        const int d2 = idx & 63;
+        const unsigned short packed = *reinterpret_cast<const unsigned short*>(
+            v_src + (long)pos * HEAD_DIM + 2 * d2);
+        const float2_vt f = unpk2_fp8(packed);
+        const bf16 b0 = f2bf(f[0]);
+        const bf16 b1 = f2bf(f[1]);
+        v_smem_t[(2 * d2) * VT_ROW + pos] =
+            *reinterpret_cast<const unsigned short*>(&b0);
+        v_smem_t[(2 * d2 + 1) * VT_ROW + pos] =
+            *reinterpret_cast<const unsigned short*>(&b1);
+      }
+    } else {
       const unsigned int* v_src = &v_base[(long)kt0 * (HEAD_DIM / 2)];
       const int total = kn * (HEAD_DIM / 2);  // dwords in the tile
       for (int idx = wave * WAVE_SIZE + lane; idx < total;
@@ -168,12 +189,25 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
       for (int nt = 0; nt < 4; ++nt) {
         const int kpos = kt0 + 16 * nt + (lane % 16);
         const int kpos_c = kpos < max_seq ? kpos : max_seq - 1;
-        const short* krow = reinterpret_cast<const short*>(
-            k_slab + (long)kpos_c * HEAD_DIM);
 #pragma unroll
         for (int kk = 0; kk < 4; ++kk) {
-          bf16x8_frag k_frag =
-              *reinterpret_cast<const bf16x8_frag*>(krow + 32 * kk + col0);
+          bf16x8_frag k_frag;
+          if constexpr (KV_FP8) {
+            const unsigned short* src =
+                reinterpret_cast<const unsigned short*>(
+                    k_slab8 + (long)kpos_c * HEAD_DIM + 32 * kk + col0);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              const float2_vt f = unpk2_fp8(src[j]);
+              k_frag[2 * j] = pf_f2bf_bits(f[0]);
+              k_frag[2 * j + 1] = pf_f2bf_bits(f[1]);
+            }
+          } else {
+            const short* krow = reinterpret_cast<const short*>(
+                k_slab + (long)kpos_c * HEAD_DIM);
+            k_frag =
+                *reinterpret_cast<const bf16x8_frag*>(krow + 32 * kk + col0);
+          }
           s_frag[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               q_frag[kk], k_frag, s_frag[nt], 0, 0, 0);
         }
@@ -294,14 +328,32 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
   }
 }
 
+extern "C" void launch_prefill_attn_ex(
+    void* out, const void* q, const void* k_cache, const void* v_cache,
+    int batch, int seq, int num_q_heads, int num_kv_heads, int max_seq,
+    float scale, int kv_fp8, hipStream_t stream);
+
 extern "C" void launch_prefill_attn(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     int batch, int seq, int num_q_heads, int num_kv_heads, int max_seq,
     float scale, hipStream_t stream) {
+  launch_prefill_attn_ex(out, q, k_cache, v_cache, batch, seq, num_q_heads,
+                         num_kv_heads, max_seq, scale, 0, stream);
+}
+
+extern "C" void launch_prefill_attn_ex(
+    void* out, const void* q, const void* k_cache, const void* v_cache,
+    int batch, int seq, int num_q_heads, int num_kv_heads, int max_seq,
+    float scale, int kv_fp8, hipStream_t stream) {
   dim3 grid(batch * num_q_heads, (seq + QT - 1) / QT);
   dim3 block(256);
-  hipLaunchKernelGGL(prefill_attn_kernel, grid, block, 0, stream,
-                     (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
-                     (const bf16*)v_cache, batch, seq, num_q_heads,
-                     num_kv_heads, max_seq, scale);
+  if (kv_fp8) {
+    hipLaunchKernelGGL(prefill_attn_kernel<true>, grid, block, 0, stream,
+                       (bf16*)out, (const bf16*)q, k_cache, v_cache, batch,
+                       seq, num_q_heads, num_kv_heads, max_seq, scale);
+  } else {
+    hipLaunchKernelGGL(prefill_attn_kernel<false>, grid, block, 0, stream,
+                       (bf16*)out, (const bf16*)q, k_cache, v_cache, batch,
+                       seq, num_q_heads, num_kv_heads, max_seq, scale);
+  }
 }
