@@ -247,3 +247,25 @@ class TestFp8KvCpu:
         tp.reset(2, 8)
         out = tp.decode_step(torch.randint(0, TINY.vocab_size, (2,)))
         assert torch.isfinite(out.float()).all()
+
+    def test_fp8_weights_decode_close_to_bf16(self):
+        import torch
+
+        from wva_amd.calibration.model import TINY, LlamaDecodeModel
+
+        torch.manual_seed(0)
+        m_bf = LlamaDecodeModel(TINY, max_batch=2, max_seq=32, device="cpu",
+                                seed=7)
+        torch.manual_seed(0)
+        m_q = LlamaDecodeModel(TINY, max_batch=2, max_seq=32, device="cpu",
+                               seed=7, weights_dtype="fp8")
+        assert m_q.lm_head is None and m_q.lm_head_q[0].dtype == \
+            torch.float8_e4m3fn
+        m_bf.context_lens.zero_()
+        m_q.context_lens.zero_()
+        t = torch.randint(0, TINY.vocab_size, (2,))
+        for _ in range(3):
+            lb = m_bf.decode_step(t)
+            lq = m_q.decode_step(t)
+        rel = (lb.float() - lq.float()).abs().max() / lb.float().abs().max()
+        assert float(rel) < 0.25  # W8A8 per-tensor quantization noise

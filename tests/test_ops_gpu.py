@@ -548,3 +548,13 @@ class TestFp8KvGpu:
         # fp8 prefill works too (templated flash kernel)
         pre = m.prefill(torch.randint(0, TINY.vocab_size, (1, 4), device=dev))
         assert torch.isfinite(pre.float()).all()
+
+    def test_fp8_weights_decode_gpu(self, dev):
+        from wva_amd.calibration.model import TINY, LlamaDecodeModel
+
+        m = LlamaDecodeModel(TINY, max_batch=2, max_seq=32, seed=3,
+                             weights_dtype="fp8")
+        m.reset(2, 8)
+        t = torch.randint(0, TINY.vocab_size, (2,), device=dev)
+        logits = m.decode_step(t)
+        assert torch.isfinite(logits.float()).all()

@@ -91,6 +91,13 @@ TINY = LlamaConfig(
 )
 
 
+def _quantize_fp8(w: torch.Tensor):
+    """Per-tensor e4m3 quantization: returns (w_fp8 [N,K], scale)."""
+    scale = (w.abs().amax().float() / 448.0).clamp_min(1e-12)
+    w8 = (w.float() / scale).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    return w8.contiguous(), scale
+
+
 class _DecoderLayer:
     def __init__(self, cfg: LlamaConfig, device, dtype, gen):
         h, std = cfg.hidden_size, 0.02
@@ -118,10 +125,19 @@ class LlamaDecodeModel:
         device: str = "cuda",
         seed: int = 0,
         kv_dtype: str = "bf16",
+        weights_dtype: str = "bf16",
     ):
         self.cfg = cfg
         self.device = torch.device(device)
         self.dtype = torch.bfloat16
+        # fp8 weights: per-tensor e4m3 quantization, GEMMs via
+        # torch._scaled_mm (W8A8 with dynamic per-tensor activation
+        # scales) — the quantized-serving configuration (vLLM --dtype
+        # fp8 analog); halves GEMM weight streaming. Calibration-variant
+        # only: the headline bench stays bf16 weights.
+        if weights_dtype not in ("bf16", "fp8"):
+            raise ValueError(f"weights_dtype must be bf16|fp8")
+        self.weights_dtype = weights_dtype
         # fp8 (e4m3) KV storage: half the attention HBM traffic, double
         # the KV capacity; compute stays bf16 (kernels up-convert at
         # fragment build — see csrc/attention.hip KV_FP8)
@@ -149,6 +165,14 @@ class LlamaDecodeModel:
         self.lm_head = torch.empty(
             cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=self.dtype
         ).normal_(0.0, 0.02, generator=gen)
+        if weights_dtype == "fp8":
+            for layer in self.layers:
+                for attr in ("wqkv", "wo", "w_gate_up", "w_down"):
+                    setattr(layer, attr + "_q",
+                            _quantize_fp8(getattr(layer, attr)))
+                    setattr(layer, attr, None)  # free the bf16 copy
+            self.lm_head_q = _quantize_fp8(self.lm_head)
+            self.lm_head = None
 
         # Head-major contiguous KV cache per layer: [B, Hk, S, D] — each
         # (sequence, kv-head) is one sequential HBM stream for the
@@ -181,6 +205,37 @@ class LlamaDecodeModel:
                         shape, device=self.device, dtype=torch.bfloat16
                     ).to(self.cache_dtype)
 
+    def _linear(self, x: torch.Tensor, layer, name: str) -> torch.Tensor:
+        """bf16: ops.linear; fp8 weights: W8A8 torch._scaled_mm with a
+        dynamic per-tensor activation scale (rows padded to 16 — the
+        hipBLASLt fp8 path's minimum M granularity)."""
+        if self.weights_dtype == "bf16":
+            w = getattr(layer, name) if layer is not None else self.lm_head
+            return ops.linear(x, w)
+        w8, w_scale = (
+            getattr(layer, name + "_q") if layer is not None else self.lm_head_q
+        )
+        a_scale = (x.abs().amax().float() / 448.0).clamp_min(1e-12)
+        x8 = (x.float() / a_scale).clamp(-448.0, 448.0).to(
+            torch.float8_e4m3fn
+        )
+        if not x.is_cuda:  # CPU reference: dequantized matmul
+            return ((x8.float() * a_scale) @ (w8.float() * w_scale).t()).to(
+                x.dtype
+            )
+        M = x8.shape[0]
+        pad = (-M) % 16
+        if pad:
+            x8 = torch.cat(
+                [x8, torch.zeros(pad, x8.shape[1], device=x8.device,
+                                 dtype=x8.dtype)]
+            )
+        y = torch._scaled_mm(
+            x8, w8.t(), scale_a=a_scale.reshape(1),
+            scale_b=w_scale.reshape(1), out_dtype=torch.bfloat16,
+        )
+        return y[:M] if pad else y
+
     @torch.no_grad()
     def decode_step(self, token_ids: torch.Tensor) -> torch.Tensor:
         """One decode iteration for `B = len(token_ids)` sequences.
@@ -200,7 +255,7 @@ class LlamaDecodeModel:
             else:
                 h = ops.rmsnorm(x, layer.input_norm, residual, cfg.rms_eps)
 
-            qkv = ops.linear(h, layer.wqkv)  # skinny MFMA GEMM (M<=64)
+            qkv = self._linear(h, layer, "wqkv")
             # fused: RoPE on the strided qkv row + KV-cache append
             q = ops.rope_append_kv(
                 qkv, self.k_cache[li][:B], self.v_cache[li][:B],
@@ -211,15 +266,15 @@ class LlamaDecodeModel:
             attn = ops.gqa_decode_attn(
                 q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
             )
-            x = ops.linear(attn.reshape(B, cfg.q_size), layer.wo)
+            x = self._linear(attn.reshape(B, cfg.q_size), layer, "wo")
 
             h2 = ops.rmsnorm(x, layer.post_attn_norm, residual, cfg.rms_eps)
-            gate_up = ops.linear(h2, layer.w_gate_up)
+            gate_up = self._linear(h2, layer, "w_gate_up")
             act = ops.silu_mul_fused(gate_up)
-            x = ops.linear(act, layer.w_down)
+            x = self._linear(act, layer, "w_down")
 
         final = ops.rmsnorm(x, self.final_norm, residual, cfg.rms_eps)
-        logits = ops.linear(final, self.lm_head)
+        logits = self._linear(final, None, "lm_head")
         self.context_lens[:B] += 1
         return logits
 
