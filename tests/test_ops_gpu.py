@@ -558,3 +558,19 @@ class TestFp8KvGpu:
         t = torch.randint(0, TINY.vocab_size, (2,), device=dev)
         logits = m.decode_step(t)
         assert torch.isfinite(logits.float()).all()
+
+    def test_skinny_linear_fp8_matches_dequant(self, dev):
+        from wva_amd.ops import _require_ext
+
+        ext = _require_ext()
+        torch.manual_seed(2)
+        for M, N, K in [(1, 6144, 4096), (8, 4096, 4096), (16, 4096, 14336),
+                        (33, 28672, 4096)]:
+            x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+            w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.05
+            scale = (w.abs().amax(dim=1).float() / 448.0).clamp_min(1e-12)
+            w8 = ((w.float() / scale[:, None]).clamp(-448, 448)
+                  .to(torch.float8_e4m3fn).contiguous())
+            y = ext.skinny_linear_fp8(x, w8, scale.contiguous())
+            ref = x.float() @ (w8.float() * scale[:, None]).t()
+            torch.testing.assert_close(y.float(), ref, atol=0.5, rtol=2e-2)

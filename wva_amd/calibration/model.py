@@ -92,10 +92,15 @@ TINY = LlamaConfig(
 
 
 def _quantize_fp8(w: torch.Tensor):
-    """Per-tensor e4m3 quantization: returns (w_fp8 [N,K], scale)."""
-    scale = (w.abs().amax().float() / 448.0).clamp_min(1e-12)
-    w8 = (w.float() / scale).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
-    return w8.contiguous(), scale
+    """Per-channel (rowwise) e4m3 quantization: (w_fp8 [N,K], scale [N]).
+    Row scales commute with the GEMM's k-sum, so they fold into the
+    output store exactly (csrc/skinny_gemm.hip W_FP8)."""
+    scale = (w.abs().amax(dim=1).float() / 448.0).clamp_min(1e-12)  # [N]
+    w8 = (
+        (w.float() / scale[:, None]).clamp(-448.0, 448.0)
+        .to(torch.float8_e4m3fn)
+    )
+    return w8.contiguous(), scale.contiguous()
 
 
 class _DecoderLayer:
@@ -206,35 +211,28 @@ class LlamaDecodeModel:
                     ).to(self.cache_dtype)
 
     def _linear(self, x: torch.Tensor, layer, name: str) -> torch.Tensor:
-        """bf16: ops.linear; fp8 weights: W8A8 torch._scaled_mm with a
-        dynamic per-tensor activation scale (rows padded to 16 — the
-        hipBLASLt fp8 path's minimum M granularity)."""
+        """bf16: ops.linear (hipBLASLt); fp8 weights: weight-only W8A16
+        on the in-tree streaming kernel (activations stay bf16; weights
+        up-convert in-fragment; per-channel scales fold into the store).
+        W8A8 via torch._scaled_mm measured 2.4× slower at decode sizes
+        (dynamic-quant launch overhead — docs/mi355x-kernels.md), so the
+        weight-only form is the fp8 GEMM path."""
         if self.weights_dtype == "bf16":
             w = getattr(layer, name) if layer is not None else self.lm_head
             return ops.linear(x, w)
         w8, w_scale = (
             getattr(layer, name + "_q") if layer is not None else self.lm_head_q
         )
-        a_scale = (x.abs().amax().float() / 448.0).clamp_min(1e-12)
-        x8 = (x.float() / a_scale).clamp(-448.0, 448.0).to(
-            torch.float8_e4m3fn
-        )
         if not x.is_cuda:  # CPU reference: dequantized matmul
-            return ((x8.float() * a_scale) @ (w8.float() * w_scale).t()).to(
-                x.dtype
-            )
-        M = x8.shape[0]
-        pad = (-M) % 16
-        if pad:
-            x8 = torch.cat(
-                [x8, torch.zeros(pad, x8.shape[1], device=x8.device,
-                                 dtype=x8.dtype)]
-            )
-        y = torch._scaled_mm(
-            x8, w8.t(), scale_a=a_scale.reshape(1),
-            scale_b=w_scale.reshape(1), out_dtype=torch.bfloat16,
-        )
-        return y[:M] if pad else y
+            return (
+                x.float() @ (w8.float() * w_scale[:, None]).t()
+            ).to(x.dtype)
+        if x.shape[0] <= 64 and x.shape[1] % 128 == 0:
+            from .. import ops as _ops
+
+            return _ops._require_ext().skinny_linear_fp8(x, w8, w_scale)
+        # large-M fallback (prefill-scale): dequantize and use hipBLASLt
+        return (x @ (w8.to(x.dtype) * w_scale[:, None].to(x.dtype)).t())
 
     @torch.no_grad()
     def decode_step(self, token_ids: torch.Tensor) -> torch.Tensor:
@@ -293,6 +291,12 @@ class LlamaDecodeModel:
         row-shaped and batch-size-agnostic).
         """
         cfg = self.cfg
+        if self.weights_dtype != "bf16":
+            raise NotImplementedError(
+                "prefill with fp8 weights is not implemented (the "
+                "weight-only fp8 mode targets decode; prefill would "
+                "dequantize every layer per pass)"
+            )
         B, S = token_ids.shape
         T = B * S
         if S > self.max_seq:

@@ -63,6 +63,10 @@ extern "C" int skinny_gemm_tile_n(int M);
 extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
                                    const void* w, int M, int N, int K,
                                    int num_splits, hipStream_t stream);
+extern "C" void launch_skinny_gemm_ex(void* c, void* ws, const void* a,
+                                      const void* w, const float* w_scale,
+                                      int M, int N, int K, int num_splits,
+                                      int w_fp8, hipStream_t stream);
 extern "C" void launch_gqa_decode_attn(void* out, void* workspace,
                                        const void* q, const void* k_cache,
                                        const void* v_cache,
@@ -317,6 +321,41 @@ torch::Tensor skinny_linear(torch::Tensor x, torch::Tensor w) {
   return y;
 }
 
+// Weight-only fp8 (W8A16): w is [N, K] e4m3 with per-channel scales [N];
+// activations stay bf16, dequantization happens in-fragment and the
+// scale folds into the store (csrc/skinny_gemm.hip W_FP8).
+torch::Tensor skinny_linear_fp8(torch::Tensor x, torch::Tensor w,
+                                torch::Tensor w_scale) {
+  check_bf16_contig(x, "x");
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous() &&
+                  w.scalar_type() == at::kFloat8_e4m3fn,
+              "w must be contiguous float8_e4m3fn");
+  TORCH_CHECK(w_scale.is_cuda() && w_scale.scalar_type() == torch::kFloat &&
+                  w_scale.numel() == w.size(0),
+              "w_scale must be float32 [N]");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2, "x [M,K], w [N,K]");
+  const int M = x.size(0);
+  const int K = x.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  TORCH_CHECK(M <= 64, "skinny_linear requires M <= 64");
+  TORCH_CHECK(K % 128 == 0, "K must be a multiple of 128");
+  auto y = torch::empty({M, N}, x.options());
+  const int nt = skinny_gemm_tile_n(M);
+  const int sk = skinny_gemm_num_splits(N, K, nt);
+  torch::Tensor ws;
+  void* ws_ptr = nullptr;
+  if (sk > 1) {
+    ws = torch::empty({(long)sk * M * N},
+                      x.options().dtype(torch::kFloat32));
+    ws_ptr = ws.data_ptr();
+  }
+  launch_skinny_gemm_ex(y.data_ptr(), ws_ptr, x.data_ptr(), w.data_ptr(),
+                        w_scale.data_ptr<float>(), M, N, K, sk, 1,
+                        current_stream());
+  return y;
+}
+
 // Causal flash-attention prefill over the (already-populated) head-major
 // KV caches (csrc/prefill_attention.hip). q/out: [B*S, Hq, 128].
 torch::Tensor prefill_attn(torch::Tensor q, torch::Tensor k_cache,
@@ -374,6 +413,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Causal flash-attention prefill over head-major KV caches",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("batch"), py::arg("seq"), py::arg("scale"));
+  m.def("skinny_linear_fp8", &skinny_linear_fp8,
+        "Weight-only fp8 decode GEMM: x[M,K]bf16 @ w[N,K]e4m3^T, M <= 64",
+        py::arg("x"), py::arg("w"), py::arg("w_scale"));
   m.def("skinny_linear", &skinny_linear,
         "Weight-streaming decode GEMM: x[M,K] @ w[N,K]^T, M <= 64",
         py::arg("x"), py::arg("w"));

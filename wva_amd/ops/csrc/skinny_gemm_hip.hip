@@ -39,12 +39,17 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_frag;
 // NSUB = n-subtiles (16-row groups) per wave: 1 for M <= 16, 2 for
 // larger M — the second subtile reuses the same A fragments, halving
 // the L2 A-read : HBM W-read ratio that bounds the M=64 case.
-template <int NSUB>
+// W_FP8: weight-only e4m3 quantization (W8A16): B fragments up-convert
+// 8 weight bytes per load (halving the HBM stream this kernel is bound
+// by); per-channel scales are folded into the output store, which is
+// exact because a row scale commutes with the k-sum.
+template <int NSUB, bool W_FP8>
 __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
     bf16* __restrict__ c,          // [M, N] (null when split-K)
     float* __restrict__ ws,        // [SK, M, N] fp32 partials (or null)
     const bf16* __restrict__ a,    // [M, K]
-    const bf16* __restrict__ w,    // [N, K] row-major (torch linear weight)
+    const void* __restrict__ w,    // [N, K] row-major, bf16 or e4m3
+    const float* __restrict__ w_scale,  // [N] per-channel (W_FP8 only)
     const int M,
     const int N,
     const int K) {
@@ -57,12 +62,15 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
   const int wg_rows = NT * NSUB;
   const int n_row0 = blockIdx.x * wg_rows + 16 * NSUB * wave + (lane % 16);
   const bf16* w_rows[NSUB];
+  const fp8_t* w_rows8[NSUB];
   bool n_live[NSUB];
 #pragma unroll
   for (int ns = 0; ns < NSUB; ++ns) {
     const int n_row = n_row0 + 16 * ns;
     n_live[ns] = n_row < N;
-    w_rows[ns] = w + (long)(n_live[ns] ? n_row : 0) * K;
+    const long off = (long)(n_live[ns] ? n_row : 0) * K;
+    w_rows[ns] = reinterpret_cast<const bf16*>(w) + off;
+    w_rows8[ns] = reinterpret_cast<const fp8_t*>(w) + off;
   }
 
   const int mt = (M + 15) / 16;          // ≤ 4 m-tiles
@@ -89,7 +97,21 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
 #pragma unroll
       for (int ns = 0; ns < NSUB; ++ns) {
         if (n_live[ns]) {
-          b_frag[ns] = *reinterpret_cast<const bf16x8_frag*>(w_rows[ns] + k);
+          if constexpr (W_FP8) {
+            const unsigned short* src =
+                reinterpret_cast<const unsigned short*>(w_rows8[ns] + k);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              const float2_vt f = unpk2_fp8(src[j]);
+              const bf16 b0 = f2bf(f[0]);
+              const bf16 b1 = f2bf(f[1]);
+              b_frag[ns][2 * j] = *reinterpret_cast<const short*>(&b0);
+              b_frag[ns][2 * j + 1] = *reinterpret_cast<const short*>(&b1);
+            }
+          } else {
+            b_frag[ns] =
+                *reinterpret_cast<const bf16x8_frag*>(w_rows[ns] + k);
+          }
         } else {
 #pragma unroll
           for (int i = 0; i < 8; ++i) b_frag[ns][i] = 0;
@@ -124,14 +146,15 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
 #pragma unroll
     for (int pm = 0; pm < 4; ++pm) {
       if (pm >= mt) break;
+      const float scl = W_FP8 ? w_scale[n_out] : 1.0f;
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int m = pm * 16 + (lane >> 4) * 4 + i;
         if (m >= M) continue;
         if (ws != nullptr) {
-          ws[((long)sk * M + m) * N + n_out] = acc[pm][ns][i];
+          ws[((long)sk * M + m) * N + n_out] = acc[pm][ns][i] * scl;
         } else {
-          c[(long)m * N + n_out] = f2bf(acc[pm][ns][i]);
+          c[(long)m * N + n_out] = f2bf(acc[pm][ns][i] * scl);
         }
       }
     }
@@ -161,20 +184,35 @@ extern "C" int skinny_gemm_num_splits(int N, int K, int nt) {
   return sk;
 }
 
-extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
-                                   const void* w, int M, int N, int K,
-                                   int num_splits, hipStream_t stream) {
+extern "C" void launch_skinny_gemm_ex(void* c, void* ws, const void* a,
+                                      const void* w, const float* w_scale,
+                                      int M, int N, int K, int num_splits,
+                                      int w_fp8, hipStream_t stream) {
   const int nt = (M > 16) ? 2 * NT : NT;
   dim3 grid((N + nt - 1) / nt, num_splits);
   dim3 block(256);
   float* ws_ptr = num_splits > 1 ? (float*)ws : nullptr;
   bf16* c_ptr = num_splits > 1 ? nullptr : (bf16*)c;
   if (nt == NT) {
-    hipLaunchKernelGGL(skinny_gemm_kernel<1>, grid, block, 0, stream, c_ptr,
-                       ws_ptr, (const bf16*)a, (const bf16*)w, M, N, K);
+    if (w_fp8) {
+      hipLaunchKernelGGL((skinny_gemm_kernel<1, true>), grid, block, 0,
+                         stream, c_ptr, ws_ptr, (const bf16*)a, w, w_scale,
+                         M, N, K);
+    } else {
+      hipLaunchKernelGGL((skinny_gemm_kernel<1, false>), grid, block, 0,
+                         stream, c_ptr, ws_ptr, (const bf16*)a, w, w_scale,
+                         M, N, K);
+    }
   } else {
-    hipLaunchKernelGGL(skinny_gemm_kernel<2>, grid, block, 0, stream, c_ptr,
-                       ws_ptr, (const bf16*)a, (const bf16*)w, M, N, K);
+    if (w_fp8) {
+      hipLaunchKernelGGL((skinny_gemm_kernel<2, true>), grid, block, 0,
+                         stream, c_ptr, ws_ptr, (const bf16*)a, w, w_scale,
+                         M, N, K);
+    } else {
+      hipLaunchKernelGGL((skinny_gemm_kernel<2, false>), grid, block, 0,
+                         stream, c_ptr, ws_ptr, (const bf16*)a, w, w_scale,
+                         M, N, K);
+    }
   }
   if (num_splits > 1) {
     const long MN = (long)M * N;
@@ -182,6 +220,13 @@ extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
     hipLaunchKernelGGL(skinny_gemm_merge_kernel, mgrid, block, 0, stream,
                        (bf16*)c, (const float*)ws, num_splits, MN);
   }
+}
+
+extern "C" void launch_skinny_gemm(void* c, void* ws, const void* a,
+                                   const void* w, int M, int N, int K,
+                                   int num_splits, hipStream_t stream) {
+  launch_skinny_gemm_ex(c, ws, a, w, nullptr, M, N, K, num_splits, 0,
+                        stream);
 }
 
 extern "C" int skinny_gemm_tile_n(int M) { return (M > 16) ? 2 * NT : NT; }
