@@ -123,8 +123,9 @@ def build_scenario(profile, seed: int):
     # whole optimization interval (improvement over reference parity 1.0)
     service_time_s = OUTPUT_TOKENS * profile.itl_ms(profile.max_num_seqs) / 1000.0
     drain = min(1.0, service_time_s / ENGINE_INTERVAL_S)
+    analyzer = os.environ.get("WVA_BENCH_ANALYZER", "saturation")
     sat = {
-        "analyzerName": "saturation",
+        "analyzerName": analyzer,
         "schedulerQueueDrainFactor": drain,
     }
     if os.environ.get("WVA_BENCH_LEAD"):
@@ -136,6 +137,44 @@ def build_scenario(profile, seed: int):
         cluster, config, source=source,
         metrics_registry=CollectorRegistry(), start_engines=False,
     )
+    if analyzer == "inferno":
+        # rate-based SLO sizing (M/M/1/K queueing on the measured service
+        # parms): immune to the token-demand analyzer's post-drain
+        # undershoot limit cycle (docs/saturation-analyzer.md) because
+        # the sizing input is the arrival rate, which is
+        # provisioning-invariant
+        from wva_amd.analyzers.modelanalyzer import InfernoAnalyzer
+        from wva_amd.inferno.system import System
+        from wva_amd.inferno.types import (
+            AcceleratorSpec,
+            ModelAcceleratorPerfData,
+            ModelTarget,
+            ServiceClassSpec,
+            ServiceParmsSpec,
+            SystemData,
+        )
+
+        system = System(SystemData(
+            accelerators=[
+                AcceleratorSpec(name="MI355X", type="MI355X", cost=50)
+            ],
+            models=[ModelAcceleratorPerfData(
+                name=MODEL_ID, acc="MI355X",
+                max_batch_size=profile.max_num_seqs,
+                at_tokens=OUTPUT_TOKENS,
+                service_parms=ServiceParmsSpec(
+                    alpha=profile.alpha_ms, beta=profile.beta_ms
+                ),
+            )],
+            service_classes=[ServiceClassSpec(
+                name="default", priority=1,
+                model_targets=[ModelTarget(
+                    model=MODEL_ID, slo_itl=SLO_ITL_MS,
+                    slo_ttft=SLO_TTFT_MS,
+                )],
+            )],
+        ))
+        app.saturation_engine.inferno_analyzer = InfernoAnalyzer(system)
     return cluster, sim, app
 
 
