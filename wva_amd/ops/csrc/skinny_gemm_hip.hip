@@ -92,7 +92,11 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
   for (int k0 = k_begin; k0 < k_end; k0 += KT) {
 #pragma unroll
     for (int kk = 0; kk < KT / 32; ++kk) {
-      const int k = k0 + kk * 32 + col0;
+      // lane→k bijection chosen for CONTIGUITY: lane's four k-subtile
+      // windows are adjacent 8-element runs (32 B fp8 / 64 B bf16 per
+      // lane per K-block), so the unrolled loads merge into wide ones.
+      // Any bijection shared by A and B is exact (probe-verified).
+      const int k = k0 + 4 * col0 + 8 * kk;
       bf16x8_frag b_frag[NSUB];
 #pragma unroll
       for (int ns = 0; ns < NSUB; ++ns) {
