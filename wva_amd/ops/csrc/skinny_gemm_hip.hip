@@ -89,6 +89,7 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm_kernel(
     for (int ns = 0; ns < NSUB; ++ns)
       acc[p][ns] = f32x4_frag{0.f, 0.f, 0.f, 0.f};
 
+#pragma unroll 2
   for (int k0 = k_begin; k0 < k_end; k0 += KT) {
 #pragma unroll
     for (int kk = 0; kk < KT / 32; ++kk) {
