@@ -1,34 +1,29 @@
 // GQA decode attention (single query token per sequence) over a
-// contiguous KV cache — MI355X (gfx950).
+// contiguous head-major KV cache — MI355X (gfx950).
 //
 // Shape contract (Llama-3 family): head_dim = 128, G = Hq/Hk ≤ 8 query
-// heads share one KV head. One 4-wave workgroup per (batch, kv_head).
+// heads share one KV head; caches [B, Hk, S_max, 128] in bf16 or fp8
+// e4m3 (KV_FP8 template — up-converted at fragment build). One 4-wave
+// workgroup per (batch, kv_head, split).
 //
-// Structure (v3 — shared tile, head-per-wave):
-//   * Each iteration stages ONE shared 64-position K/V tile into LDS
-//     cooperatively (all 4 waves, coalesced row loads), rows padded to 65
-//     dwords so every position-per-lane sweep is bank-conflict-free
-//     (bank = (65·row + d) % 32 = (row + d) % 32).
-//   * Query heads are OWNED by waves (head g → wave g % 4): each wave
-//     runs the full online softmax for its heads, so there is no
-//     cross-wave flash-decoding merge at all — the owning wave writes the
-//     final output directly.
-//   * Scores: lane ↔ position (full 64-lane utilization), K swept with
-//     ds_read_b128; ONE max+sum wave-reduction per (tile, head) — the v1
-//     kernel did 6-shuffle reductions per position per head and ran at
-//     ~10% of HBM speed (profiles/decode8b_b64_ctx512_kernel_stats.txt).
-//   * PV: p staged to LDS (broadcast reads), lane ↔ output dword.
-// fp32 softmax and accumulation throughout.
+// Three kernel generations, all kept (measured dispatch in ops.cpp):
+//   v3  shared-LDS-tile, VALU scores + VALU PV (head-per-wave softmax
+//       ownership; regression baseline)
+//   v4  MFMA scores: K streams HBM→VGPR fragments (no K staging), V via
+//       LDS, VALU PV — best for G < 8 with split-KV
+//   v5  v4 + MFMA PV (P cast to bf16, corr/s broadcast via LDS) — best
+//       for G = 8 or unsplit grids
+// plus split-KV partial/merge kernels shared by all three. The
+// optimization ladder with per-step measurements (13.04 → 6.26 ms for
+// the 8B decode step) and the negative results along the way are in
+// docs/mi355x-kernels.md; MFMA fragment conventions were probe-verified
+// on hardware (scripts/mfma_probe.hip): identical lane→k bijections for
+// A and B suffice, C/D is col = lane&15, row = (lane>>4)·4 + reg.
 //
-// LDS: K+V tile 2×64×260 B ≈ 33 KiB + q 4 KiB + p 2 KiB ≈ 39 KiB/WG
-// → 4 workgroups/CU admissible; grid B×Hk.
+// LDS notes: 66-dword row stride (64 + 2 pad) keeps every b64 access
+// aligned AND bank-conflict-free; a 65-dword pad is conflict-free but
+// leaves odd rows 4-byte-aligned, silently splitting vector LDS ops.
 //
-// Layouts: q [B, Hq, 128], k/v cache [B, Hk, S_max, 128], out [B, Hq, 128].
-// The head-major cache layout makes each (b, kvh)'s K and V tiles fully
-// CONTIGUOUS in HBM (a 64-position tile = one 16 KiB sequential stream),
-// where the token-major [B, S, Hk, D] layout fetched 256 B rows at a
-// 2 KiB stride — head-major measured ~1.3x on the staging-bound kernel.
-
 #include "common.h"
 
 #define HEAD_DIM 128
