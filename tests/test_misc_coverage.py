@@ -470,3 +470,15 @@ class TestCliFlags:
         setup_logging(4)
         assert get_logger("x").getEffectiveLevel() <= logging.DEBUG
         setup_logging(2)
+
+
+class TestMetricsTlsConfig:
+    def test_loader_maps_cert_envs(self, monkeypatch):
+        from wva_amd.config.loader import load_config
+
+        monkeypatch.setenv("METRICS_CERT_PATH", "/certs/tls.crt")
+        monkeypatch.setenv("METRICS_KEY_PATH", "/certs/tls.key")
+        monkeypatch.setenv("PROMETHEUS_BASE_URL", "http://p:9090")
+        cfg = load_config()
+        assert cfg.infra.metrics_cert_path == "/certs/tls.crt"
+        assert cfg.infra.metrics_key_path == "/certs/tls.key"

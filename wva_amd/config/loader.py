@@ -27,6 +27,8 @@ _DEFAULTS: Dict[str, Any] = {
     "LEADER_ELECTION_RETRY_PERIOD": "10s",
     "REST_CLIENT_TIMEOUT": "60s",
     "METRICS_SECURE": True,
+    "METRICS_CERT_PATH": "",
+    "METRICS_KEY_PATH": "",
     "ENABLE_HTTP2": False,
     "WATCH_NAMESPACE": "",
     "V": 0,
@@ -112,6 +114,8 @@ def load_config(
         merged["REST_CLIENT_TIMEOUT"]
     )
     infra.secure_metrics = _coerce_bool(merged["METRICS_SECURE"])
+    infra.metrics_cert_path = str(merged["METRICS_CERT_PATH"])
+    infra.metrics_key_path = str(merged["METRICS_KEY_PATH"])
     infra.enable_http2 = _coerce_bool(merged["ENABLE_HTTP2"])
     infra.watch_namespace = str(merged["WATCH_NAMESPACE"])
     infra.logger_verbosity = int(merged["V"])
