@@ -536,24 +536,54 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
   __shared__ float sum_smem[MAX_G];
   __shared__ float m_smem[MAX_G];
 
-  // Q fragments (same as v4)
+  // Q fragments (same as v4; fp8 mode pre-quantizes once per WG)
   bf16x8_frag q_frag[4];
+  long q8_frag[4] = {0, 0, 0, 0};
+  float q_scale = 1.0f;
   {
     const int g = lane % 16;
     const int k0 = 8 * (lane / 16);
+    float qv[4][8];
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+      for (int i = 0; i < 8; ++i) qv[kk][i] = 0.0f;
     if (g < G) {
       const bf16* qrow = q + ((long)b * num_q_heads + kvh * G + g) * HEAD_DIM;
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         const short* src = reinterpret_cast<const short*>(qrow + 32 * kk + k0);
 #pragma unroll
-        for (int i = 0; i < 8; ++i) q_frag[kk][i] = src[i];
+        for (int i = 0; i < 8; ++i) {
+          q_frag[kk][i] = src[i];
+          const bf16 bv = *reinterpret_cast<const bf16*>(&src[i]);
+          qv[kk][i] = bf2f(bv);
+        }
       }
     } else {
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk)
 #pragma unroll
         for (int i = 0; i < 8; ++i) q_frag[kk][i] = 0;
+    }
+    if constexpr (KV_FP8) {
+      float amax = 0.0f;
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk)
+#pragma unroll
+        for (int i = 0; i < 8; ++i) amax = fmaxf(amax, fabsf(qv[kk][i]));
+      amax = wave_reduce_max(amax);
+      __shared__ float qmax_smem5[NUM_WAVES];
+      if (lane == 0) qmax_smem5[wave] = amax;
+      __syncthreads();
+      float m0 = qmax_smem5[0];
+#pragma unroll
+      for (int w2 = 1; w2 < NUM_WAVES; ++w2) m0 = fmaxf(m0, qmax_smem5[w2]);
+      q_scale = fmaxf(m0 / 448.0f, 1e-12f);
+      const float inv = 1.0f / q_scale;
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) q8_frag[kk] = pack8_fp8(qv[kk], inv);
+      __syncthreads();
     }
   }
 
@@ -619,33 +649,35 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
       const int pos_c = pos < max_seq ? pos : max_seq - 1;
       const int k0 = 8 * (lane / 16);
       f32x4_frag sf = {0.f, 0.f, 0.f, 0.f};
+      if constexpr (KV_FP8) {
+        const long* ksrc = reinterpret_cast<const long*>(
+            k_slab8 + (long)pos_c * HEAD_DIM);
+        const int l0 = k0 / 8;
 #pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
-        bf16x8_frag a_frag;
-        if constexpr (KV_FP8) {
-          const unsigned short* src = reinterpret_cast<const unsigned short*>(
-              k_slab8 + (long)pos_c * HEAD_DIM + 32 * kk + k0);
+        for (int kk = 0; kk < 4; ++kk) {
+          const long a8 = ksrc[kk * 4 + l0];
+          sf = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              a8, q8_frag[kk], sf, 0, 0, 0);
+        }
+      } else {
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            const float2_vt f = unpk2_fp8(src[j]);
-            a_frag[2 * j] = f2bf_bits(f[0]);
-            a_frag[2 * j + 1] = f2bf_bits(f[1]);
-          }
-        } else {
+        for (int kk = 0; kk < 4; ++kk) {
+          bf16x8_frag a_frag;
           const short* src = reinterpret_cast<const short*>(
               k_slab + (long)pos_c * HEAD_DIM) + 32 * kk + k0;
 #pragma unroll
           for (int i = 0; i < 8; ++i) a_frag[i] = src[i];
+          sf = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, q_frag[kk],
+                                                       sf, 0, 0, 0);
         }
-        sf = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, q_frag[kk], sf,
-                                                     0, 0, 0);
       }
+      const float s_eff = KV_FP8 ? scale * q_scale : scale;
       const int g = lane & 15;
       if (g < G) {
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
           const int t = 16 * wave + (lane >> 4) * 4 + i;
-          s_smem[g][t] = (t < tn) ? sf[i] * scale : -INFINITY;
+          s_smem[g][t] = (t < tn) ? sf[i] * s_eff : -INFINITY;
         }
       }
     }
