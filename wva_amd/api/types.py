@@ -252,11 +252,19 @@ class VariantAutoscalingStatus:
     conditions: List[Condition] = field(default_factory=list)
 
     def to_dict(self) -> Dict[str, Any]:
-        return {
-            "desiredOptimizedAlloc": self.desired_optimized_alloc.to_dict(),
-            "actuation": self.actuation.to_dict(),
-            "conditions": [c.to_dict() for c in self.conditions],
-        }
+        d: Dict[str, Any] = {"actuation": self.actuation.to_dict()}
+        alloc = self.desired_optimized_alloc
+        # omit the zero-valued alloc: a real stored CR has no
+        # desiredOptimizedAlloc until the first valid status write (the
+        # CRD requires accelerator minLength=2, so the zero struct could
+        # never be stored), and the reference's #731 patch-base trick
+        # exists precisely to exclude the zero struct from merge patches
+        # (variantautoscaling_controller.go:244-252)
+        if alloc.accelerator or alloc.num_replicas or alloc.last_run_time:
+            d["desiredOptimizedAlloc"] = alloc.to_dict()
+        if self.conditions:
+            d["conditions"] = [c.to_dict() for c in self.conditions]
+        return d
 
     @classmethod
     def from_dict(cls, d: Dict[str, Any]) -> "VariantAutoscalingStatus":

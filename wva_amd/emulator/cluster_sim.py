@@ -155,10 +155,15 @@ class ClusterSim:
                 pod = self.cluster.try_get("Pod", ns, pod_name)
                 if pod is not None and pod.status.ready != is_ready:
                     pod.status.ready = is_ready
+                    # blind write (no resourceVersion precondition): the sim
+                    # plays kubelet/controller-manager and must not conflict
+                    # with concurrent WVA writes through the API surface
+                    pod.metadata.resource_version = 0
                     self.cluster.update(pod)
             deploy.status.replicas = len(pods)
             deploy.status.ready_replicas = ready
             deploy.status.available_replicas = ready
+            deploy.metadata.resource_version = 0
             self.cluster.update(deploy)
 
     def _pick_node(self) -> str:

@@ -537,8 +537,20 @@ class SaturationEngine:
         self,
         decisions: List[VariantDecision],
         va_map: Dict[str, VariantAutoscaling],
+        current_allocations: Optional[Dict[str, tuple]] = None,
     ) -> None:
+        """Apply decisions to VA status + decision cache.
+
+        ``current_allocations`` maps ``ns/name`` -> ``(num_replicas,
+        accelerator)`` and mirrors the reference's ``currentAllocations``
+        fallback on the no-decision path (engine.go:861-880).  NOTE: the
+        reference declares that map but never writes to it (verified by
+        grep over engine.go/engine_v2.go — no assignment exists), so the
+        fallback is structurally present here for parity but, like the
+        reference, fires only if a caller supplies allocations.
+        """
         decision_map = {f"{d.namespace}/{d.variant_name}": d for d in decisions}
+        current_allocations = current_allocations or {}
 
         for va_key, va in va_map.items():
             decision = decision_map.get(va_key)
@@ -555,12 +567,21 @@ class SaturationEngine:
                 accelerator_name = decision.accelerator_name
                 reason = decision.reason
             else:
-                target_replicas = (
-                    update_va.status.desired_optimized_alloc.num_replicas
-                    if update_va.status.desired_optimized_alloc.num_replicas > 0
-                    else 0
-                )
+                # No decision: keep the previously desired allocation, then
+                # fall back to the current allocation if one was collected
+                # (engine.go:866-880; see docstring re: reference parity).
+                curr = current_allocations.get(va_key)
+                if update_va.status.desired_optimized_alloc.num_replicas > 0:
+                    target_replicas = (
+                        update_va.status.desired_optimized_alloc.num_replicas
+                    )
+                elif curr is not None and curr[0] > 0:
+                    target_replicas = curr[0]
+                else:
+                    target_replicas = 0
                 accelerator_name = update_va.status.desired_optimized_alloc.accelerator
+                if not accelerator_name and curr is not None:
+                    accelerator_name = curr[1]
                 reason = "No scaling decision (optimization loop)"
 
             if decision is not None and not accelerator_name:

@@ -164,10 +164,26 @@ class FakeCluster:
             return out
 
     def update(self, obj: Any, bump_generation: bool = False) -> Any:
+        """PUT semantics with optimistic concurrency: a nonzero
+        metadata.resourceVersion on the incoming object is a precondition
+        — mismatch raises ConflictError (HTTP 409 on the REST surface),
+        exactly as kube-apiserver rejects stale writes. An unset/zero
+        resourceVersion is a blind write (no precondition), matching the
+        API server's behavior for PUT without resourceVersion."""
         with self._lock:
             key = self._key(obj)
-            if key not in self._objects:
+            existing = self._objects.get(key)
+            if existing is None:
                 raise NotFoundError(*key)
+            incoming_rv = getattr(_meta(obj), "resource_version", 0) or 0
+            stored_rv = _meta(existing).resource_version
+            if incoming_rv and incoming_rv != stored_rv:
+                raise ConflictError(
+                    f"Operation cannot be fulfilled on {key}: the object "
+                    f"has been modified (resourceVersion {incoming_rv} != "
+                    f"{stored_rv}); please apply your changes to the "
+                    f"latest version and try again"
+                )
             stored = _clone(obj)
             self._rv += 1
             m = _meta(stored)

@@ -307,6 +307,9 @@ class _WatchPump(threading.Thread):
                 )
         except Exception as e:  # noqa: BLE001 — keep pumping
             log.warning("watch list %s failed: %s", self.kind, e)
+        # initial-list-complete marker: CachedCluster.wait_for_sync()
+        # (the WaitForCacheSync analog) blocks on one SYNC per kind
+        self.out.put(WatchEvent("SYNC", self.kind, None))
 
         while not self._stop.is_set():
             try:

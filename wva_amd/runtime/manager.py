@@ -89,9 +89,16 @@ class LeaderElector:
                 return True
             except Exception:  # noqa: BLE001 — lost the race
                 return False
+        from ..kube.fake import ConflictError
+
         if lease.holder_identity == self.identity:
             lease.renew_time = now
-            self.cluster.update(lease)
+            try:
+                self.cluster.update(lease)
+            except ConflictError:
+                # a competitor wrote between our read and renew — the
+                # optimistic-concurrency loss IS the election loss
+                return False
             return True
         expired = (
             lease.renew_time is None
@@ -101,7 +108,10 @@ class LeaderElector:
             lease.holder_identity = self.identity
             lease.acquire_time = now
             lease.renew_time = now
-            self.cluster.update(lease)
+            try:
+                self.cluster.update(lease)
+            except ConflictError:
+                return False
             return True
         return False
 
@@ -260,6 +270,8 @@ class Manager:
                 continue
             if event is None:
                 return
+            if event.type not in ("ADDED", "MODIFIED", "DELETED"):
+                continue  # SYNC/BOOKMARK markers from cache/REST layers
             for r in self._registrations:
                 if event.kind not in r.kinds:
                     continue
