@@ -387,6 +387,10 @@ class K8sTestServer:
             self._fail_remaining = n
             self._fail_code = code
 
+    def active_watch_count(self) -> int:
+        with self._watch_lock:
+            return len(self._watch_drops)
+
     def drop_watches(self) -> int:
         """Force-close every active watch stream; returns how many were
         dropped. Clients must transparently reconnect from their last

@@ -97,6 +97,40 @@ class CachedCluster:
                 lambda: all(self._synced.values()), timeout=timeout
             )
 
+    def wait_caught_up(self, timeout: float = 10.0) -> bool:
+        """Test/bench barrier: block until the store matches the
+        upstream cluster's current contents for every cached kind
+        (same key set, resourceVersions at least as new). Only
+        meaningful while upstream is quiescent; costs one LIST per
+        kind per poll, so not for production paths."""
+        import time as _time
+
+        deadline = _time.monotonic() + timeout
+        while _time.monotonic() < deadline:
+            if self._caught_up():
+                return True
+            _time.sleep(0.02)
+        return self._caught_up()
+
+    def _caught_up(self) -> bool:
+        for kind in self.kinds:
+            upstream = {
+                (o.metadata.namespace, o.metadata.name):
+                    o.metadata.resource_version or 0
+                for o in self.cluster.list(kind)
+            }
+            with self._lock:
+                mine = {
+                    (ns, n): o.metadata.resource_version or 0
+                    for (k, ns, n), o in self._store.items()
+                    if k == kind
+                }
+            if set(upstream) != set(mine):
+                return False
+            if any(mine[key] < rv for key, rv in upstream.items()):
+                return False
+        return True
+
     # --- pump ---
 
     def _pump_loop(self) -> None:
