@@ -50,3 +50,23 @@ def test_tp2_gloo_decode():
     )
     assert out.returncode == 0, out.stderr[-2000:]
     assert "TP2_OK" in out.stdout
+
+
+@pytest.mark.timeout(300)
+def test_tp2_numerics_matches_tp1_gloo():
+    """TP=2 sharded decode equals the unsharded reference (weights
+    reassembled from the shard generator streams) — the CPU/gloo twin of
+    tests/test_tp_gpu.py::test_tp2_decode_matches_tp1, so the GPU
+    numerics check is validated logic, not dead code."""
+    env = dict(os.environ, WVA_TP_DEVICE="cpu", WVA_TP_BACKEND="gloo")
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29543",
+            os.path.join(REPO, "tests", "tp_numerics_worker.py"),
+        ],
+        capture_output=True, text=True, timeout=280, env=env,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "TP2_NUMERICS_OK" in out.stdout

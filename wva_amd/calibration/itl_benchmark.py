@@ -38,6 +38,10 @@ class CalibrationResult:
     block_size: int
     decode_tokens_per_s_peak: float
     prefill_tokens_per_s: float = 0.0
+    # GPU-busy marker phase (driver-verifiable evidence that the
+    # calibration really ran on the device: long enough for SMI sampling)
+    busy_marker_s: float = 0.0
+    busy_tokens: int = 0
 
     def to_json(self) -> str:
         return json.dumps(asdict(self), indent=2)
@@ -152,6 +156,7 @@ def calibrate_service_profile(
     iters: int = 8,
     use_graph: bool = False,
     kv_dtype: str = "bf16",
+    busy_seconds: float = 0.0,
 ) -> Tuple[ServiceProfile, CalibrationResult]:
     """Measure on the current GPU and return an emulator ServiceProfile
     + the raw calibration record. kv_dtype="fp8" measures the e4m3
@@ -182,6 +187,25 @@ def calibrate_service_profile(
     peak_tps = max(
         b / (t / 1000.0) for b, t in zip(batch_sizes, itl)
     )
+    # GPU-busy marker: keep the device saturated with real decode steps
+    # for a fixed wall-clock window so out-of-process SMI sampling (the
+    # bench driver's gpu_busy probe) cannot miss the GPU phase
+    busy_tokens = 0
+    busy_elapsed = 0.0
+    if busy_seconds > 0:
+        b = max_batch
+        model.reset(b, context_len)
+        toks = torch.randint(0, cfg.vocab_size, (b,), device=model.device)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        while time.perf_counter() - t0 < busy_seconds:
+            for _ in range(8):
+                model.decode_step(toks)
+                busy_tokens += b
+            torch.cuda.synchronize()
+            if model.context_lens[0].item() >= model.max_seq - 8:
+                model.reset(b, context_len)
+        busy_elapsed = time.perf_counter() - t0
     result = CalibrationResult(
         model=cfg.name,
         device=torch.cuda.get_device_name(0),
@@ -196,6 +220,8 @@ def calibrate_service_profile(
         block_size=16,
         decode_tokens_per_s_peak=peak_tps,
         prefill_tokens_per_s=prefill_tps,
+        busy_marker_s=busy_elapsed,
+        busy_tokens=busy_tokens,
     )
     profile = ServiceProfile(
         alpha_ms=max(alpha, 0.1),
