@@ -1,0 +1,326 @@
+"""Inferno as a DEPLOYABLE analyzer:
+`analyzerName: inferno` must work from YAML (ConfigMaps) alone, with
+live reload, and the EKF tuner must refine α/β online from emulated
+drift.
+
+Reference parity anchors: the dormant service-class ConfigMap
+(charts/.../templates/manager/wva-configmap-service-class.yaml), the
+SystemData adapters (internal/utils/utils.go:125-196), and the tuner
+(internal/engines/analyzers/queueingmodel/tuner/tuner.go:29-143).
+"""
+import pytest
+
+from wva_amd.config.config import Config
+from wva_amd.constants import (
+    ACCELERATOR_CONFIG_MAP_NAME,
+    MODEL_PERF_CONFIG_MAP_NAME,
+    SERVICE_CLASS_CONFIG_MAP_NAME,
+)
+from wva_amd.inferno.types import (
+    parse_accelerator_configmap,
+    parse_model_perf_configmap,
+    parse_service_class_configmap,
+)
+from wva_amd.kube.objects import ConfigMap
+from wva_amd.api.types import ObjectMeta
+
+from test_e2e_emulated import MODEL, NS, VARIANT, make_stack, run_sim
+
+CONTROLLER_NS = "wva-system"
+
+SERVICE_CLASS_YAML = """\
+name: Premium
+priority: 1
+data:
+  - model: {model}
+    slo-tpot: 80
+    slo-ttft: 2000
+  - model: default/default
+    slo-tpot: 24
+    slo-ttft: 500
+"""
+
+FREEMIUM_YAML = """\
+name: Freemium
+priority: 10
+data:
+  - model: ibm/granite-13b
+    slo-tpot: 200
+    slo-ttft: 2000
+"""
+
+ACCELERATOR_YAML = """\
+device: AMD-Instinct-MI355X-288GB
+cost: "50.0"
+multiplicity: 1
+memSize: 288
+"""
+
+PERF_YAML = """\
+name: {model}
+acc: MI355X
+accCount: 1
+maxBatchSize: {maxb}
+atTokens: 50
+decodeParms: {{alpha: {alpha}, beta: {beta}}}
+"""
+
+
+def inferno_configmaps(model=MODEL, alpha=50.0, beta=2.0, maxb=8):
+    return [
+        ConfigMap(
+            metadata=ObjectMeta(
+                name=SERVICE_CLASS_CONFIG_MAP_NAME, namespace=CONTROLLER_NS
+            ),
+            data={
+                "premium.yaml": SERVICE_CLASS_YAML.format(model=model),
+                "freemium.yaml": FREEMIUM_YAML,
+            },
+        ),
+        ConfigMap(
+            metadata=ObjectMeta(
+                name=ACCELERATOR_CONFIG_MAP_NAME, namespace=CONTROLLER_NS
+            ),
+            data={"MI355X": ACCELERATOR_YAML},
+        ),
+        ConfigMap(
+            metadata=ObjectMeta(
+                name=MODEL_PERF_CONFIG_MAP_NAME, namespace=CONTROLLER_NS
+            ),
+            data={"llama-8b-tp1.yaml": PERF_YAML.format(
+                model=model, alpha=alpha, beta=beta, maxb=maxb
+            )},
+        ),
+    ]
+
+
+class TestParsers:
+    def test_service_class_reference_format(self):
+        out = parse_service_class_configmap({
+            "premium.yaml": SERVICE_CLASS_YAML.format(model="m/a"),
+            "freemium.yaml": FREEMIUM_YAML,
+            "junk": ": not yaml [",
+        })
+        names = {s.name for s in out}
+        assert names == {"Premium", "Freemium"}
+        prem = next(s for s in out if s.name == "Premium")
+        assert prem.priority == 1
+        assert prem.model_targets[0].slo_itl == 80  # slo-tpot key
+
+    def test_accelerator_reference_format(self):
+        out = parse_accelerator_configmap({
+            "MI355X": ACCELERATOR_YAML,
+            "BAD": "device: x\ncost: not-a-number",
+        })
+        # unparseable cost → accelerator skipped (utils.go:144-148)
+        assert len(out) == 1
+        acc = out[0]
+        assert acc.name == "MI355X"
+        assert acc.type == "AMD-Instinct-MI355X-288GB"
+        assert acc.cost == 50.0
+        assert acc.mem_size == 288
+
+    def test_perf_configmap(self):
+        out = parse_model_perf_configmap({
+            "one": PERF_YAML.format(model="m/a", alpha=5.0, beta=0.03, maxb=256),
+            "many": (
+                "- name: m/b\n  acc: MI300X\n  decodeParms: {alpha: 9, beta: 0.1}\n"
+                "- name: m/c\n  acc: MI355X\n  decodeParms: {alpha: 4, beta: 0.02}\n"
+            ),
+        })
+        assert {p.name for p in out} == {"m/a", "m/b", "m/c"}
+        pa = next(p for p in out if p.name == "m/a")
+        assert pa.service_parms.alpha == 5.0 and pa.acc == "MI355X"
+
+
+class TestConfigMapPlumbing:
+    def test_reconciler_routes_and_live_reloads(self):
+        cluster, sim, app = make_stack(replicas=1)
+        assert app.config.inferno_system_data() is None
+        for cm in inferno_configmaps():
+            cluster.create(cm)
+            app.configmap_reconciler.reconcile(CONTROLLER_NS, cm.metadata.name)
+        sd = app.config.inferno_system_data()
+        assert sd is not None
+        assert sd.accelerators[0].name == "MI355X"
+        assert {c.name for c in sd.service_classes} == {"Premium", "Freemium"}
+        v1 = app.config.inferno_config_version()
+
+        # live reload: change the perf alpha, version bumps, data updates
+        cm = cluster.get("ConfigMap", CONTROLLER_NS, MODEL_PERF_CONFIG_MAP_NAME)
+        cm.data = {"llama-8b-tp1.yaml": PERF_YAML.format(
+            model=MODEL, alpha=99.0, beta=3.0, maxb=8
+        )}
+        cluster.update(cm)
+        app.configmap_reconciler.reconcile(
+            CONTROLLER_NS, MODEL_PERF_CONFIG_MAP_NAME
+        )
+        assert app.config.inferno_config_version() > v1
+        assert app.config.inferno_system_data().models[0].service_parms.alpha == 99.0
+
+    def test_non_controller_namespace_ignored(self):
+        cluster, sim, app = make_stack(replicas=1)
+        cm = inferno_configmaps()[1]
+        cm.metadata.namespace = "default"
+        cluster.create(cm)
+        app.configmap_reconciler.reconcile("default", cm.metadata.name)
+        assert app.config.inferno_config_version() == 0
+
+    def test_deletion_degrades_to_fallback(self):
+        cluster, sim, app = make_stack(replicas=1)
+        for cm in inferno_configmaps():
+            cluster.create(cm)
+            app.configmap_reconciler.reconcile(CONTROLLER_NS, cm.metadata.name)
+        assert app.config.inferno_system_data() is not None
+        cluster.delete("ConfigMap", CONTROLLER_NS, ACCELERATOR_CONFIG_MAP_NAME)
+        app.configmap_reconciler.reconcile(
+            CONTROLLER_NS, ACCELERATOR_CONFIG_MAP_NAME
+        )
+        assert app.config.inferno_system_data() is None
+
+
+class TestInfernoFromYamlAlone:
+    def _stack_with_yaml(self, alpha=50.0, beta=2.0, maxb=8, profile=None):
+        from wva_amd.emulator.vllm_sim import ServiceProfile
+
+        prof = profile or ServiceProfile(
+            alpha_ms=alpha, beta_ms=beta, max_num_seqs=maxb,
+            num_gpu_blocks=500,
+        )
+        cluster, sim, app = make_stack(
+            replicas=1, profile=prof, analyzer="inferno"
+        )
+        for cm in inferno_configmaps(alpha=alpha, beta=beta, maxb=maxb):
+            cluster.create(cm)
+        # bootstrap AGAIN the way a real process start would see them
+        app.configmap_reconciler.bootstrap_initial_configmaps()
+        return cluster, sim, app
+
+    def test_scales_up_from_yaml_alone(self):
+        """No programmatic System injection anywhere: the engine builds
+        its Inferno analyzer from the three ConfigMaps."""
+        cluster, sim, app = self._stack_with_yaml()
+        assert app.saturation_engine.inferno_analyzer is None
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=20, seconds=30)
+        app.saturation_engine.optimize()
+        # analyzer was auto-built from YAML
+        assert app.saturation_engine.inferno_analyzer is not None
+        assert app.saturation_engine._inferno_built_version > 0
+        d = app.decision_cache.get(NS, VARIANT)
+        assert d is not None
+        assert d.target_replicas >= 2  # one tiny replica can't hold 20 qps
+
+    def test_live_reload_rebuilds_analyzer(self):
+        cluster, sim, app = self._stack_with_yaml()
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=5, seconds=10)
+        app.saturation_engine.optimize()
+        first = app.saturation_engine.inferno_analyzer
+        assert first is not None
+
+        cm = cluster.get("ConfigMap", CONTROLLER_NS, MODEL_PERF_CONFIG_MAP_NAME)
+        cm.data = {"llama-8b-tp1.yaml": PERF_YAML.format(
+            model=MODEL, alpha=60.0, beta=2.5, maxb=8
+        )}
+        cluster.update(cm)
+        app.configmap_reconciler.reconcile(
+            CONTROLLER_NS, MODEL_PERF_CONFIG_MAP_NAME
+        )
+        run_sim(sim, model, qps=5, seconds=10)
+        app.saturation_engine.optimize()
+        second = app.saturation_engine.inferno_analyzer
+        assert second is not None and second is not first
+        # rebuilt from the reloaded YAML seed (60.0) — the online tuner
+        # may already pull it toward the emulated truth (α=50) within
+        # the same tick, but it cannot still be the ORIGINAL seed (120
+        # was never in this ConfigMap generation)
+        alpha = second.system.perf[(MODEL, "MI355X")].service_parms.alpha
+        assert 40.0 <= alpha <= 62.0, alpha
+
+    def test_partial_yaml_falls_back_to_v2(self):
+        """Only two of three ConfigMaps: engine must not crash — it
+        falls back to the V2 token analyzer (loud log) and still
+        produces decisions."""
+        from wva_amd.emulator.vllm_sim import ServiceProfile
+
+        prof = ServiceProfile(
+            alpha_ms=50.0, beta_ms=2.0, max_num_seqs=8, num_gpu_blocks=500
+        )
+        cluster, sim, app = make_stack(
+            replicas=1, profile=prof, analyzer="inferno"
+        )
+        for cm in inferno_configmaps()[:2]:  # no perf data
+            cluster.create(cm)
+        app.configmap_reconciler.bootstrap_initial_configmaps()
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=20, seconds=10)
+        app.saturation_engine.optimize()
+        assert app.saturation_engine.inferno_analyzer is None
+        d = app.decision_cache.get(NS, VARIANT)
+        assert d is not None and d.target_replicas >= 2  # V2 fallback
+
+
+class TestOnlineTuner:
+    def test_tuner_corrects_drifted_alpha(self):
+        """Emulated drift: the ConfigMap seeds α=120 ms but the
+        emulated replicas actually serve at α=50 ms. After a few engine
+        ticks consuming observed (TTFT, ITL), the EKF pulls the System's
+        α toward the true value (VERDICT r01 #4 done criterion)."""
+        from wva_amd.emulator.vllm_sim import ServiceProfile
+
+        true_alpha, seeded_alpha = 50.0, 120.0
+        prof = ServiceProfile(
+            alpha_ms=true_alpha, beta_ms=2.0, max_num_seqs=8,
+            num_gpu_blocks=4000,
+        )
+        cluster, sim, app = TestInfernoFromYamlAlone()._stack_with_yaml(
+            alpha=seeded_alpha, beta=2.0, maxb=8, profile=prof
+        )
+        model = sim.model(MODEL, NS)
+        # moderate steady load: replicas see real traffic, latencies
+        # reflect the TRUE profile
+        for _ in range(6):
+            run_sim(sim, model, qps=3, seconds=20)
+            app.saturation_engine.optimize()
+        analyzer = app.saturation_engine.inferno_analyzer
+        assert analyzer is not None
+        perf = analyzer.system.perf[(MODEL, "MI355X")]
+        # α moved from the seeded 120 toward the true 50
+        assert perf.service_parms.alpha < seeded_alpha - 10, (
+            f"tuner did not move alpha: {perf.service_parms.alpha}"
+        )
+        assert len(analyzer._tuners) == 1
+
+    def test_tuner_rejects_outliers(self):
+        from wva_amd.analyzers.modelanalyzer import InfernoAnalyzer
+        from wva_amd.inferno.system import System
+        from wva_amd.inferno.types import (
+            AcceleratorSpec,
+            ModelAcceleratorPerfData,
+            ModelTarget,
+            ServiceClassSpec,
+            ServiceParmsSpec,
+            SystemData,
+        )
+
+        system = System(SystemData(
+            accelerators=[AcceleratorSpec(name="MI355X", cost=50)],
+            models=[ModelAcceleratorPerfData(
+                name="m", acc="MI355X", max_batch_size=8, at_tokens=50,
+                service_parms=ServiceParmsSpec(alpha=50.0, beta=2.0),
+            )],
+            service_classes=[ServiceClassSpec(
+                name="default", priority=1,
+                model_targets=[ModelTarget(model="m", slo_itl=80,
+                                           slo_ttft=2000)],
+            )],
+        ))
+        analyzer = InfernoAnalyzer(system)
+        # absurd observation (1000x latencies) → NIS gate rejects
+        accepted = analyzer.observe_latency(
+            "m", "MI355X", per_replica_rate=1.0, avg_in=100, avg_out=50,
+            ttft_ms=60_000.0, itl_ms=50_000.0,
+        )
+        assert accepted is False
+        assert system.perf[("m", "MI355X")].service_parms.alpha == 50.0

@@ -74,11 +74,17 @@ class InfernoAnalyzer:
     as the V2 analyzer so the CostAwareOptimizer consumes them unchanged.
     """
 
-    def __init__(self, system: System, service_class: str = "default"):
+    def __init__(self, system: System, service_class: str = "default",
+                 enable_tuner: bool = True):
         self.system = system
         self.service_class = service_class
         # model_id → (arrival_rate req/s, avg_in, avg_out) fed by the engine
         self._observed_load: Dict[str, tuple] = {}
+        # online EKF tuners per (model, accelerator) refining the
+        # ConfigMap-seeded α/β/γ from live (TTFT, ITL) observations
+        # (reference tuner.go:29-143 — dormant there, live here)
+        self.enable_tuner = enable_tuner
+        self._tuners: Dict[tuple, object] = {}
 
     def name(self) -> str:
         return "inferno-slo"
@@ -87,6 +93,83 @@ class InfernoAnalyzer:
         self, model_id: str, arrival_rate_per_s: float, avg_in: float, avg_out: float
     ) -> None:
         self._observed_load[model_id] = (arrival_rate_per_s, avg_in, avg_out)
+
+    def observe_latency(
+        self,
+        model_id: str,
+        accelerator: str,
+        per_replica_rate: float,
+        avg_in: float,
+        avg_out: float,
+        ttft_ms: float,
+        itl_ms: float,
+    ) -> bool:
+        """One EKF step on the (model, accelerator) service parameters
+        from a live (TTFT, ITL) observation at the per-REPLICA request
+        rate (the queueing model is single-server). Accepted updates are
+        written back into the System's perf record, so the very next
+        sizing pass uses the refined α/β/γ. Returns True if accepted
+        (NIS gate), False if rejected as an outlier."""
+        if not self.enable_tuner:
+            return False
+        perf = self.system.perf.get((model_id, accelerator))
+        if perf is None or per_replica_rate <= 0:
+            return False
+        from ..inferno.queue_analyzer import ServiceParms
+        from ..inferno.tuner import Observation, ServiceParmsTuner, TunerConfig
+
+        key = (model_id, accelerator)
+        tuner = self._tuners.get(key)
+        if tuner is None:
+            tuner = ServiceParmsTuner(
+                ServiceParms(
+                    alpha=perf.service_parms.alpha,
+                    beta=perf.service_parms.beta,
+                    gamma=perf.service_parms.gamma,
+                ),
+                TunerConfig(
+                    max_batch_size=max(perf.max_batch_size, 1),
+                    max_queue_size=max(perf.max_batch_size, 1) * 10,
+                ),
+            )
+            self._tuners[key] = tuner
+        # saturation gate: when the observed per-replica rate reaches
+        # the CURRENT model's maximum, the real queue is unbounded and
+        # the observed TTFT cannot be explained by the bounded-queue
+        # model — update from ITL alone (still identifies α, β at the
+        # running batch) instead of letting backlogged TTFT corrupt θ
+        itl_only = False
+        try:
+            from ..inferno.queue_analyzer import (
+                Configuration as _QCfg,
+                QueueAnalyzer as _QA,
+                RequestSize as _QReq,
+            )
+
+            probe = _QA(
+                _QCfg(
+                    max_batch_size=tuner.config.max_batch_size,
+                    max_queue_size=tuner.config.max_queue_size,
+                    service_parms=tuner.parms(),
+                ),
+                _QReq(avg_input_tokens=avg_in, avg_output_tokens=avg_out),
+            )
+            itl_only = per_replica_rate >= 0.9 * probe.rate_max
+        except (ValueError, ZeroDivisionError):
+            itl_only = True
+        accepted = tuner.update(Observation(
+            request_rate=per_replica_rate,
+            avg_input_tokens=avg_in,
+            avg_output_tokens=avg_out,
+            ttft_ms=ttft_ms,
+            itl_ms=itl_ms,
+        ), itl_only=itl_only)
+        if accepted:
+            p = tuner.parms()
+            perf.service_parms.alpha = p.alpha
+            perf.service_parms.beta = p.beta
+            perf.service_parms.gamma = p.gamma
+        return accepted
 
     def analyze(self, input: AnalyzerInput) -> AnalyzerResult:
         cfg = input.config
@@ -172,13 +255,29 @@ class InfernoAnalyzer:
         from ..inferno.types import MAX_QUEUE_TO_BATCH_RATIO
 
         perf = self.system.perf.get((model_id, accelerator))
-        sc = self.system.service_classes.get(self.service_class)
-        if perf is None or sc is None:
+        if perf is None:
             return 0.0
+        # SLO target resolution: the configured class first, then any
+        # other class carrying this model, highest priority (lowest
+        # number) first — so a multi-class ConfigMap (premium/freemium,
+        # reference chart wva-configmap-service-class.yaml) works
+        # without per-VA class labels
         target = None
-        for t in sc.model_targets:
-            if t.model == model_id:
-                target = t
+        classes = []
+        sc = self.system.service_classes.get(self.service_class)
+        if sc is not None:
+            classes.append(sc)
+        classes.extend(sorted(
+            (c for n, c in self.system.service_classes.items()
+             if sc is None or n != sc.name),
+            key=lambda c: c.priority,
+        ))
+        for c in classes:
+            for t in c.model_targets:
+                if t.model == model_id:
+                    target = t
+                    break
+            if target is not None:
                 break
         if target is None:
             return 0.0

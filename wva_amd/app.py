@@ -121,6 +121,7 @@ def build_app(
     reg.register_saturation_queries(source_registry)
     reg.register_scale_to_zero_queries(source_registry)
     reg.register_arrival_rate_query(source_registry)
+    reg.register_latency_queries(source_registry)
 
     datastore = Datastore(
         cluster,

@@ -34,6 +34,8 @@ QUERY_SCHEDULER_QUEUE_SIZE = "scheduler_queue_size"
 QUERY_SCHEDULER_QUEUE_BYTES = "scheduler_queue_bytes"
 QUERY_MODEL_REQUEST_COUNT = "model_request_count"
 QUERY_MODEL_ARRIVAL_RATE = "model_arrival_rate"
+QUERY_AVG_ITL = "avg_itl"
+QUERY_AVG_TTFT = "avg_ttft"
 
 
 def register_saturation_queries(source_registry: SourceRegistry) -> None:
@@ -118,6 +120,61 @@ def register_arrival_rate_query(source_registry: SourceRegistry) -> None:
         params=[PARAM_NAMESPACE, PARAM_MODEL_ID],
         description="Model-level request completion rate (req/s, 2m window)",
     ))
+
+
+def register_latency_queries(source_registry: SourceRegistry) -> None:
+    """Observed (TTFT, ITL) averages from the vLLM latency histograms
+    (constants/metrics.go:40-46 names) — the EKF tuner's measurement
+    vector (tuner.go observation = solved (TTFT, ITL); here the real
+    ones close the loop online)."""
+    src = source_registry.get(PROMETHEUS_SOURCE_NAME)
+    if src is None:
+        return
+    src.query_list().must_register(QueryTemplate(
+        name=QUERY_AVG_ITL,
+        type=QUERY_TYPE_PROMQL,
+        template=(
+            'sum(rate(vllm:time_per_output_token_seconds_sum'
+            '{namespace="{{.namespace}}",model_name="{{.modelID}}"}[2m]))'
+            ' / sum(rate(vllm:time_per_output_token_seconds_count'
+            '{namespace="{{.namespace}}",model_name="{{.modelID}}"}[2m]))'
+        ),
+        params=[PARAM_NAMESPACE, PARAM_MODEL_ID],
+        description="Mean inter-token latency (s, 2m window)",
+    ))
+    src.query_list().must_register(QueryTemplate(
+        name=QUERY_AVG_TTFT,
+        type=QUERY_TYPE_PROMQL,
+        template=(
+            'sum(rate(vllm:time_to_first_token_seconds_sum'
+            '{namespace="{{.namespace}}",model_name="{{.modelID}}"}[2m]))'
+            ' / sum(rate(vllm:time_to_first_token_seconds_count'
+            '{namespace="{{.namespace}}",model_name="{{.modelID}}"}[2m]))'
+        ),
+        params=[PARAM_NAMESPACE, PARAM_MODEL_ID],
+        description="Mean time to first token (s, 2m window)",
+    ))
+
+
+def collect_model_latency(
+    metrics_source: MetricsSource, model_id: str, namespace: str
+) -> Optional[tuple]:
+    """(ttft_ms, itl_ms) observed averages, or None when unavailable."""
+    results = metrics_source.refresh(RefreshSpec(
+        queries=[QUERY_AVG_TTFT, QUERY_AVG_ITL],
+        params={PARAM_MODEL_ID: model_id, PARAM_NAMESPACE: namespace},
+    ))
+    ttft = results.get(QUERY_AVG_TTFT)
+    itl = results.get(QUERY_AVG_ITL)
+    if (
+        ttft is None or ttft.has_error() or not ttft.values
+        or itl is None or itl.has_error() or not itl.values
+    ):
+        return None
+    return (
+        ttft.first_value().value * 1000.0,
+        itl.first_value().value * 1000.0,
+    )
 
 
 def collect_model_arrival_rate(

@@ -98,6 +98,14 @@ class Config:
         self._stz_global: ScaleToZeroConfigData = {}
         self._stz_by_ns: Dict[str, ScaleToZeroConfigData] = {}
         self._bootstrap_complete = False
+        # Inferno system config (service classes + accelerators + model
+        # perf data) from the wva-{service-class,accelerator,model-perf}
+        # ConfigMaps; version bumps on every update so the engine can
+        # rebuild its analyzer lazily on live reload
+        self._inferno_service_classes: list = []
+        self._inferno_accelerators: list = []
+        self._inferno_perf: list = []
+        self._inferno_version = 0
 
     # --- feature flags ---
 
@@ -208,6 +216,47 @@ class Config:
         return 1
 
     # --- bootstrap gating (readyz depends on this, cmd/main.go:486-498) ---
+
+    # --- Inferno system config (live-reloadable SLO analyzer inputs) ---
+
+    def update_inferno_service_classes(self, specs: list) -> None:
+        with self._lock:
+            self._inferno_service_classes = list(specs)
+            self._inferno_version += 1
+
+    def update_inferno_accelerators(self, specs: list) -> None:
+        with self._lock:
+            self._inferno_accelerators = list(specs)
+            self._inferno_version += 1
+
+    def update_inferno_perf(self, specs: list) -> None:
+        with self._lock:
+            self._inferno_perf = list(specs)
+            self._inferno_version += 1
+
+    def inferno_config_version(self) -> int:
+        with self._lock:
+            return self._inferno_version
+
+    def inferno_system_data(self):
+        """Build a SystemData from the ConfigMap-fed pieces; None until
+        accelerators, perf data AND service classes have all arrived
+        (the analyzer cannot size replicas without SLO targets and
+        measured service parms)."""
+        from ..inferno.types import SystemData
+
+        with self._lock:
+            if not (
+                self._inferno_accelerators
+                and self._inferno_perf
+                and self._inferno_service_classes
+            ):
+                return None
+            return SystemData(
+                accelerators=list(self._inferno_accelerators),
+                models=list(self._inferno_perf),
+                service_classes=list(self._inferno_service_classes),
+            )
 
     def mark_bootstrap_complete(self) -> None:
         with self._lock:

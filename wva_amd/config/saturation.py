@@ -13,6 +13,12 @@ from typing import Any, Dict, Optional
 from .. import constants as C
 
 ANALYZER_NAME_V2 = "saturation"
+ANALYZER_NAME_INFERNO = "inferno"
+# analyzers that run the V2 optimizer flow (token/SLO capacity +
+# CostAwareOptimizer) and therefore need the V2 thresholds defaulted —
+# the inferno path falls back to the V2 token analyzer when its system
+# config is incomplete, so it must carry valid thresholds too
+_V2_FLOW_ANALYZERS = (ANALYZER_NAME_V2, ANALYZER_NAME_INFERNO)
 
 
 class ConfigValidationError(ValueError):
@@ -52,7 +58,7 @@ class SaturationScalingConfig:
         return self.analyzer_name
 
     def apply_defaults(self) -> "SaturationScalingConfig":
-        if self.analyzer_name == ANALYZER_NAME_V2:
+        if self.analyzer_name in _V2_FLOW_ANALYZERS:
             if self.scale_up_threshold == 0:
                 self.scale_up_threshold = C.DEFAULT_SCALE_UP_THRESHOLD
             if self.scale_down_boundary == 0:
@@ -81,7 +87,7 @@ class SaturationScalingConfig:
                 f"kvCacheThreshold ({self.kv_cache_threshold:.2f}) should be >= "
                 f"kvSpareTrigger ({self.kv_spare_trigger:.2f})"
             )
-        if self.analyzer_name == ANALYZER_NAME_V2:
+        if self.analyzer_name in _V2_FLOW_ANALYZERS:
             if not 0 < self.scale_up_threshold <= 1:
                 raise ConfigValidationError(
                     f"scaleUpThreshold must be in (0, 1], got {self.scale_up_threshold:.2f}"

@@ -79,3 +79,10 @@ DEFAULT_SCALE_DOWN_BOUNDARY = 0.70
 WVA_CONFIG_MAP_NAME = "wva-variantautoscaling-config"
 SATURATION_CONFIG_MAP_NAME = "wva-saturation-scaling-config"
 SCALE_TO_ZERO_CONFIG_MAP_NAME = "wva-model-scale-to-zero-config"
+# Inferno SLO-analyzer system config (the reference ships the
+# service-class ConfigMap dormant in its chart,
+# charts/.../templates/manager/wva-configmap-service-class.yaml; here
+# all three feed the live `analyzerName: inferno` path)
+SERVICE_CLASS_CONFIG_MAP_NAME = "wva-service-class-config"
+ACCELERATOR_CONFIG_MAP_NAME = "wva-accelerator-config"
+MODEL_PERF_CONFIG_MAP_NAME = "wva-model-perf-config"

@@ -15,8 +15,11 @@ from typing import Callable
 
 from ..api.types import VariantAutoscaling
 from ..constants import (
+    ACCELERATOR_CONFIG_MAP_NAME,
+    MODEL_PERF_CONFIG_MAP_NAME,
     SATURATION_CONFIG_MAP_NAME,
     SCALE_TO_ZERO_CONFIG_MAP_NAME,
+    SERVICE_CLASS_CONFIG_MAP_NAME,
     WVA_CONFIG_MAP_NAME,
 )
 from ..kube.fake import ADDED, DELETED, MODIFIED, FakeCluster, WatchEvent
@@ -28,6 +31,9 @@ WELL_KNOWN_CONFIGMAPS = {
     WVA_CONFIG_MAP_NAME,
     SATURATION_CONFIG_MAP_NAME,
     SCALE_TO_ZERO_CONFIG_MAP_NAME,
+    SERVICE_CLASS_CONFIG_MAP_NAME,
+    ACCELERATOR_CONFIG_MAP_NAME,
+    MODEL_PERF_CONFIG_MAP_NAME,
 }
 
 

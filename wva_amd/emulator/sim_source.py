@@ -128,6 +128,20 @@ class SimMetricsSource:
             elapsed = min(window, self.sim.now) or 1.0
             return [MetricValue(value=count / elapsed, timestamp=ts)]
 
+        if query in (reg.QUERY_AVG_TTFT, reg.QUERY_AVG_ITL):
+            # mean over the last 2m of completed requests (the PromQL
+            # rate(sum)/rate(count) histogram-average analog); values in
+            # SECONDS like the vllm histograms
+            cutoff = self.sim.now - 120.0
+            recent = [c for c in model.completed if c.finish_time >= cutoff]
+            if not recent:
+                return []
+            if query == reg.QUERY_AVG_TTFT:
+                mean = sum(c.ttft for c in recent) / len(recent)
+            else:
+                mean = sum(c.itl for c in recent) / len(recent)
+            return [MetricValue(value=mean, timestamp=ts)]
+
         if query == reg.QUERY_MODEL_REQUEST_COUNT:
             retention = parse_go_duration(params.get("retentionPeriod", "10m"))
             cutoff = self.sim.now - retention

@@ -242,3 +242,72 @@ def parse_service_class_configmap(data: Dict[str, str]):
             continue
         out.append(ServiceClassSpec.from_dict(parsed))
     return out
+
+
+def parse_accelerator_configmap(data: Dict[str, str]) -> List[AcceleratorSpec]:
+    """Parse the accelerator ConfigMap (the reference's acceleratorCm in
+    utils.go:142-156): each data key is an accelerator NAME, each value a
+    YAML doc {device, cost, [multiplicity], [memSize], [memBW]}; cost may
+    be a quoted decimal string exactly as the reference parses it.
+
+      MI355X: |
+        device: AMD-Instinct-MI355X-288GB
+        cost: "50.0"
+        multiplicity: 1
+        memSize: 288
+    """
+    import yaml
+
+    out: List[AcceleratorSpec] = []
+    for key in sorted(data or {}):
+        try:
+            parsed = yaml.safe_load(data[key]) or {}
+        except Exception:  # noqa: BLE001
+            continue
+        if not isinstance(parsed, dict):
+            continue
+        try:
+            cost = float(parsed.get("cost", 0.0))
+        except (TypeError, ValueError):
+            # reference: unparseable cost → skip accelerator (utils.go:144-148)
+            continue
+        out.append(AcceleratorSpec(
+            name=key,
+            type=str(parsed.get("device", key)),
+            multiplicity=int(parsed.get("multiplicity", 1)),
+            mem_size=int(parsed.get("memSize", 0)),
+            mem_bw=int(parsed.get("memBW", 0)),
+            cost=cost,
+        ))
+    return out
+
+
+def parse_model_perf_configmap(data: Dict[str, str]) -> List[ModelAcceleratorPerfData]:
+    """Parse the model-perf ConfigMap: each data key holds one YAML doc
+    (or a list of docs) in the SystemData perf schema — the on-cluster
+    home for this repo's measured calibration records
+    (profiles/calibration_tp{N}.json → alpha/beta per (model, acc,
+    accCount)):
+
+      llama-8b-tp1.yaml: |
+        name: meta-llama/Llama-3.1-8B
+        acc: MI355X
+        accCount: 1
+        maxBatchSize: 256
+        atTokens: 50
+        decodeParms: {alpha: 4.9, beta: 0.025}
+    """
+    import yaml
+
+    out: List[ModelAcceleratorPerfData] = []
+    for key in sorted(data or {}):
+        try:
+            parsed = yaml.safe_load(data[key])
+        except Exception:  # noqa: BLE001
+            continue
+        docs = parsed if isinstance(parsed, list) else [parsed]
+        for doc in docs:
+            if not isinstance(doc, dict) or not doc.get("name"):
+                continue
+            out.append(ModelAcceleratorPerfData.from_dict(doc))
+    return out
