@@ -20,3 +20,6 @@ crd:               ## regenerate the CRD manifest
 
 clean:
 	rm -rf wva_amd/ops/build wva_amd/ops/csrc/*_hip.hip .pytest_cache
+
+chart-render: ## render the Helm chart (all features on) with the in-repo renderer
+	python scripts/render_chart.py --set hpa.enabled=true --set vllmService.enabled=true --set inferno.enabled=true
