@@ -123,9 +123,12 @@ def build_app(
     reg.register_arrival_rate_query(source_registry)
     reg.register_latency_queries(source_registry)
 
+    epp_secret_name, epp_secret_key = config.epp_metrics_reader_secret()
     datastore = Datastore(
         cluster,
         epp_bearer_token=config.epp_metric_reader_bearer_token(),
+        epp_metrics_reader_secret_name=epp_secret_name,
+        epp_metrics_reader_secret_key=epp_secret_key,
         scrape_fetch=scrape_fetch,
         source_registry=source_registry,
     )

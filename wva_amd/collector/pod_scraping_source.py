@@ -85,18 +85,24 @@ class PodScrapingSource:
         cluster: FakeCluster,
         pool: EndpointPool,
         bearer_token: str = "",
+        metrics_reader_secret_name: str = "",
+        metrics_reader_secret_key: str = "token",
         scrape_timeout_seconds: float = DEFAULT_SCRAPE_TIMEOUT_SECONDS,
         max_concurrent_scrapes: int = DEFAULT_MAX_CONCURRENT_SCRAPES,
         fetch: Optional[FetchFunc] = None,
         metrics_path: str = "/metrics",
+        metrics_scheme: str = "http",
     ):
         self.cluster = cluster
         self.pool = pool
         self.bearer_token = bearer_token
+        self.metrics_reader_secret_name = metrics_reader_secret_name
+        self.metrics_reader_secret_key = metrics_reader_secret_key
         self.scrape_timeout_seconds = scrape_timeout_seconds
         self.max_concurrent_scrapes = max(1, max_concurrent_scrapes)
         self.fetch = fetch or _default_fetch
         self.metrics_path = metrics_path
+        self.metrics_scheme = metrics_scheme
         self._query_list = QueryList()
         self._query_list.must_register(QueryTemplate(
             name=ALL_METRICS_QUERY,
@@ -131,13 +137,35 @@ class PodScrapingSource:
                 ready.append(pod)
         return ready
 
-    def _scrape_pod(self, pod: Pod) -> List[MetricValue]:
+    def _auth_token(self) -> str:
+        """Token resolution per reference getAuthToken
+        (pod_scraping_source.go:300-331): explicit token wins; else read
+        the metrics-reader Secret from the EPP service namespace (fresh
+        read each refresh — live rotation); a missing Secret or key
+        means auth is OPTIONAL, not an error."""
+        if self.bearer_token:
+            return self.bearer_token
+        if not self.metrics_reader_secret_name:
+            return ""
+        secret = self.cluster.try_get(
+            "Secret",
+            self.pool.endpoint_picker.namespace,
+            self.metrics_reader_secret_name,
+        )
+        if secret is None:
+            return ""
+        return secret.data.get(self.metrics_reader_secret_key, "")
+
+    def _scrape_pod(self, pod: Pod, token: str) -> List[MetricValue]:
         picker = self.pool.endpoint_picker
         host = pod.status.pod_ip or pod.name
-        url = f"http://{host}:{picker.metrics_port_number}{self.metrics_path}"
+        url = (
+            f"{self.metrics_scheme}://{host}:"
+            f"{picker.metrics_port_number}{self.metrics_path}"
+        )
         headers = {}
-        if self.bearer_token:
-            headers["Authorization"] = f"Bearer {self.bearer_token}"
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
         text = self.fetch(url, headers, self.scrape_timeout_seconds)
         values = parse_prometheus_text(text)
         for v in values:
@@ -148,13 +176,16 @@ class PodScrapingSource:
 
     def refresh(self, spec: RefreshSpec) -> Dict[str, MetricResult]:
         pods = self._discover_pods()
+        token = self._auth_token()
         all_values: List[MetricValue] = []
         errors: List[Exception] = []
         if pods:
             with concurrent.futures.ThreadPoolExecutor(
                 max_workers=min(self.max_concurrent_scrapes, len(pods))
             ) as pool:
-                futures = {pool.submit(self._scrape_pod, p): p for p in pods}
+                futures = {
+                    pool.submit(self._scrape_pod, p, token): p for p in pods
+                }
                 for fut in concurrent.futures.as_completed(futures):
                     try:
                         all_values.extend(fut.result())

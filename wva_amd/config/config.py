@@ -91,6 +91,8 @@ class Config:
         self._limited_mode_enabled = False
         self._scale_from_zero_max_concurrency = 10
         self._epp_metric_reader_bearer_token = ""
+        self._epp_metrics_reader_secret_name = ""
+        self._epp_metrics_reader_secret_key = "token"
         # global saturation config + per-namespace overrides
         self._saturation_global = SaturationScalingConfig()
         self._saturation_by_ns: Dict[str, SaturationScalingConfig] = {}
@@ -140,6 +142,20 @@ class Config:
     def set_epp_metric_reader_bearer_token(self, v: str) -> None:
         with self._lock:
             self._epp_metric_reader_bearer_token = v
+
+    def epp_metrics_reader_secret(self) -> tuple:
+        """(secret_name, key) for the Secret-sourced EPP bearer token
+        (reference pod_scraping_source.go:300-331)."""
+        with self._lock:
+            return (
+                self._epp_metrics_reader_secret_name,
+                self._epp_metrics_reader_secret_key,
+            )
+
+    def set_epp_metrics_reader_secret(self, name: str, key: str = "token") -> None:
+        with self._lock:
+            self._epp_metrics_reader_secret_name = name
+            self._epp_metrics_reader_secret_key = key or "token"
 
     # --- saturation config (global vs namespace-local) ---
 

@@ -36,6 +36,8 @@ _DEFAULTS: Dict[str, Any] = {
     "WVA_LIMITED_MODE": False,
     "SCALE_FROM_ZERO_ENGINE_MAX_CONCURRENCY": 10,
     "EPP_METRIC_READER_BEARER_TOKEN": "",
+    "EPP_METRICS_READER_SECRET_NAME": "",
+    "EPP_METRICS_READER_SECRET_KEY": "token",
     "GLOBAL_OPT_INTERVAL": "60s",
     "PROMETHEUS_BASE_URL": "",
     "PROMETHEUS_BEARER_TOKEN": "",
@@ -130,6 +132,10 @@ def load_config(
     )
     cfg.set_epp_metric_reader_bearer_token(
         str(merged["EPP_METRIC_READER_BEARER_TOKEN"])
+    )
+    cfg.set_epp_metrics_reader_secret(
+        str(merged["EPP_METRICS_READER_SECRET_NAME"]),
+        str(merged["EPP_METRICS_READER_SECRET_KEY"]),
     )
 
     prom = cfg.prometheus

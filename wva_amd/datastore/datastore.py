@@ -20,11 +20,15 @@ class Datastore:
         self,
         cluster: FakeCluster,
         epp_bearer_token: str = "",
+        epp_metrics_reader_secret_name: str = "",
+        epp_metrics_reader_secret_key: str = "token",
         scrape_fetch: Optional[FetchFunc] = None,
         source_registry=None,
     ):
         self.cluster = cluster
         self.epp_bearer_token = epp_bearer_token
+        self.epp_metrics_reader_secret_name = epp_metrics_reader_secret_name
+        self.epp_metrics_reader_secret_key = epp_metrics_reader_secret_key
         self.scrape_fetch = scrape_fetch
         # Optional collector SourceRegistry: pool sources register there
         # under their pool name (reference datastore.go registers one
@@ -50,6 +54,8 @@ class Datastore:
                 self.cluster,
                 pool,
                 bearer_token=self.epp_bearer_token,
+                metrics_reader_secret_name=self.epp_metrics_reader_secret_name,
+                metrics_reader_secret_key=self.epp_metrics_reader_secret_key,
                 fetch=self.scrape_fetch,
             )
             self._pool_sources[key] = source

@@ -142,6 +142,29 @@ class ConfigMap:
 
 
 @dataclass
+class Secret:
+    """core/v1 Secret subset — the EPP metrics-reader token source
+    (reference pod_scraping_source.go:300-331). `data` holds DECODED
+    string values; the serde layer base64-encodes on the wire like the
+    API server does."""
+
+    metadata: ObjectMeta = field(default_factory=ObjectMeta)
+    data: Dict[str, str] = field(default_factory=dict)
+    type: str = "Opaque"
+
+    kind: str = "Secret"
+    api_version: str = "v1"
+
+    @property
+    def name(self) -> str:
+        return self.metadata.name
+
+    @property
+    def namespace(self) -> str:
+        return self.metadata.namespace
+
+
+@dataclass
 class ServicePort:
     name: str = ""
     port: int = 0

@@ -263,6 +263,45 @@ def configmap_from_dict(d: Dict[str, Any]) -> ConfigMap:
     )
 
 
+# --- Secret ---
+
+def secret_to_dict(o) -> Dict[str, Any]:
+    import base64
+
+    return {
+        "apiVersion": o.api_version,
+        "kind": o.kind,
+        "metadata": _meta_to(o.metadata),
+        "type": o.type,
+        # API-server wire format: base64-encoded values under `data`
+        "data": {
+            k: base64.b64encode(str(v).encode()).decode()
+            for k, v in o.data.items()
+        },
+    }
+
+
+def secret_from_dict(d: Dict[str, Any]):
+    import base64
+
+    from .objects import Secret
+
+    data: Dict[str, str] = {}
+    for k, v in (d.get("data") or {}).items():
+        try:
+            data[k] = base64.b64decode(str(v)).decode()
+        except Exception:  # noqa: BLE001 — tolerate unencoded test data
+            data[k] = str(v)
+    # stringData convenience field (API server merges it into data)
+    for k, v in (d.get("stringData") or {}).items():
+        data[k] = str(v)
+    return Secret(
+        metadata=_meta_from(d.get("metadata") or {}),
+        data=data,
+        type=d.get("type", "Opaque"),
+    )
+
+
 # --- Service ---
 
 def service_to_dict(o: Service) -> Dict[str, Any]:
@@ -425,6 +464,9 @@ SERDE: Dict[str, Tuple[Callable, Callable, Tuple[str, str, bool]]] = {
     ),
     "ConfigMap": (
         configmap_to_dict, configmap_from_dict, ("api/v1", "configmaps", True)
+    ),
+    "Secret": (
+        secret_to_dict, secret_from_dict, ("api/v1", "secrets", True)
     ),
     "Service": (
         service_to_dict, service_from_dict, ("api/v1", "services", True)
