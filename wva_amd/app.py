@@ -103,6 +103,7 @@ def build_app(
     scrape_fetch=None,
     start_engines: bool = True,
     serve_http: bool = False,
+    measured_profiles_dir: Optional[str] = None,
 ) -> App:
     # Metrics source: explicit (sim) or Prometheus from config
     source_registry = SourceRegistry()
@@ -149,6 +150,15 @@ def build_app(
     inventory = TypeInventory("gpu", discovery)
     limiter = DefaultLimiter("gpu-limiter", inventory, GreedyBySaturation())
     capacity_store = CapacityKnowledgeStore()
+    if measured_profiles_dir:
+        # seed the MI355X calibration registry (profiles/*.json): new
+        # variants get hardware-measured k1 (incl. fp8 2x) before any
+        # live cache_config_info arrives
+        from .analyzers.capacity_store import load_calibration_dir
+
+        n = load_calibration_dir(capacity_store, measured_profiles_dir)
+        log.info("loaded %d measured capacity profiles from %s",
+                 n, measured_profiles_dir)
 
     saturation_engine = SaturationEngine(
         cluster=cluster,
