@@ -338,6 +338,19 @@ class SaturationEngine:
                     d.reason = (
                         f"V2 {d.action} (optimizer: {self.optimizer.name()}, enforced)"
                     )
+
+        # GPU limiter on the V2 path as well (improvement over the
+        # reference, whose V2 flow is unlimited-only — its
+        # CostAwareOptimizer ignores ResourceConstraints,
+        # cost_aware_optimizer.go:23,38 — while V1 gets GPULimiter at
+        # engine.go:379): without this, a V2 scale-up on an exhausted
+        # pool requests replicas that can never schedule
+        sat_cfg = self.config.saturation_config()
+        if sat_cfg.enable_limiter and self.limiter is not None and all_decisions:
+            try:
+                self.limiter.limit(all_decisions)
+            except Exception as e:  # noqa: BLE001
+                log.error("GPU limiter failed (V2), proceeding unlimited: %s", e)
         return all_decisions
 
     def _run_v2_analysis(self, model_id, namespace, data: _ModelData, config):

@@ -77,6 +77,12 @@ class DefaultLimiter:
         """V1 path: modify decisions in place based on available GPUs."""
         if not decisions:
             return
+        for d in decisions:
+            # preserve the pre-limit desire for observability (the V1
+            # target builder sets this; the V2 optimizer's decisions
+            # arrive with the field unset)
+            if d.original_target_replicas <= 0:
+                d.original_target_replicas = d.target_replicas
         self.inventory.refresh()
         self.inventory.set_used(self._calculate_used_gpus(decisions))
         allocator = self.inventory.create_allocator()
