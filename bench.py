@@ -131,6 +131,9 @@ def build_scenario(profile, seed: int):
     if os.environ.get("WVA_BENCH_LEAD"):
         # experiment knob: predictive scale-up lead (seconds)
         sat["scaleUpLeadSeconds"] = float(os.environ["WVA_BENCH_LEAD"])
+    if os.environ.get("WVA_BENCH_UP_THRESHOLD"):
+        # experiment knob: V2 scaleUpThreshold (default 0.85)
+        sat["scaleUpThreshold"] = float(os.environ["WVA_BENCH_UP_THRESHOLD"])
     config.update_saturation_config(SaturationScalingConfig.from_dict(sat))
     config.mark_bootstrap_complete()
     app = build_app(
