@@ -27,7 +27,11 @@ def main():
 
     import torch
     from wva_amd.calibration.model import LLAMA_3_8B, TINY, LlamaDecodeModel
+    from wva_amd.ops import enable_tuned_gemms
 
+    # profile the SERVING configuration: committed TunableOp solutions
+    # (bench.py enables the same table before calibration)
+    enable_tuned_gemms()
     cfg = LLAMA_3_8B if args.model == "8b" else TINY
     model = LlamaDecodeModel(cfg, max_batch=args.batch, max_seq=args.context + args.iters + 8)
     model.reset(args.batch, args.context)

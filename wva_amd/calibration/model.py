@@ -243,6 +243,9 @@ class LlamaDecodeModel:
         B = token_ids.shape[0]
         positions = self.context_lens[:B].clone()
 
+        ctx = positions + 1  # includes the new token (hoisted: one
+        # launch per step, not per layer — profiling showed 32 redundant
+        # int-add launches/step, profiles/decode8b_r2_kernel_stats.txt)
         x = self.embed.index_select(0, token_ids)  # [B, H]
         residual: Optional[torch.Tensor] = None
 
@@ -260,7 +263,6 @@ class LlamaDecodeModel:
                 positions, cfg.num_q_heads, cfg.num_kv_heads, cfg.rope_theta,
             )
 
-            ctx = positions + 1  # includes the new token
             attn = ops.gqa_decode_attn(
                 q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
             )

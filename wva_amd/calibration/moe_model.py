@@ -266,6 +266,7 @@ class MixtralDecodeModel:
         cfg = self.cfg
         B = token_ids.shape[0]
         positions = self.context_lens[:B].clone()
+        ctx = positions + 1  # hoisted out of the layer loop
 
         x = self.embed.index_select(0, token_ids)
         residual: Optional[torch.Tensor] = None
@@ -282,7 +283,6 @@ class MixtralDecodeModel:
                 qkv, self.k_cache[li][:B], self.v_cache[li][:B],
                 positions, cfg.num_q_heads, cfg.num_kv_heads, cfg.rope_theta,
             )
-            ctx = positions + 1
             attn = ops.gqa_decode_attn(
                 q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
             )

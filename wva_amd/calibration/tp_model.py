@@ -156,6 +156,9 @@ class TPLlamaDecodeModel:
         B = token_ids.shape[0]
         positions = self.context_lens[:B].clone()
 
+        ctx = positions + 1  # includes the new token (hoisted: one
+        # launch per step, not per layer — profiling showed 32 redundant
+        # int-add launches/step, profiles/decode8b_r2_kernel_stats.txt)
         x = self.embed.index_select(0, token_ids)
         residual: Optional[torch.Tensor] = None
 
@@ -172,7 +175,6 @@ class TPLlamaDecodeModel:
                 positions, shard.num_q_heads, shard.num_kv_heads,
                 cfg.rope_theta,
             )
-            ctx = positions + 1
             attn = ops.gqa_decode_attn(
                 q, self.k_cache[li][:B], self.v_cache[li][:B], ctx, self.scale
             )
