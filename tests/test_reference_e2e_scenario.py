@@ -27,7 +27,7 @@ NS = "default"
 VARIANT = "vllm-sim"
 
 
-def build(analyzer_name: str):
+def build(analyzer_name: str, backend: str = "fake"):
     from prometheus_client import CollectorRegistry
     from wva_amd.app import build_app
 
@@ -79,20 +79,42 @@ def build(analyzer_name: str):
         SaturationScalingConfig.from_dict({"analyzerName": analyzer_name})
     )
     config.mark_bootstrap_complete()
+
+    app_cluster = cluster
+    teardown = []
+    if backend == "rest":
+        from wva_amd.kube.cache import CachedCluster
+        from wva_amd.kube.rest import RestCluster
+        from k8s_test_server import K8sTestServer
+
+        server = K8sTestServer(cluster).start()
+        cache = CachedCluster(RestCluster(server.url)).start()
+        assert cache.wait_for_sync(10)
+        teardown = [cache.stop, server.stop]
+        app_cluster = cache
+
     app = build_app(
-        cluster, config, source=SimMetricsSource(sim),
+        app_cluster, config, source=SimMetricsSource(sim),
         metrics_registry=CollectorRegistry(), start_engines=False,
     )
+    if backend == "rest":
+        cache = app_cluster
+        orig_opt = app.saturation_engine.optimize
+        app.saturation_engine.optimize = (
+            lambda: (cache.wait_caught_up(), orig_opt())[1]
+        )
+    app._teardown = teardown  # type: ignore[attr-defined]
     return cluster, sim, app
 
 
+@pytest.mark.parametrize("backend", ["fake", "rest"])
 @pytest.mark.parametrize("analyzer", ["saturation", "percentage"])
-def test_scale_up_within_600s(analyzer):
+def test_scale_up_within_600s(analyzer, backend):
     """8 req/s at 100/50 tokens vs max-num-seqs=5 replicas: one replica
     holds ~5 concurrent (≈ 5/(50·ITL(5)) ≈ 4.4 req/s) — saturation must
     drive scale-up well within the reference's 600 s ceiling, on BOTH
     analyzer paths (V2 token and V1 percentage)."""
-    cluster, sim, app = build(analyzer)
+    cluster, sim, app = build(analyzer, backend)
     model = sim.model(MODEL, NS)
 
     scaled_at = None
@@ -113,3 +135,5 @@ def test_scale_up_within_600s(analyzer):
     assert scaled_at <= 600
     va = cluster.get("VariantAutoscaling", NS, VARIANT)
     assert va.status.desired_optimized_alloc.num_replicas > 1
+    for fn in getattr(app, "_teardown", []):
+        fn()
