@@ -608,3 +608,48 @@ class TestSecretBackedEppToken:
         import base64
         wire = serde.encode(raw)["data"]["token"]
         assert base64.b64decode(wire).decode() == "p@ss/w0rd=="
+
+
+class TestMainEntryRestMode:
+    def test_main_connects_through_cache_and_serves(self, server):
+        """`python -m wva_amd --kube-api-url ...` boots the full stack
+        through RestCluster + CachedCluster against the API server
+        (production wiring, cmd/main.go analog), then shuts down
+        cleanly on SIGTERM."""
+        import os
+        import signal
+        import subprocess
+        import sys
+        import time as _time
+
+        env = dict(os.environ)
+        env["PROMETHEUS_BASE_URL"] = "http://127.0.0.1:1"  # never queried
+        env["PYTHONPATH"] = os.path.dirname(
+            os.path.dirname(os.path.abspath(__file__))
+        )
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "wva_amd",
+             "--kube-api-url", server.url,
+             "--health-probe-bind-address", "127.0.0.1:0",
+             "--metrics-bind-address", "0"],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+            text=True,
+        )
+        try:
+            # the cache's initial LISTs hit the server → request counts
+            deadline = _time.time() + 30
+            while _time.time() < deadline:
+                if server.request_counts.get("GET", 0) > 5:
+                    break
+                _time.sleep(0.1)
+            assert server.request_counts.get("GET", 0) > 5
+            assert proc.poll() is None  # still running (synced + serving)
+        finally:
+            proc.send_signal(signal.SIGTERM)
+            try:
+                out, _ = proc.communicate(timeout=15)
+            except subprocess.TimeoutExpired:
+                proc.kill()
+                out, _ = proc.communicate()
+        assert proc.returncode == 0, out[-2000:]
+        assert "REST mode against" in out

@@ -52,6 +52,12 @@ def parse_flags(argv=None):
 
 
 def main(argv=None) -> int:
+    # install signal handlers FIRST: a SIGTERM during startup (informer
+    # sync, ConfigMap bootstrap) must still shut down cleanly
+    stop = []
+    signal.signal(signal.SIGINT, lambda *_: stop.append(1))
+    signal.signal(signal.SIGTERM, lambda *_: stop.append(1))
+
     args = parse_flags(argv)
     flags = {
         "METRICS_BIND_ADDRESS": args.metrics_bind_address,
@@ -74,10 +80,11 @@ def main(argv=None) -> int:
         source = SimMetricsSource(sim)
         log.info("running in emulated mode (in-memory cluster)")
     elif args.kube_api_url or os.environ.get("KUBERNETES_SERVICE_HOST"):
+        from .kube.cache import CachedCluster
         from .kube.rest import RestCluster
 
         if args.kube_api_url:
-            cluster = RestCluster(
+            rest = RestCluster(
                 args.kube_api_url,
                 token=args.kube_token,
                 ca_cert_path=args.kube_ca_cert,
@@ -85,8 +92,15 @@ def main(argv=None) -> int:
             )
             log.info("REST mode against %s", args.kube_api_url)
         else:
-            cluster = RestCluster.in_cluster()
+            rest = RestCluster.in_cluster()
             log.info("in-cluster REST mode")
+        # informer-style read cache in front of the API server (the
+        # controller-runtime cache analog, cmd/main.go:289-297): engine
+        # ticks read locally; writes pass through
+        cluster = CachedCluster(rest).start()
+        if not cluster.wait_for_sync(60.0):
+            log.error("informer cache failed to sync within 60s")
+            return 1
     else:
         cluster = FakeCluster()
         log.warning(
@@ -99,14 +113,14 @@ def main(argv=None) -> int:
     app.start()
     log.info("manager started (leader_elect=%s)", config.infra.enable_leader_election)
 
-    stop = []
-    signal.signal(signal.SIGINT, lambda *_: stop.append(1))
-    signal.signal(signal.SIGTERM, lambda *_: stop.append(1))
     try:
         while not stop:
             time.sleep(0.5)
     finally:
         app.stop()
+        cache_stop = getattr(cluster, "stop", None)
+        if cache_stop is not None and hasattr(cluster, "wait_for_sync"):
+            cache_stop()  # CachedCluster: stop pump + close REST watches
     return 0
 
 
