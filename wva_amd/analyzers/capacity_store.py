@@ -267,3 +267,159 @@ def load_calibration_dir(
             )
         records.append(d)
     return store.load_measured_profiles(records, accelerator=accelerator)
+
+
+# --- ConfigMap persistence (improvement over the reference) -----------------
+#
+# The reference's CapacityKnowledgeStore is memory-only: on controller
+# restart every live-learned record is lost and zero-replica capacity
+# estimates degrade until re-learned (SURVEY §5 checkpoint/resume:
+# "In-memory-only state that is lost on restart ... optionally persist
+# capacity records in a ConfigMap as an improvement"). These helpers
+# serialize the store to the `wva-capacity-store` ConfigMap so a
+# restarted controller resumes with its learned capacities, ages intact
+# (eviction and staleness keep working across restarts).
+
+CAPACITY_STORE_CONFIG_MAP_NAME = "wva-capacity-store"
+
+
+def _params_to_dict(p: Optional[VLLMEngineParams]) -> Optional[dict]:
+    if p is None:
+        return None
+    import dataclasses
+
+    return dataclasses.asdict(p)
+
+
+def _params_from_dict(d: Optional[dict]) -> Optional[VLLMEngineParams]:
+    if not d:
+        return None
+    import dataclasses
+
+    fields = {f.name for f in dataclasses.fields(VLLMEngineParams)}
+    return VLLMEngineParams(**{k: v for k, v in d.items() if k in fields})
+
+
+def snapshot_store(store: CapacityKnowledgeStore) -> dict:
+    """Serializable snapshot: records with ages (learned_at is a
+    monotonic clock, meaningless across processes)."""
+    now = time.monotonic()
+    with store._lock:
+        return {
+            "records": {
+                key: {
+                    "accelerator_name": r.accelerator_name,
+                    "gpu_count": r.gpu_count,
+                    "num_gpu_blocks": r.num_gpu_blocks,
+                    "block_size": r.block_size,
+                    "total_kv_capacity_tokens": r.total_kv_capacity_tokens,
+                    "effective_capacity": r.effective_capacity,
+                    "learned_from": r.learned_from,
+                    "age_seconds": max(now - r.learned_at, 0.0),
+                    "vllm_params": _params_to_dict(r.vllm_params),
+                }
+                for key, r in store._records.items()
+            },
+        }
+
+
+def restore_store(store: CapacityKnowledgeStore, snap: dict) -> int:
+    """Restore records from a snapshot; ages are preserved so staleness
+    and eviction behave as if the process had never restarted. Never
+    overwrites records learned in THIS process. Returns count restored."""
+    now = time.monotonic()
+    n = 0
+    records = (snap or {}).get("records") or {}
+    with store._lock:
+        for key, d in records.items():
+            if key in store._records:
+                continue
+            try:
+                rec = CapacityRecord(
+                    accelerator_name=str(d.get("accelerator_name", "")),
+                    gpu_count=int(d.get("gpu_count", 1)),
+                    num_gpu_blocks=int(d.get("num_gpu_blocks", 0)),
+                    block_size=int(d.get("block_size", 0)),
+                    total_kv_capacity_tokens=int(
+                        d.get("total_kv_capacity_tokens", 0)
+                    ),
+                    effective_capacity=int(d.get("effective_capacity", 0)),
+                    learned_from=str(d.get("learned_from", "")),
+                    vllm_params=_params_from_dict(d.get("vllm_params")),
+                )
+            except (TypeError, ValueError):
+                continue
+            rec.learned_at = now - float(d.get("age_seconds", 0.0))
+            store._records[key] = rec
+            n += 1
+    return n
+
+
+class CapacityStorePersistence:
+    """Periodic ConfigMap writer + bootstrap reader for the store."""
+
+    def __init__(self, cluster, store: CapacityKnowledgeStore,
+                 namespace: str, write_interval_seconds: float = 60.0):
+        self.cluster = cluster
+        self.store = store
+        self.namespace = namespace
+        self.write_interval_seconds = write_interval_seconds
+        self._last_write = 0.0
+        self._last_payload = ""
+
+    def restore(self) -> int:
+        """Read the ConfigMap (if present) into the store."""
+        import json
+
+        cm = self.cluster.try_get(
+            "ConfigMap", self.namespace, CAPACITY_STORE_CONFIG_MAP_NAME
+        )
+        if cm is None:
+            return 0
+        try:
+            snap = json.loads(cm.data.get("records", "{}"))
+        except ValueError:
+            return 0
+        return restore_store(self.store, {"records": snap})
+
+    def maybe_persist(self) -> bool:
+        """Write the snapshot if the interval elapsed and something
+        changed; best-effort (persistence failures never fail a tick)."""
+        import json
+
+        now = time.monotonic()
+        if now - self._last_write < self.write_interval_seconds:
+            return False
+        snap = snapshot_store(self.store)
+        payload = json.dumps(snap["records"], sort_keys=True)
+        # change detection must ignore ages (they advance every call)
+        canonical = json.dumps({
+            k: {f: v for f, v in rec.items() if f != "age_seconds"}
+            for k, rec in snap["records"].items()
+        }, sort_keys=True)
+        if canonical == self._last_payload:
+            self._last_write = now
+            return False
+        from ..api.types import ObjectMeta
+        from ..kube.objects import ConfigMap
+
+        cm = self.cluster.try_get(
+            "ConfigMap", self.namespace, CAPACITY_STORE_CONFIG_MAP_NAME
+        )
+        try:
+            if cm is None:
+                self.cluster.create(ConfigMap(
+                    metadata=ObjectMeta(
+                        name=CAPACITY_STORE_CONFIG_MAP_NAME,
+                        namespace=self.namespace,
+                    ),
+                    data={"records": payload},
+                ))
+            else:
+                cm.data = {"records": payload}
+                self.cluster.update(cm)
+        except Exception:  # noqa: BLE001 — best-effort (incl. 409s)
+            return False
+        self._last_write = now
+        self._last_payload = canonical
+        return True

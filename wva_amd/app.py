@@ -228,6 +228,21 @@ def build_app(
     # ConfigMap bootstrap before runnables (cmd/main.go:322-336)
     configmap_reconciler.bootstrap_initial_configmaps()
 
+    # capacity-store persistence (checkpoint/resume improvement): restore
+    # learned capacity records from the wva-capacity-store ConfigMap and
+    # write them back periodically from the engine tick
+    from .analyzers.capacity_store import CapacityStorePersistence
+    from .controllers.configmap import controller_namespace
+
+    persistence = CapacityStorePersistence(
+        cluster, capacity_store, controller_namespace()
+    )
+    restored = persistence.restore()
+    if restored:
+        log.info("restored %d capacity records from %s/%s",
+                 restored, controller_namespace(), "wva-capacity-store")
+    saturation_engine.capacity_persistence = persistence
+
     # HTTP surfaces (cmd/main.go:266-287,482-498): probe address serves
     # /healthz + /readyz (readyz gated on ConfigMap bootstrap); metrics
     # address serves /metrics from the emitter's registry.

@@ -128,6 +128,9 @@ class SaturationEngine:
         )
         # model_key -> (ts, total_demand) for the scale-up lead trend
         self._demand_history: Dict[str, tuple] = {}
+        # optional ConfigMap persistence of the capacity store
+        # (checkpoint/resume improvement — SURVEY §5); set by build_app
+        self.capacity_persistence = None
         self.v1_analyzer = SaturationAnalyzerV1()
         self.v2_analyzer = SaturationAnalyzerV2(self.capacity_store)
         # Optional Inferno SLO analyzer (analyzerName: "inferno"); the
@@ -185,6 +188,12 @@ class SaturationEngine:
             all_decisions = self._optimize_v1(model_groups)
 
         self.apply_saturation_decisions(all_decisions, va_map)
+
+        if self.capacity_persistence is not None:
+            try:
+                self.capacity_persistence.maybe_persist()
+            except Exception as e:  # noqa: BLE001 — never fail a tick
+                log.debug("capacity-store persistence failed: %s", e)
 
     # --- V1 path ---
 
