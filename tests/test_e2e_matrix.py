@@ -226,3 +226,40 @@ class TestKedaShapedMetricsQuery:
         # the companion gauges KEDA dashboards read
         names = {v.labels.get("__name__") for v in parse_prometheus_text(text)}
         assert {"wva_current_replicas", "wva_desired_ratio"} <= names
+
+
+class TestWatchNamespaceScoping:
+    def test_engine_ignores_other_namespaces(self):
+        """WATCH_NAMESPACE set: the engine only decides for VAs in that
+        namespace (cmd/main.go:289-297 single-namespace cache analog)."""
+        from wva_amd.kube.objects import Node
+
+        prof = ServiceProfile(**OVERLOAD_PROFILE)
+        cluster, sim, app = make_stack(replicas=1, profile=prof)
+        # a second VA in another namespace
+        other_ns = "other-team"
+        cluster.create(Deployment(
+            metadata=ObjectMeta(name="vllm-b", namespace=other_ns),
+            replicas=1,
+            selector={"app": "vllm-b"},
+            template=PodTemplateSpec(
+                labels={"app": "vllm-b"},
+                containers=[Container(requests={"amd.com/gpu": "1"})],
+            ),
+        ))
+        cluster.create(VariantAutoscaling(
+            metadata=ObjectMeta(
+                name="vllm-b", namespace=other_ns,
+                labels={"inference.optimization/acceleratorName": "MI355X"},
+            ),
+            spec=VariantAutoscalingSpec(
+                scale_target_ref=CrossVersionObjectReference(name="vllm-b"),
+                model_id=MODEL,
+            ),
+        ))
+        app.config.infra.watch_namespace = NS
+        model = sim.model(MODEL, NS)
+        run_sim(sim, model, qps=20, seconds=10)
+        app.saturation_engine.optimize()
+        assert app.decision_cache.get(NS, VARIANT) is not None
+        assert app.decision_cache.get(other_ns, "vllm-b") is None

@@ -163,7 +163,10 @@ class SaturationEngine:
         self.capacity_store.evict_stale(CAPACITY_EVICTION_TIMEOUT_S)
         self.v2_analyzer.evict_stale_history(HISTORY_EVICTION_TIMEOUT_S)
 
-        active_vas = active_variant_autoscalings(self.cluster)
+        # single-namespace scoping (cmd/main.go:289-297: the reference
+        # restricts the manager cache to WATCH_NAMESPACE when set)
+        watch_ns = self.config.infra.watch_namespace or None
+        active_vas = active_variant_autoscalings(self.cluster, watch_ns)
         if not active_vas:
             log.debug("no active VariantAutoscalings found, skipping optimization")
             return

@@ -61,7 +61,8 @@ class ScaleFromZeroEngine:
         self.executor.stop()
 
     def optimize(self) -> None:
-        inactive = inactive_variant_autoscalings(self.cluster)
+        watch_ns = self.config.infra.watch_namespace or None
+        inactive = inactive_variant_autoscalings(self.cluster, watch_ns)
         if not inactive:
             return
         max_conc = self.config.scale_from_zero_max_concurrency()
