@@ -653,3 +653,69 @@ class TestMainEntryRestMode:
                 out, _ = proc.communicate()
         assert proc.returncode == 0, out[-2000:]
         assert "REST mode against" in out
+
+
+class TestTwoManagerFailover:
+    def test_leader_failover_moves_engines(self, server):
+        """Two full Manager instances against ONE API server: exactly
+        one holds the lease and runs its engines; when it stops
+        (release-on-cancel), the standby takes over within its retry
+        period — the reference's HA deployment shape
+        (cmd/main.go:266-287, LeaderElectionReleaseOnCancel)."""
+        import time as _time
+
+        from wva_amd.config.config import Config
+        from wva_amd.runtime.manager import Manager
+
+        class CountingRunnable:
+            def __init__(self):
+                self.running = False
+                self.starts = 0
+
+            def start(self):
+                self.running = True
+                self.starts += 1
+
+            def stop(self):
+                self.running = False
+
+        managers, runnables, clients = [], [], []
+        for ident in ("pod-a", "pod-b"):
+            c = RestCluster(server.url)
+            clients.append(c)
+            cfg = Config()
+            cfg.infra.enable_leader_election = True
+            cfg.infra.leader_election_id = "wva-failover-test"
+            cfg.infra.lease_duration_seconds = 2.0
+            cfg.infra.renew_deadline_seconds = 1.5
+            cfg.infra.retry_period_seconds = 0.2
+            cfg.mark_bootstrap_complete()
+            m = Manager(c, cfg)
+            r = CountingRunnable()
+            m.add_runnable(r)
+            managers.append(m)
+            runnables.append(r)
+
+        managers[0].start()
+        deadline = _time.time() + 10
+        while _time.time() < deadline and not runnables[0].running:
+            _time.sleep(0.05)
+        assert runnables[0].running and managers[0].is_leader()
+
+        managers[1].start()
+        _time.sleep(1.0)  # several retry periods
+        # standby must NOT have started its engines
+        assert not runnables[1].running
+        assert not managers[1].is_leader()
+
+        # leader stops → release-on-cancel → standby takes over fast
+        managers[0].stop()
+        deadline = _time.time() + 10
+        while _time.time() < deadline and not runnables[1].running:
+            _time.sleep(0.05)
+        assert runnables[1].running and managers[1].is_leader()
+        assert not runnables[0].running
+
+        managers[1].stop()
+        for c in clients:
+            c.close()
