@@ -315,10 +315,14 @@ def main() -> None:
         # enough that driver-side SMI sampling records nonzero gpu_busy
         # (untimed — runs before the timed region)
         busy_s = float(os.environ.get("WVA_BENCH_GPU_BUSY_S", "8"))
+        # calibrate across the FULL serving batch range (replicas run up
+        # to max_num_seqs=256): the emulator then interpolates the
+        # measured ITL table instead of extrapolating a [1..64] fit
+        cal_batches = [1, 8, 32, 64, 128, 256]
         try:
             profile, calibration = calibrate_service_profile(
                 LLAMA_3_8B,
-                batch_sizes=[1, 8, 32, 64],
+                batch_sizes=cal_batches,
                 context_len=512,
                 max_seq=1024,
                 iters=5,
@@ -330,12 +334,22 @@ def main() -> None:
                   "re-measuring eager", file=sys.stderr)
             profile, calibration = calibrate_service_profile(
                 LLAMA_3_8B,
-                batch_sizes=[1, 8, 32, 64],
+                batch_sizes=cal_batches,
                 context_len=512,
                 max_seq=1024,
                 iters=5,
                 busy_seconds=busy_s,
             )
+        # measured-table profile: exact interpolation inside [1,256]
+        from wva_amd.emulator.vllm_sim import ServiceProfile as _SP
+
+        profile = _SP.from_itl_table(
+            calibration.batch_sizes, calibration.itl_ms,
+            max_num_seqs=profile.max_num_seqs,
+            num_gpu_blocks=profile.num_gpu_blocks,
+            block_size=profile.block_size,
+            prefill_tokens_per_s=profile.prefill_tokens_per_s,
+        )
     else:
         from wva_amd.emulator.vllm_sim import ServiceProfile
 
