@@ -165,8 +165,9 @@ def build_scenario(profile, seed: int):
                 name=MODEL_ID, acc="MI355X",
                 max_batch_size=profile.max_num_seqs,
                 at_tokens=OUTPUT_TOKENS,
-                service_parms=ServiceParmsSpec(
-                    alpha=profile.alpha_ms, beta=profile.beta_ms
+                service_parms=ServiceParmsSpec.from_itl_fit(
+                    profile.alpha_ms, profile.beta_ms,
+                    INPUT_TOKENS, OUTPUT_TOKENS,
                 ),
             )],
             service_classes=[ServiceClassSpec(
@@ -363,6 +364,19 @@ def main() -> None:
             profile.beta_ms = float(os.environ["WVA_BENCH_BETA"])
         if os.environ.get("WVA_BENCH_BLOCKS"):
             profile.num_gpu_blocks = int(os.environ["WVA_BENCH_BLOCKS"])
+        if os.environ.get("WVA_BENCH_ITL_TABLE"):
+            # replay a measured table: "1:4.92,8:4.96,...,256:12.25"
+            pairs = [
+                kv.split(":")
+                for kv in os.environ["WVA_BENCH_ITL_TABLE"].split(",")
+            ]
+            profile = ServiceProfile.from_itl_table(
+                [int(b) for b, _ in pairs], [float(t) for _, t in pairs],
+                max_num_seqs=profile.max_num_seqs,
+                num_gpu_blocks=profile.num_gpu_blocks,
+                block_size=profile.block_size,
+                prefill_tokens_per_s=profile.prefill_tokens_per_s,
+            )
 
     cal_wall_s = time.perf_counter() - cal_t0
 
@@ -373,20 +387,32 @@ def main() -> None:
     b_slo = int(0.9 * (SLO_ITL_MS - profile.alpha_ms) / max(profile.beta_ms, 1e-6))
     profile.max_num_seqs = max(1, min(profile.max_num_seqs, b_slo))
 
-    cluster, sim, app = build_scenario(profile, seed=rank)
-    model = sim.model(MODEL_ID, NS)
+    # Independent scenario shards per rank (different arrival-process
+    # seeds): one bench "step" advances EVERY shard by one autoscaler
+    # tick, and the rank's score is the mean over shards — same workload
+    # definition, ~1/sqrt(S) of the single-ramp sampling noise.
+    n_shards = max(1, int(os.environ.get("WVA_BENCH_SHARDS", "2")))
+    shards = []
+    for s_i in range(n_shards):
+        cluster, sim, app = build_scenario(profile, seed=rank * 97 + s_i)
+        shards.append({
+            "cluster": cluster, "sim": sim, "app": app,
+            "model": sim.model(MODEL_ID, NS), "hpa": HPAActuator(),
+            "desired": [], "oracle": [], "actuated": [],
+            "completed_before": 0, "t_start": 0.0,
+        })
 
     rate_per_replica = per_replica_req_rate(profile)
     peak_qps = args.peak_qps or 4.0 * rate_per_replica
-
-    hpa = HPAActuator()
     floor_qps = qps_ramp_frac(0.0, peak_qps)
 
     # --- warmup (untimed) ---
     for step in range(args.warmup):
-        run_step(sim, app, cluster, model, lambda t: floor_qps, hpa)
-
-    completed_before = len(model.completed)
+        for sh in shards:
+            run_step(sh["sim"], sh["app"], sh["cluster"], sh["model"],
+                     lambda t: floor_qps, sh["hpa"])
+    for sh in shards:
+        sh["completed_before"] = len(sh["model"].completed)
 
     def barrier_sync():
         if dist is not None:
@@ -394,72 +420,85 @@ def main() -> None:
         if has_gpu:
             torch.cuda.synchronize()
 
-    # --- timed region: exactly K steps ---
+    # --- timed region: exactly K steps (each advancing every shard) ---
     barrier_sync()
     t0 = time.perf_counter()
-    desired_series = []
-    oracle_series = []
-    t_start = sim.now
     total_sim = args.steps * ENGINE_INTERVAL_S
+    for sh in shards:
+        sh["t_start"] = sh["sim"].now
 
-    def qps_of_time(t):
-        return qps_ramp_frac((t - t_start) / total_sim, peak_qps)
+    def make_qps_of_time(t_start):
+        return lambda t: qps_ramp_frac((t - t_start) / total_sim, peak_qps)
 
-    raw_series = []
+    for sh in shards:
+        sh["qps_of_time"] = make_qps_of_time(sh["t_start"])
+
     for step in range(args.steps):
         qps = qps_ramp(step, args.steps, peak_qps)
-        # backlog visible at step start: requests an ideal controller must
-        # ALSO drain this interval (they accumulated while pods were coming
-        # ready — an environment property, not a controller error)
-        backlog = len(model.scheduler_queue) + sum(
-            len(rep.waiting)
-            for rep, _ready, dep, ns in sim.replicas.values()
-            if dep == VARIANT and ns == NS
-        )
-        actuated = run_step(sim, app, cluster, model, qps_of_time, hpa)
-        # Score the RAW WVA signal (wva_desired_replicas — the product
-        # this controller emits) against the raw oracle. HPA stabilization
-        # stays in the ACTUATION path (the sim scales through it, so SLO
-        # reflects stabilized provisioning), but is not scored: its 120 s
-        # MAX-window stretches any one-tick sizing transient into an
-        # 8-tick plateau and would score cluster policy, not the
-        # controller.
-        d_raw = app.decision_cache.get(NS, VARIANT)
-        raw_series.append(actuated)
-        desired_series.append(
-            d_raw.target_replicas if d_raw is not None else actuated
-        )
-        oracle_raw = max(
-            1,
-            math.ceil(
-                (qps + backlog / ENGINE_INTERVAL_S)
-                / (rate_per_replica * UTILIZATION_SETPOINT)
-            ),
-        )
-        oracle_series.append(oracle_raw)
+        for sh in shards:
+            sim, app, cluster, model = (
+                sh["sim"], sh["app"], sh["cluster"], sh["model"]
+            )
+            # backlog visible at step start: requests an ideal controller
+            # must ALSO drain this interval (they accumulated while pods
+            # were coming ready — an environment property, not a
+            # controller error)
+            backlog = len(model.scheduler_queue) + sum(
+                len(rep.waiting)
+                for rep, _ready, dep, ns in sim.replicas.values()
+                if dep == VARIANT and ns == NS
+            )
+            actuated = run_step(
+                sim, app, cluster, model, sh["qps_of_time"], sh["hpa"]
+            )
+            # Score the RAW WVA signal (wva_desired_replicas — the
+            # product this controller emits) against the raw oracle. HPA
+            # stabilization stays in the ACTUATION path (the sim scales
+            # through it, so SLO reflects stabilized provisioning), but
+            # is not scored: its 120 s MAX-window would score cluster
+            # policy, not the controller.
+            d_raw = app.decision_cache.get(NS, VARIANT)
+            sh["actuated"].append(actuated)
+            sh["desired"].append(
+                d_raw.target_replicas if d_raw is not None else actuated
+            )
+            sh["oracle"].append(max(
+                1,
+                math.ceil(
+                    (qps + backlog / ENGINE_INTERVAL_S)
+                    / (rate_per_replica * UTILIZATION_SETPOINT)
+                ),
+            ))
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
-    # --- score ---
+    # --- score (per shard, then mean over shards) ---
     # ±1 replica tolerance: integer replica counts and the oracle's own
     # boundary ambiguity make off-by-one a perfect score
-    errs = [
-        min(1.0, max(0, abs(d - o) - 1) / max(o, 1))
-        for d, o in zip(desired_series, oracle_series)
-    ]
-    # strict accuracy (no tolerance) reported alongside so the headline
-    # ±1 number can't be mistaken for it
-    errs_strict = [
-        min(1.0, abs(d - o) / max(o, 1))
-        for d, o in zip(desired_series, oracle_series)
-    ]
-    accuracy_strict = 100.0 * max(0.0, 1.0 - sum(errs_strict) / len(errs_strict))
-    if os.environ.get("WVA_BENCH_DEBUG"):
-        print(f"[bench-debug] desired(raw)={desired_series}\n"
-              f"[bench-debug] actuated    ={raw_series}\n"
-              f"[bench-debug] oracle(raw) ={oracle_series}", file=sys.stderr)
-    accuracy = 100.0 * max(0.0, 1.0 - sum(errs) / len(errs))
-    slo = compute_slo_attainment(model.completed[completed_before:])
+    accs, accs_strict, slos = [], [], []
+    for sh in shards:
+        pairs = list(zip(sh["desired"], sh["oracle"]))
+        errs = [
+            min(1.0, max(0, abs(d - o) - 1) / max(o, 1)) for d, o in pairs
+        ]
+        # strict accuracy (no tolerance) reported alongside so the
+        # headline ±1 number can't be mistaken for it
+        errs_strict = [min(1.0, abs(d - o) / max(o, 1)) for d, o in pairs]
+        accs.append(100.0 * max(0.0, 1.0 - sum(errs) / len(errs)))
+        accs_strict.append(
+            100.0 * max(0.0, 1.0 - sum(errs_strict) / len(errs_strict))
+        )
+        slos.append(compute_slo_attainment(
+            sh["model"].completed[sh["completed_before"]:]
+        ))
+        if os.environ.get("WVA_BENCH_DEBUG"):
+            print(f"[bench-debug] desired(raw)={sh['desired']}\n"
+                  f"[bench-debug] actuated    ={sh['actuated']}\n"
+                  f"[bench-debug] oracle(raw) ={sh['oracle']}",
+                  file=sys.stderr)
+    accuracy = sum(accs) / len(accs)
+    accuracy_strict = sum(accs_strict) / len(accs_strict)
+    slo = sum(slos) / len(slos)
     score = 0.5 * accuracy + 0.5 * slo
 
     ms_per_step = elapsed * 1000.0 / args.steps

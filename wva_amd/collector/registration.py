@@ -116,9 +116,18 @@ def register_arrival_rate_query(source_registry: SourceRegistry) -> None:
     src.query_list().must_register(QueryTemplate(
         name=QUERY_MODEL_ARRIVAL_RATE,
         type=QUERY_TYPE_PROMQL,
-        template='sum(rate(vllm:request_success_total{namespace="{{.namespace}}",model_name="{{.modelID}}"}[2m]))',
+        template=(
+            'sum(rate(vllm:request_success_total'
+            '{namespace="{{.namespace}}",model_name="{{.modelID}}"}[2m]))'
+            ' + clamp_min(sum(deriv(vllm:num_requests_waiting'
+            '{namespace="{{.namespace}}",model_name="{{.modelID}}"}[2m])), 0)'
+        ),
         params=[PARAM_NAMESPACE, PARAM_MODEL_ID],
-        description="Model-level request completion rate (req/s, 2m window)",
+        description=(
+            "Model-level ARRIVAL rate estimate (req/s, 2m window): "
+            "completion rate plus queue-growth rate — completions alone "
+            "understate arrivals under backlog and overstate during drain"
+        ),
     ))
 
 

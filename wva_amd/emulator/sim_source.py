@@ -122,9 +122,30 @@ class SimMetricsSource:
             return []
 
         if query == reg.QUERY_MODEL_ARRIVAL_RATE:
+            # true arrivals over the window — what the registered PromQL
+            # (completion rate + queue-depth derivative) estimates: a
+            # completions-only rate goes stale under backlog (arrivals >
+            # completions while under-provisioned) and overshoots during
+            # drain, destabilizing rate-based sizing
             window = 120.0
             cutoff = self.sim.now - window
-            count = sum(1 for c in model.completed if c.finish_time >= cutoff)
+            count = sum(
+                1 for c in model.completed
+                if c.spec.arrival_time >= cutoff
+            )
+            count += sum(
+                1 for spec in model.scheduler_queue
+                if spec.arrival_time >= cutoff
+            )
+            for _pod, rep in self._pods_of_model(model):
+                count += sum(
+                    1 for r in rep.waiting
+                    if r.spec.arrival_time >= cutoff
+                )
+                count += sum(
+                    1 for r in rep.running
+                    if r.spec.arrival_time >= cutoff
+                )
             elapsed = min(window, self.sim.now) or 1.0
             return [MetricValue(value=count / elapsed, timestamp=ts)]
 

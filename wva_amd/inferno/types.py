@@ -59,6 +59,27 @@ class ServiceParmsSpec:
             gamma=float(d.get("gamma", 0.0)),
         )
 
+    @classmethod
+    def from_itl_fit(
+        cls, alpha_ms: float, beta_itl_ms: float,
+        avg_input_tokens: float, avg_output_tokens: float,
+    ) -> "ServiceParmsSpec":
+        """Convert a RAW ITL fit (ITL(n) = α + β_itl·n, the calibration
+        harness' model — profiles/calibration_*.json) into the Inferno
+        parameter convention, whose iteration time multiplies β by
+        tokensCompute = (in+out)/(out+1) (queueanalyzer.go:261-279).
+        Feeding β_itl directly overestimates service time ~3× at the
+        default 100/50 token mix and makes the SLO sizer over-provision
+        by the same factor."""
+        tokens_compute = (avg_input_tokens + avg_output_tokens) / (
+            avg_output_tokens + 1.0
+        )
+        return cls(
+            alpha=alpha_ms,
+            beta=beta_itl_ms / max(tokens_compute, 1e-9),
+            gamma=0.0,
+        )
+
 
 @dataclass
 class ModelAcceleratorPerfData:
@@ -71,13 +92,33 @@ class ModelAcceleratorPerfData:
 
     @classmethod
     def from_dict(cls, d: Dict[str, Any]) -> "ModelAcceleratorPerfData":
+        # two parameter conventions:
+        #  * decodeParms/serviceParms: the reference's SystemData fields
+        #    (β multiplied by tokensCompute inside the queueing model)
+        #  * itlFit: a RAW calibration fit (ITL(n) = α + β·n, the
+        #    profiles/calibration_*.json numbers) with the request token
+        #    mix it should be converted at — feeding a raw ITL β as
+        #    decodeParms overestimates service time ~3× (see
+        #    ServiceParmsSpec.from_itl_fit)
+        itl_fit = d.get("itlFit")
+        if itl_fit:
+            parms = ServiceParmsSpec.from_itl_fit(
+                float(itl_fit.get("alpha", 0.0)),
+                float(itl_fit.get("beta", 0.0)),
+                float(itl_fit.get("avgInputTokens", 100.0)),
+                float(itl_fit.get("avgOutputTokens", 50.0)),
+            )
+        else:
+            parms = ServiceParmsSpec.from_dict(
+                d.get("decodeParms") or d.get("serviceParms") or {}
+            )
         return cls(
             name=d.get("name", ""),
             acc=d.get("acc", ""),
             acc_count=int(d.get("accCount", 1)),
             max_batch_size=int(d.get("maxBatchSize", 0)),
             at_tokens=int(d.get("atTokens", 0)),
-            service_parms=ServiceParmsSpec.from_dict(d.get("decodeParms") or d.get("serviceParms") or {}),
+            service_parms=parms,
         )
 
 
