@@ -263,3 +263,133 @@ class TestWatchNamespaceScoping:
         app.saturation_engine.optimize()
         assert app.decision_cache.get(NS, VARIANT) is not None
         assert app.decision_cache.get(other_ns, "vllm-b") is None
+
+
+class TestMultiNamespaceConfigIsolation:
+    def test_ns_local_thresholds_change_decisions(self):
+        """Two namespaces, same model + identical load: the namespace
+        with an ultra-sensitive ns-local saturation config scales up
+        while the global-config namespace doesn't — the e2e proof of
+        the global vs namespace-local override chain (config.go:360,
+        configmap_reconciler.go:154-168)."""
+        from wva_amd.kube.objects import ConfigMap
+
+        prof_kwargs = dict(
+            alpha_ms=20.0, beta_ms=1.0, max_num_seqs=64,
+            num_gpu_blocks=20_000,
+        )
+        prof = ServiceProfile(**prof_kwargs)
+        cluster, sim, app = make_stack(replicas=2, profile=prof)
+        other_ns = "team-b"
+        cluster.create(Deployment(
+            metadata=ObjectMeta(name="vllm-b", namespace=other_ns),
+            replicas=2,
+            selector={"app": "vllm-b"},
+            template=PodTemplateSpec(
+                labels={"app": "vllm-b"},
+                containers=[Container(requests={"amd.com/gpu": "1"})],
+            ),
+        ))
+        cluster.create(VariantAutoscaling(
+            metadata=ObjectMeta(
+                name="vllm-b", namespace=other_ns,
+                labels={"inference.optimization/acceleratorName": "MI355X"},
+            ),
+            spec=VariantAutoscalingSpec(
+                scale_target_ref=CrossVersionObjectReference(name="vllm-b"),
+                model_id=MODEL,
+            ),
+        ))
+        sim.register_variant(MODEL, other_ns, "vllm-b",
+                             ServiceProfile(**prof_kwargs))
+        sim.reconcile_deployments()
+
+        # ns-local override for team-b: ANY kv usage is "saturated"
+        # (kvCacheThreshold 0.01) → scale-up at the slightest load
+        app.datastore.namespace_track(other_ns)
+        cluster.create(ConfigMap(
+            metadata=ObjectMeta(name="wva-saturation-scaling-config",
+                                namespace=other_ns),
+            data={"default": (
+                "kvCacheThreshold: 0.01\n"
+                "queueLengthThreshold: 1\n"
+                "kvSpareTrigger: 0.01\n"
+                "queueSpareTrigger: 1\n"
+            )},
+        ))
+        app.configmap_reconciler.reconcile(
+            other_ns, "wva-saturation-scaling-config"
+        )
+
+        # identical light load on both namespaces
+        for ns_, variant in ((NS, VARIANT), (other_ns, "vllm-b")):
+            m = sim.model(MODEL, ns_)
+            run_sim(sim, m, qps=3, seconds=10)
+        app.saturation_engine.optimize()
+
+        d_global = app.decision_cache.get(NS, VARIANT)
+        d_local = app.decision_cache.get(other_ns, "vllm-b")
+        assert d_global is not None and d_local is not None
+        # global thresholds (0.80/5): light load → no scale-up
+        assert d_global.target_replicas <= 2
+        # ns-local hair-trigger config → scale-up
+        assert d_local.target_replicas > 2
+
+
+class TestFullZeroCycle:
+    def test_scale_to_zero_then_back_from_zero(self):
+        """The complete zero cycle in one scenario: idle model scales
+        to 0 (enforcer retention query), then queued EPP traffic brings
+        it back 0→1 (scale-from-zero engine) — the two reference
+        engines cooperating end to end."""
+        from wva_amd.config.scale_to_zero import ModelScaleToZeroConfig
+        from wva_amd.kube.objects import (
+            InferencePool, Pod, PodStatus, Service, ServicePort,
+        )
+
+        cluster, sim, app = make_stack(replicas=1)
+        model = sim.model(MODEL, NS)
+        app.config.update_scale_to_zero_config({
+            "default": ModelScaleToZeroConfig(
+                enable_scale_to_zero=True, retention_period="1m"
+            )
+        })
+        # EPP infra for the comeback path
+        cluster.create(Service(
+            metadata=ObjectMeta(name="pool-epp", namespace=NS),
+            selector={"app": "epp"},
+            ports=[ServicePort(name="metrics", port=9090)],
+        ))
+        cluster.create(Pod(
+            metadata=ObjectMeta(name="epp-0", namespace=NS,
+                                labels={"app": "epp"}),
+            status=PodStatus(phase="Running", ready=True,
+                             pod_ip="10.1.0.9"),
+        ))
+        app.datastore.scrape_fetch = (
+            lambda url, headers, timeout: sim.epp_metrics_text(NS)
+        )
+        cluster.create(InferencePool(
+            metadata=ObjectMeta(name="pool", namespace=NS),
+            selector={"app": VARIANT},
+            epp_service_name="pool-epp",
+        ))
+        app.inferencepool_reconciler.reconcile(NS, "pool")
+
+        # phase 1: idle past retention → scale to zero
+        run_sim(sim, model, qps=0, seconds=90)
+        app.saturation_engine.optimize()
+        d = app.decision_cache.get(NS, VARIANT)
+        assert d is not None and d.target_replicas == 0
+        cluster.scale("Deployment", NS, VARIANT, 0)  # HPA applies 0
+        sim.reconcile_deployments()
+
+        # phase 2: traffic arrives with zero replicas → EPP queue grows
+        run_sim(sim, model, qps=2, seconds=3)
+        assert len(model.scheduler_queue) > 0
+
+        app.scale_from_zero_engine.optimize()
+        deploy = cluster.get("Deployment", NS, VARIANT)
+        assert deploy.replicas == 1  # DirectActuator 0→1
+        va = cluster.get("VariantAutoscaling", NS, VARIANT)
+        assert cond.is_condition_true(va, "ScaleFromZeroMode")

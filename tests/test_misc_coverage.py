@@ -503,3 +503,27 @@ class TestBenchControlplaneScript:
         )
         assert out.returncode == 0, out.stderr[-1500:]
         assert "engine tick over" in out.stdout
+
+
+class TestDebugThreadsEndpoint:
+    def test_stack_dump_served(self):
+        """/debug/threads (the pprof analog): live stacks of
+        every thread over HTTP."""
+        import urllib.request
+
+        from wva_amd.runtime.http import ProbeServer
+
+        srv = ProbeServer(
+            "127.0.0.1:0", healthz=lambda: True, readyz=lambda: True,
+            serve_metrics=False,
+        )
+        srv.start()
+        try:
+            with urllib.request.urlopen(
+                f"http://127.0.0.1:{srv.port}/debug/threads", timeout=5
+            ) as resp:
+                text = resp.read().decode()
+        finally:
+            srv.stop()
+        assert "--- thread" in text
+        assert "MainThread" in text

@@ -61,6 +61,32 @@ class ProbeServer:
                     self.send_header("Content-Length", str(len(body)))
                     self.end_headers()
                     self.wfile.write(body)
+                elif self.path.startswith("/debug/threads"):
+                    # pprof-analog (SURVEY §5 "optionally add pprof"):
+                    # live stack dump of every thread — the first tool
+                    # for a wedged dispatcher/engine in production
+                    import sys as _sys
+                    import threading as _threading
+                    import traceback as _traceback
+
+                    lines = []
+                    names = {
+                        t.ident: t.name for t in _threading.enumerate()
+                    }
+                    for tid, frame in _sys._current_frames().items():
+                        lines.append(
+                            f"--- thread {names.get(tid, '?')} ({tid}) ---"
+                        )
+                        lines.extend(
+                            l.rstrip() for l in
+                            _traceback.format_stack(frame)
+                        )
+                    body = ("\n".join(lines) + "\n").encode()
+                    self.send_response(200)
+                    self.send_header("Content-Type", "text/plain")
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
                 else:
                     self.send_response(404)
                     self.end_headers()
