@@ -324,3 +324,35 @@ class TestOnlineTuner:
         )
         assert accepted is False
         assert system.perf[("m", "MI355X")].service_parms.alpha == 50.0
+
+
+class TestBetaConventionRegression:
+    def test_raw_itl_beta_would_underestimate_capacity(self):
+        """Regression guard for the β-convention bug: with realistic 8B
+        parameters (α≈4.7 ms, β·256 comparable to α), feeding the RAW
+        calibration ITL slope as the queueing-model β understates the
+        max service rate >2×; from_itl_fit restores agreement with the
+        directly computed saturated rate."""
+        from wva_amd.inferno.queue_analyzer import (
+            Configuration, QueueAnalyzer, RequestSize, ServiceParms,
+        )
+        from wva_amd.inferno.types import ServiceParmsSpec
+
+        ALPHA, BETA_ITL = 4.73, 0.0296  # profiles/calibration_8b.json scale
+        AVG_IN, AVG_OUT, B = 100.0, 50.0, 256
+
+        def rate_max(parms):
+            qa = QueueAnalyzer(
+                Configuration(max_batch_size=B, max_queue_size=B * 10,
+                              service_parms=parms),
+                RequestSize(avg_input_tokens=AVG_IN,
+                            avg_output_tokens=AVG_OUT),
+            )
+            return qa.rate_max
+
+        true_rate = B / ((ALPHA + BETA_ITL * B) / 1000.0) / AVG_OUT
+        conv = ServiceParmsSpec.from_itl_fit(ALPHA, BETA_ITL, AVG_IN, AVG_OUT)
+        r_conv = rate_max(ServiceParms(alpha=conv.alpha, beta=conv.beta))
+        r_raw = rate_max(ServiceParms(alpha=ALPHA, beta=BETA_ITL))
+        assert r_conv == pytest.approx(true_rate, rel=0.25)
+        assert r_raw < true_rate * 0.6  # the bug: >2x understated

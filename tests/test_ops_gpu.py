@@ -574,3 +574,49 @@ class TestFp8KvGpu:
             y = ext.skinny_linear_fp8(x, w8, scale.contiguous())
             ref = x.float() @ (w8.float() * scale[:, None]).t()
             torch.testing.assert_close(y.float(), ref, atol=0.5, rtol=2e-2)
+
+
+@pytest.mark.gpu
+class TestInfernoSizingOnMeasuredParms:
+    def test_sized_rate_tracks_measured_saturation(self):
+        """Bind the SLO sizer to hardware: calibrate the tiny engine on
+        THIS GPU, feed the RAW ITL fit through from_itl_fit (the β
+        convention converter), and check the queueing model's rate_max
+        against the directly measured saturated throughput — the two
+        must agree within modeling tolerance. Guards the ~3× β
+        convention bug class with a measurement, not a fixture."""
+        from wva_amd.calibration.itl_benchmark import calibrate_service_profile
+        from wva_amd.calibration.model import TINY
+        from wva_amd.inferno.queue_analyzer import (
+            Configuration, QueueAnalyzer, RequestSize, ServiceParms,
+        )
+        from wva_amd.inferno.types import ServiceParmsSpec
+
+        AVG_IN, AVG_OUT = 100.0, 50.0
+        profile, cal = calibrate_service_profile(
+            TINY, batch_sizes=[1, 8, 32, 64], context_len=256,
+            max_seq=512, iters=5,
+        )
+        parms = ServiceParmsSpec.from_itl_fit(
+            cal.alpha_ms, cal.beta_ms, AVG_IN, AVG_OUT
+        )
+        qa = QueueAnalyzer(
+            Configuration(
+                max_batch_size=64, max_queue_size=640,
+                service_parms=ServiceParms(
+                    alpha=parms.alpha, beta=parms.beta, gamma=parms.gamma
+                ),
+            ),
+            RequestSize(avg_input_tokens=AVG_IN, avg_output_tokens=AVG_OUT),
+        )
+        # directly measured saturated request rate at batch 64
+        measured_rate = 64.0 / (cal.itl_ms[-1] / 1000.0) / AVG_OUT
+        # the model's rate_max adds a prefill term the direct number
+        # ignores — agree within 35%
+        assert qa.rate_max == pytest.approx(measured_rate, rel=0.35), (
+            qa.rate_max, measured_rate
+        )
+        # (the raw-β inflation only bites when β·B is comparable to α —
+        # true for 8B at batch 256 but not for TINY at 64; the bug-class
+        # regression check with realistic parameters lives in
+        # tests/test_inferno_deployable.py on CPU)
