@@ -45,6 +45,10 @@ def main():
     p.add_argument("--replicas", type=int, default=10)
     p.add_argument("--ticks", type=int, default=10)
     p.add_argument("--analyzer", default="saturation")
+    p.add_argument("--backend", default="fake",
+                   choices=["fake", "rest", "cached"],
+                   help="cluster client the CONTROLLER uses: in-memory, "
+                        "uncached REST, or REST behind the informer cache")
     args = p.parse_args()
 
     cluster = FakeCluster()
@@ -90,7 +94,28 @@ def main():
         {"analyzerName": args.analyzer} if args.analyzer else {}
     ))
     config.mark_bootstrap_complete()
-    app = build_app(cluster, config, source=SimMetricsSource(sim),
+
+    app_cluster = cluster
+    server = cache = None
+    if args.backend in ("rest", "cached"):
+        import sys as _sys, os as _os
+        _sys.path.insert(0, _os.path.join(
+            _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))),
+            "tests"))
+        from k8s_test_server import K8sTestServer
+        from wva_amd.kube.rest import RestCluster
+
+        server = K8sTestServer(cluster).start()
+        app_cluster = RestCluster(server.url)
+        if args.backend == "cached":
+            from wva_amd.kube.cache import CachedCluster
+
+            cache = CachedCluster(app_cluster).start()
+            assert cache.wait_for_sync(30)
+            assert cache.wait_caught_up(30)
+            app_cluster = cache
+
+    app = build_app(app_cluster, config, source=SimMetricsSource(sim),
                     metrics_registry=CollectorRegistry(), start_engines=False)
 
     # brief traffic so metrics exist
@@ -109,11 +134,20 @@ def main():
     times.sort()
     pods = args.models * args.variants * args.replicas
     vas = args.models * args.variants
-    print(
+    line = (
         f"engine tick over {args.models} models / {vas} VAs / {pods} pods "
-        f"({args.analyzer or 'v1'}): median {times[len(times)//2]:.1f} ms, "
+        f"({args.analyzer or 'v1'}, backend={args.backend}): "
+        f"median {times[len(times)//2]:.1f} ms, "
         f"p90 {times[int(len(times)*0.9)]:.1f} ms"
     )
+    if server is not None:
+        gets = server.request_counts.get("GET", 0)
+        line += f", api-server GETs total {gets}"
+    print(line)
+    if cache is not None:
+        cache.stop()
+    if server is not None:
+        server.stop()
 
 
 if __name__ == "__main__":
