@@ -853,7 +853,19 @@ extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
   const int base = batch * num_kv_heads;
   if (base >= 1792) return 1;
   int splits = 1792 / base;
-  if (splits > 16) splits = 16;
+  // Split cap: 16 for moderate contexts (diminishing returns once each
+  // split's tile count is small), but LONG contexts lift it — at B=1
+  // ctx=128k the 16-split grid is 128 WGs on 256 CUs (7%% of resident
+  // capacity) and the KV sweep ran at 0.43 TB/s (profiles/
+  // longctx_itl.json, pre-fix). Allow as many splits as keep >= 8
+  // tiles of work per split; the merge kernel's cost is O(splits) per
+  // (b,h) row and stays negligible.
+  int cap = 16;
+  if (max_ctx_hint > 0) {
+    const int by_work = max_ctx_hint / (8 * TILE);
+    if (by_work > cap) cap = by_work;
+  }
+  if (splits > cap) splits = cap;
   const int max_useful = max_ctx_hint > 0 ? (max_ctx_hint + 2 * TILE - 1) / (2 * TILE) : splits;
   if (splits > max_useful && max_useful >= 1) splits = max_useful;
   return splits < 1 ? 1 : splits;
