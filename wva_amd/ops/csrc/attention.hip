@@ -860,9 +860,14 @@ extern "C" int gqa_decode_attn_num_splits(int batch, int num_kv_heads,
   // longctx_itl.json, pre-fix). Allow as many splits as keep >= 8
   // tiles of work per split; the merge kernel's cost is O(splits) per
   // (b,h) row and stays negligible.
+  static int min_tiles = [] {
+    const char* e = getenv("WVA_ATTN_MIN_TILES");  // tuning knob
+    int v = e ? atoi(e) : 8;
+    return v > 0 ? v : 8;
+  }();
   int cap = 16;
   if (max_ctx_hint > 0) {
-    const int by_work = max_ctx_hint / (8 * TILE);
+    const int by_work = max_ctx_hint / (min_tiles * TILE);
     if (by_work > cap) cap = by_work;
   }
   if (splits > cap) splits = cap;
