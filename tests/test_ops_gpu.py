@@ -275,6 +275,8 @@ class TestAttnV4:
     @pytest.mark.parametrize("batch,hq,hk,ctx", [
         (2, 8, 2, 33), (4, 32, 8, 512), (2, 64, 8, 300), (3, 4, 4, 100),
         (2, 8, 1, 2000),  # split path (small grid)
+        (1, 4, 1, 20000),  # long-context high-split regime (cap lifted
+                           # past 16: ~39 splits at min-tiles 8)
     ])
     def test_v4_matches_reference(self, dev, batch, hq, hk, ctx):
         from wva_amd.ops import _require_ext, gqa_decode_attn_ref
