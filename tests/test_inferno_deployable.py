@@ -356,3 +356,63 @@ class TestBetaConventionRegression:
         r_raw = rate_max(ServiceParms(alpha=ALPHA, beta=BETA_ITL))
         assert r_conv == pytest.approx(true_rate, rel=0.25)
         assert r_raw < true_rate * 0.6  # the bug: >2x understated
+
+
+class TestItlSurfaceGamma:
+    """The FULL 3-parameter reference model (α, β, γ) fitted from a
+    measured (batch, context) surface — γ is the per-context-token
+    memory term a fixed-context calibration silently absorbs into β."""
+
+    def test_surface_fit_and_conversion(self):
+        from wva_amd.calibration.itl_benchmark import fit_itl_surface
+        from wva_amd.inferno.types import ServiceParmsSpec
+
+        TRUE_A, TRUE_BEFF, TRUE_G = 4.7, 0.02, 3e-6
+        pts = [
+            (b, c, TRUE_A + b * (TRUE_BEFF + TRUE_G * c))
+            for b in (1, 8, 32, 64) for c in (512, 4096, 16384, 65536)
+        ]
+        a, beff, g, r2 = fit_itl_surface(pts)
+        assert r2 > 0.999999
+        parms = ServiceParmsSpec.from_itl_surface(a, beff, g, 100.0, 50.0)
+        tc = 150.0 / 51.0
+        assert parms.alpha == pytest.approx(TRUE_A)
+        assert parms.beta == pytest.approx(TRUE_BEFF / tc)
+        assert parms.gamma == pytest.approx(TRUE_G)
+
+    def test_gamma_raises_long_context_iter_time(self):
+        """With γ > 0 the queueing model's iteration time grows with the
+        request's context footprint — long-context SLO sizing stops
+        being optimistic."""
+        from wva_amd.inferno.queue_analyzer import (
+            Configuration, QueueAnalyzer, RequestSize, ServiceParms,
+        )
+
+        short = RequestSize(avg_input_tokens=100.0, avg_output_tokens=50.0)
+        long_ = RequestSize(avg_input_tokens=60_000.0,
+                            avg_output_tokens=500.0)
+        parms = ServiceParms(alpha=4.7, beta=0.0068, gamma=3e-6)
+        qa_s = QueueAnalyzer(Configuration(
+            max_batch_size=64, max_queue_size=640, service_parms=parms,
+        ), short)
+        qa_l = QueueAnalyzer(Configuration(
+            max_batch_size=64, max_queue_size=640, service_parms=parms,
+        ), long_)
+        # same parameters, longer contexts → strictly lower max rate
+        assert qa_l.rate_max < qa_s.rate_max * 0.5
+
+    def test_itl_surface_configmap_form(self):
+        """The model-perf ConfigMap accepts the measured-surface form
+        (profiles/calibration_8b_gamma.json numbers) and γ lands in the
+        system parms."""
+        from wva_amd.inferno.types import parse_model_perf_configmap
+
+        out = parse_model_perf_configmap({"surf": (
+            "name: m/s\nacc: MI355X\nmaxBatchSize: 256\natTokens: 50\n"
+            "itlSurface: {alpha: 5.346, betaEff: 0.00188, gamma: 2.68e-5,"
+            " avgInputTokens: 100, avgOutputTokens: 50}\n"
+        )})
+        p = out[0].service_parms
+        assert p.alpha == pytest.approx(5.346)
+        assert p.gamma == pytest.approx(2.68e-5)
+        assert p.beta == pytest.approx(0.00188 / (150 / 51), rel=1e-6)

@@ -235,3 +235,32 @@ def calibrate_service_profile(
     del model
     torch.cuda.empty_cache()
     return profile, result
+
+
+def fit_itl_surface(points) -> Tuple[float, float, float, float]:
+    """Fit the reference's FULL 3-parameter service model from a
+    measured (batch, context, itl_ms) surface.
+
+    The Inferno iteration-time model (queueanalyzer.go:261-279) is
+      iterTime = α + n·(β·tokensCompute + γ·tokensMemory)
+    with tokensMemory ≈ the per-request KV footprint (context tokens).
+    A fixed-context calibration can only produce α and a β that silently
+    absorbs γ·ctx at that one context; long-context serving then looks
+    as cheap as short. This fits the separable surface
+      ITL(n, ctx) = α + n·(β_eff + γ·ctx)
+    by least squares over the measured grid (β_eff = β·tokensCompute is
+    converted back by ServiceParmsSpec.from_itl_surface at the request
+    mix). Returns (alpha_ms, beta_eff_ms, gamma_ms_per_ctx_token, R²).
+    """
+    import numpy as np
+
+    pts = list(points)
+    A = np.array([[1.0, b, b * c] for b, c, _ in pts], dtype=np.float64)
+    y = np.array([t for _, _, t in pts], dtype=np.float64)
+    coef, *_ = np.linalg.lstsq(A, y, rcond=None)
+    alpha, beta_eff, gamma = (float(v) for v in coef)
+    pred = A @ coef
+    ss_res = float(((y - pred) ** 2).sum())
+    ss_tot = float(((y - y.mean()) ** 2).sum())
+    r2 = 1.0 - ss_res / ss_tot if ss_tot > 0 else 1.0
+    return alpha, beta_eff, gamma, r2

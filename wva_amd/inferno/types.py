@@ -80,6 +80,28 @@ class ServiceParmsSpec:
             gamma=0.0,
         )
 
+    @classmethod
+    def from_itl_surface(
+        cls, alpha_ms: float, beta_eff_ms: float,
+        gamma_ms_per_ctx_token: float,
+        avg_input_tokens: float, avg_output_tokens: float,
+    ) -> "ServiceParmsSpec":
+        """Convert a measured (batch, context) ITL SURFACE fit
+        (calibration.itl_benchmark.fit_itl_surface:
+        ITL = α + n·(β_eff + γ·ctx)) into the reference parameter
+        convention. γ maps directly: the model's tokensMemory term
+        (I + O/2) IS the average per-request context footprint, so the
+        per-context-token slope measured on hardware is the reference's
+        γ. β_eff divides out tokensCompute like from_itl_fit."""
+        tokens_compute = (avg_input_tokens + avg_output_tokens) / (
+            avg_output_tokens + 1.0
+        )
+        return cls(
+            alpha=alpha_ms,
+            beta=beta_eff_ms / max(tokens_compute, 1e-9),
+            gamma=gamma_ms_per_ctx_token,
+        )
+
 
 @dataclass
 class ModelAcceleratorPerfData:
@@ -101,7 +123,19 @@ class ModelAcceleratorPerfData:
         #    decodeParms overestimates service time ~3× (see
         #    ServiceParmsSpec.from_itl_fit)
         itl_fit = d.get("itlFit")
-        if itl_fit:
+        itl_surface = d.get("itlSurface")
+        if itl_surface:
+            # full 3-parameter measured surface (fit_itl_surface):
+            # includes the per-context-token γ the fixed-context fit
+            # absorbs into β — profiles/calibration_8b_gamma.json
+            parms = ServiceParmsSpec.from_itl_surface(
+                float(itl_surface.get("alpha", 0.0)),
+                float(itl_surface.get("betaEff", 0.0)),
+                float(itl_surface.get("gamma", 0.0)),
+                float(itl_surface.get("avgInputTokens", 100.0)),
+                float(itl_surface.get("avgOutputTokens", 50.0)),
+            )
+        elif itl_fit:
             parms = ServiceParmsSpec.from_itl_fit(
                 float(itl_fit.get("alpha", 0.0)),
                 float(itl_fit.get("beta", 0.0)),
