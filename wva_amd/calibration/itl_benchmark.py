@@ -259,8 +259,22 @@ def fit_itl_surface(points) -> Tuple[float, float, float, float]:
     y = np.array([t for _, _, t in pts], dtype=np.float64)
     coef, *_ = np.linalg.lstsq(A, y, rcond=None)
     alpha, beta_eff, gamma = (float(v) for v in coef)
+    if beta_eff < 0.0 or gamma < 0.0:
+        # physical constraint: service time is non-decreasing in batch
+        # and context (a weight-streaming-dominated model like 70B can
+        # fit a slightly negative β that would make the queueing model's
+        # service rates non-monotone) — refit with the offending term
+        # dropped
+        cols = [0] + ([1] if beta_eff >= 0.0 else []) \
+            + ([2] if gamma >= 0.0 else [])
+        coef_c, *_ = np.linalg.lstsq(A[:, cols], y, rcond=None)
+        full = [0.0, 0.0, 0.0]
+        for idx, c_ in zip(cols, coef_c):
+            full[idx] = float(c_)
+        alpha, beta_eff, gamma = full
+        coef = np.array(full)
     pred = A @ coef
     ss_res = float(((y - pred) ** 2).sum())
     ss_tot = float(((y - y.mean()) ** 2).sum())
     r2 = 1.0 - ss_res / ss_tot if ss_tot > 0 else 1.0
-    return alpha, beta_eff, gamma, r2
+    return alpha, max(beta_eff, 0.0), max(gamma, 0.0), r2
