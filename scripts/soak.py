@@ -51,6 +51,13 @@ def main():
     moe.reset(B, CTX)
     engines["mixtral8x7b"] = (moe, moe.decode_step)
 
+    # long-context engine: B=4 at 64k context (the lifted split-cap
+    # regime) — sustained deep-KV sweeps alongside the short-ctx engines
+    LCTX = 65536
+    lc = LlamaDecodeModel(LLAMA_3_8B, max_batch=4, max_seq=LCTX + 2048)
+    lc.reset(4, LCTX)
+    engines["llama8b_longctx64k"] = (lc, lc.decode_step)
+
     # CPU-side control plane on the emulated cluster, ticking between bursts
     from prometheus_client import CollectorRegistry
 
