@@ -35,28 +35,29 @@ def main():
     B = args.batch
     CTX, MAXSEQ = 512, 1024
 
+    # engine registry: name -> (model, step_fn, batch, reset_ctx, max_seq)
     engines = {}
     bf16 = LlamaDecodeModel(LLAMA_3_8B, max_batch=B, max_seq=MAXSEQ)
     bf16.reset(B, CTX)
     dec = GraphedDecoder(bf16, B, warmup_steps=2)
     dec.reset_to(B, CTX)
-    engines["llama8b_bf16_graph"] = (bf16, dec.decode_step)
+    engines["llama8b_bf16_graph"] = (bf16, dec.decode_step, B, CTX, MAXSEQ)
 
     fp8 = LlamaDecodeModel(LLAMA_3_8B, max_batch=B, max_seq=MAXSEQ,
                            kv_dtype="fp8")
     fp8.reset(B, CTX)
-    engines["llama8b_fp8kv"] = (fp8, fp8.decode_step)
+    engines["llama8b_fp8kv"] = (fp8, fp8.decode_step, B, CTX, MAXSEQ)
 
     moe = MixtralDecodeModel(MIXTRAL_8X7B, max_batch=B, max_seq=MAXSEQ)
     moe.reset(B, CTX)
-    engines["mixtral8x7b"] = (moe, moe.decode_step)
+    engines["mixtral8x7b"] = (moe, moe.decode_step, B, CTX, MAXSEQ)
 
     # long-context engine: B=4 at 64k context (the lifted split-cap
     # regime) — sustained deep-KV sweeps alongside the short-ctx engines
     LCTX = 65536
     lc = LlamaDecodeModel(LLAMA_3_8B, max_batch=4, max_seq=LCTX + 2048)
     lc.reset(4, LCTX)
-    engines["llama8b_longctx64k"] = (lc, lc.decode_step)
+    engines["llama8b_longctx64k"] = (lc, lc.decode_step, 4, LCTX, LCTX + 2048)
 
     # CPU-side control plane on the emulated cluster, ticking between bursts
     from prometheus_client import CollectorRegistry
@@ -77,8 +78,8 @@ def main():
     cp_model = sim.model(MODEL, NS)
 
     tokens = {
-        name: torch.randint(0, eng.cfg.vocab_size, (B,), device="cuda")
-        for name, (eng, _) in engines.items()
+        name: torch.randint(0, eng.cfg.vocab_size, (eb,), device="cuda")
+        for name, (eng, _, eb, _, _) in engines.items()
     }
     counts = {name: 0 for name in engines}
     ticks = 0
@@ -88,16 +89,16 @@ def main():
     last_report = t0
 
     while time.monotonic() < deadline:
-        for name, (eng, step) in engines.items():
+        for name, (eng, step, eb, ectx, emax) in engines.items():
             for _ in range(16):
                 logits = step(tokens[name])
-                counts[name] += B
+                counts[name] += eb
             torch.cuda.synchronize()
             assert torch.isfinite(logits.float()).all(), name
-            if eng.context_lens[0].item() >= MAXSEQ - 20:
-                eng.reset(B, CTX)
+            if eng.context_lens[0].item() >= emax - 20:
+                eng.reset(eb, ectx)
                 if name == "llama8b_bf16_graph":
-                    dec.reset_to(B, CTX)
+                    dec.reset_to(eb, ectx)
         # control-plane tick on the CPU between GPU bursts
         run_sim(sim, cp_model, qps=30, seconds=2)
         app.saturation_engine.optimize()
