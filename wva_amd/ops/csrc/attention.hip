@@ -798,7 +798,13 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
 }
 
 // Merge split-KV partials: one 64-thread wave per (b, q_head).
-__global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
+// 4 waves per (b, h) row: each wave reduces a strided subset of splits
+// into a partial (m, s, acc[128]), then wave 0 combines the 4 partials
+// through LDS. A single-wave merge looping over ~224 splits serially
+// was HALF the long-context step time (profiles/longctx_kernel_stats
+// .txt pre-fix: 101 µs vs the 115 µs attention sweep).
+#define MERGE_WAVES 4
+__global__ __launch_bounds__(64 * MERGE_WAVES) void gqa_decode_attn_merge_kernel(
     bf16* __restrict__ out,          // [B, Hq, 128]
     const float* __restrict__ workspace,  // [B, Hk, G, splits, 2+128]
     const int num_q_heads,
@@ -809,17 +815,23 @@ __global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
   const int G = num_q_heads / num_kv_heads;
   const int kvh = h / G;
   const int g = h % G;
-  const int lane = threadIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  __shared__ float sm_m[MERGE_WAVES];
+  __shared__ float sm_s[MERGE_WAVES];
+  __shared__ float sm_o[MERGE_WAVES][HEAD_DIM];
 
   const float* base = workspace +
       ((((long)b * num_kv_heads + kvh) * G + g) * (long)num_splits) *
           (2 + HEAD_DIM);
+  // wave-local pass over its strided split subset
   float M = -INFINITY;
-  for (int sp = 0; sp < num_splits; ++sp) {
+  for (int sp = wave; sp < num_splits; sp += MERGE_WAVES) {
     M = fmaxf(M, base[sp * (2 + HEAD_DIM)]);
   }
   float S = 0.0f, o0 = 0.0f, o1 = 0.0f;
-  for (int sp = 0; sp < num_splits; ++sp) {
+  for (int sp = wave; sp < num_splits; sp += MERGE_WAVES) {
     const float* wsp = base + sp * (2 + HEAD_DIM);
     const float mw = wsp[0];
     if (mw == -INFINITY) continue;
@@ -828,10 +840,31 @@ __global__ __launch_bounds__(64) void gqa_decode_attn_merge_kernel(
     o0 = fmaf(wsp[2 + 2 * lane], f, o0);
     o1 = fmaf(wsp[2 + 2 * lane + 1], f, o1);
   }
-  const float inv = S > 0.0f ? 1.0f / S : 0.0f;
+  if (lane == 0) {
+    sm_m[wave] = M;
+    sm_s[wave] = S;
+  }
+  sm_o[wave][2 * lane] = o0;
+  sm_o[wave][2 * lane + 1] = o1;
+  __syncthreads();
+
+  if (wave != 0) return;
+  // combine the wave partials (online-softmax merge across <=4 terms)
+  float Mg = -INFINITY;
+  for (int w = 0; w < MERGE_WAVES; ++w) Mg = fmaxf(Mg, sm_m[w]);
+  float Sg = 0.0f;
+  float g0 = 0.0f, g1 = 0.0f;
+  for (int w = 0; w < MERGE_WAVES; ++w) {
+    if (sm_m[w] == -INFINITY) continue;
+    const float f = __expf(sm_m[w] - Mg);
+    Sg += sm_s[w] * f;
+    g0 = fmaf(sm_o[w][2 * lane], f, g0);
+    g1 = fmaf(sm_o[w][2 * lane + 1], f, g1);
+  }
+  const float inv = Sg > 0.0f ? 1.0f / Sg : 0.0f;
   bf16x2* orow = reinterpret_cast<bf16x2*>(
       out + ((long)b * num_q_heads + h) * HEAD_DIM);
-  orow[lane] = bf16x2{f2bf(o0 * inv), f2bf(o1 * inv)};
+  orow[lane] = bf16x2{f2bf(g0 * inv), f2bf(g1 * inv)};
 }
 
 extern "C" void launch_gqa_decode_attn_v4_ex(
@@ -889,7 +922,7 @@ extern "C" void launch_gqa_decode_attn(
                      num_q_heads, num_kv_heads, max_seq, scale);
   if (num_splits > 1) {
     dim3 mgrid(batch, num_q_heads);
-    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64), 0,
+    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64 * MERGE_WAVES), 0,
                        stream, (bf16*)out, (const float*)workspace,
                        num_q_heads, num_kv_heads, num_splits);
   }
@@ -925,7 +958,7 @@ extern "C" void launch_gqa_decode_attn_v5_ex(
   }
   if (num_splits > 1) {
     dim3 mgrid(batch, num_q_heads);
-    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64), 0,
+    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64 * MERGE_WAVES), 0,
                        stream, (bf16*)out, (const float*)workspace,
                        num_q_heads, num_kv_heads, num_splits);
   }
@@ -961,7 +994,7 @@ extern "C" void launch_gqa_decode_attn_v4_ex(
   }
   if (num_splits > 1) {
     dim3 mgrid(batch, num_q_heads);
-    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64), 0,
+    hipLaunchKernelGGL(gqa_decode_attn_merge_kernel, mgrid, dim3(64 * MERGE_WAVES), 0,
                        stream, (bf16*)out, (const float*)workspace,
                        num_q_heads, num_kv_heads, num_splits);
   }
