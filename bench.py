@@ -5,11 +5,14 @@
 
 What runs:
   1. GPU calibration (per rank, on its own MI355X): real bf16 decode steps
-     of random-init Llama-3-1-8B built from this repo's HIP/CDNA4 kernels
-     (wva_amd.ops) + hipBLASLt GEMMs, sweeping batch sizes → measured
-     ITL(batch) = α+β·batch and 288 GB-derived KV capacity → the replica
-     ServiceProfile. No GPU ⇒ falls back to a documented synthetic profile
-     (data field still says synthetic either way — weights are random-init).
+     of random-init Llama-3.1-8B built from this repo's HIP/CDNA4 kernels
+     (wva_amd.ops) + hipBLASLt GEMMs across the FULL serving batch range
+     [1..256] → the emulator interpolates the measured ITL table exactly
+     (α/β also fitted for the linear-model consumers) + 288 GB-derived KV
+     capacity → the replica ServiceProfile; an 8 s GPU-busy decode marker
+     makes the phase driver-visible. No GPU ⇒ documented synthetic
+     fallback profile (data says synthetic either way — random-init
+     weights).
   2. Timed region: K autoscaler steps. One step = advance the emulated
      cluster (simulated vLLM replicas with the measured service profile)
      through ENGINE_INTERVAL seconds of the QPS ramp, run one saturation-
@@ -27,9 +30,12 @@ What runs:
      actuation (what the sim scales to) still passes through the HPA
      stabilization analog, which shapes SLO but is not itself scored.
 
-Multi-rank (torchrun, one rank per GPU): weak scaling — each rank runs an
-independent cluster shard end-to-end on its own GPU-calibrated profile;
-value is the mean score over ranks, ms_per_step the MAX over ranks.
+Each rank runs WVA_BENCH_SHARDS (default 2) independent ramp shards with
+distinct arrival seeds — one bench step advances every shard one
+autoscaler tick; the rank score is the shard mean (~1/sqrt(S) sampling
+noise). Multi-rank (torchrun, one rank per GPU): weak scaling — each
+rank additionally has its own GPU-calibrated profile; value is the mean
+score over ranks, ms_per_step the MAX over ranks.
 """
 from __future__ import annotations
 
