@@ -200,3 +200,26 @@ class TestKustomizeTree:
         )) as f:
             doc = yaml.safe_load(f)
         assert validate(schema, doc) == []
+
+
+class TestVaTemplate:
+    def test_sample_va_renders_and_validates(self):
+        from wva_amd.api.crd import variantautoscaling_crd
+        from wva_amd.kube.openapi import validate
+
+        docs = render_chart(CHART, {"va.enabled": True})
+        va = next(d for d in docs if d["kind"] == "VariantAutoscaling")
+        assert va["spec"]["modelID"] == "meta-llama/Llama-3.1-8B"
+        assert va["metadata"]["labels"][
+            "inference.optimization/acceleratorName"] == "MI355X"
+        schema = variantautoscaling_crd()["spec"]["versions"][0]["schema"][
+            "openAPIV3Schema"]
+        assert validate(schema, va) == []
+
+    def test_controller_instance_label_gated(self):
+        docs = render_chart(CHART, {
+            "va.enabled": True, "va.controllerInstance": "team-a",
+        })
+        va = next(d for d in docs if d["kind"] == "VariantAutoscaling")
+        assert va["metadata"]["labels"][
+            "wva.llmd.ai/controller-instance"] == "team-a"
