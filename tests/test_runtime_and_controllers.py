@@ -427,3 +427,93 @@ class TestConfigLoader:
                 },
                 require_prometheus=False,
             )
+
+
+class TestInferencePoolReconciler:
+    """Reference inferencepool_reconciler.go:41-118 + pool.go:34-148:
+    InferencePool → EndpointPool conversion, metrics-port-by-name, and
+    deletion-driven pool removal."""
+
+    def _stack(self):
+        from wva_amd.controllers.inferencepool import InferencePoolReconciler
+        from wva_amd.datastore.datastore import Datastore
+
+        cluster = FakeCluster()
+        ds = Datastore(cluster)
+        return cluster, ds, InferencePoolReconciler(cluster, ds)
+
+    def test_conversion_and_metrics_port_by_name(self):
+        from wva_amd.kube.objects import (
+            InferencePool, Service, ServicePort,
+        )
+
+        cluster, ds, rec = self._stack()
+        cluster.create(Service(
+            metadata=ObjectMeta(name="pool-epp", namespace="ns"),
+            selector={"app": "epp"},
+            ports=[
+                ServicePort(name="grpc", port=9002),
+                ServicePort(name="epp-metrics", port=9091),  # "metric" hit
+            ],
+        ))
+        cluster.create(InferencePool(
+            metadata=ObjectMeta(name="pool", namespace="ns"),
+            selector={"app": "vllm"},
+            epp_service_name="pool-epp",
+        ))
+        rec.reconcile("ns", "pool")
+        pool = ds.pool_get("ns", "pool")
+        assert pool is not None
+        assert pool.selector == {"app": "vllm"}
+        assert pool.endpoint_picker.service_name == "pool-epp"
+        # port named *metric* wins over other ports (pool.go:117)
+        assert pool.endpoint_picker.metrics_port_number == 9091
+
+    def test_missing_service_falls_back_to_9090(self):
+        from wva_amd.kube.objects import InferencePool
+
+        cluster, ds, rec = self._stack()
+        cluster.create(InferencePool(
+            metadata=ObjectMeta(name="pool", namespace="ns"),
+            selector={"app": "vllm"},
+            epp_service_name="absent-svc",
+        ))
+        rec.reconcile("ns", "pool")
+        assert ds.pool_get("ns", "pool").endpoint_picker \
+            .metrics_port_number == 9090
+
+    def test_default_epp_service_name_convention(self):
+        from wva_amd.kube.objects import InferencePool
+
+        cluster, ds, rec = self._stack()
+        cluster.create(InferencePool(
+            metadata=ObjectMeta(name="mypool", namespace="ns"),
+            selector={"a": "b"},
+        ))
+        rec.reconcile("ns", "mypool")
+        assert ds.pool_get("ns", "mypool").endpoint_picker \
+            .service_name == "mypool-epp"
+
+    def test_deletion_removes_pool_and_source(self):
+        from wva_amd.collector.registry import SourceRegistry
+        from wva_amd.controllers.inferencepool import InferencePoolReconciler
+        from wva_amd.datastore.datastore import Datastore
+        from wva_amd.kube.objects import InferencePool
+
+        cluster = FakeCluster()
+        registry = SourceRegistry()
+        ds = Datastore(cluster, source_registry=registry)
+        rec = InferencePoolReconciler(cluster, ds)
+        cluster.create(InferencePool(
+            metadata=ObjectMeta(name="pool", namespace="ns"),
+            selector={"a": "b"},
+        ))
+        rec.reconcile("ns", "pool")
+        assert ds.pool_get("ns", "pool") is not None
+        src = ds.pool_source("ns", "pool")
+        assert src is not None and registry.get(src.name()) is not None
+
+        cluster.delete("InferencePool", "ns", "pool")
+        rec.reconcile("ns", "pool")
+        assert ds.pool_get("ns", "pool") is None
+        assert registry.get(src.name()) is None
