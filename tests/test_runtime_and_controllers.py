@@ -517,3 +517,36 @@ class TestInferencePoolReconciler:
         rec.reconcile("ns", "pool")
         assert ds.pool_get("ns", "pool") is None
         assert registry.get(src.name()) is None
+
+
+class TestDedupWorkQueue:
+    def test_burst_collapses_to_one(self):
+        from wva_amd.runtime.manager import DedupWorkQueue
+
+        q = DedupWorkQueue()
+        fn = lambda ns, n: None  # noqa: E731
+        for _ in range(50):
+            q.put((fn, "ns", "a"))
+        q.put((fn, "ns", "b"))
+        assert q.get(0.1) == (fn, "ns", "a")
+        assert q.get(0.1) == (fn, "ns", "b")
+        assert q.get(0.05) is None  # the 49 duplicates collapsed
+
+    def test_readd_after_pickup_allowed(self):
+        from wva_amd.runtime.manager import DedupWorkQueue
+
+        q = DedupWorkQueue()
+        fn = lambda ns, n: None  # noqa: E731
+        q.put((fn, "ns", "a"))
+        item = q.get(0.1)
+        q.put(item)  # re-add while "processing" — must queue again
+        assert q.get(0.1) == item
+
+    def test_fifo_order_preserved(self):
+        from wva_amd.runtime.manager import DedupWorkQueue
+
+        q = DedupWorkQueue()
+        fn = lambda ns, n: None  # noqa: E731
+        for name in ("x", "y", "z"):
+            q.put((fn, "ns", name))
+        assert [q.get(0.1)[2] for _ in range(3)] == ["x", "y", "z"]

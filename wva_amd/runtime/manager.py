@@ -32,6 +32,41 @@ ReconcileFunc = Callable[[str, str], None]  # (namespace, name)
 MapFunc = Callable[[WatchEvent], Optional[Tuple[str, str]]]
 
 
+class DedupWorkQueue:
+    """controller-runtime workqueue semantics: an item already queued
+    is not queued again (dedup while PENDING); once a worker picks it
+    up it may be re-added — a burst of watch events for one object
+    collapses into at most one queued + one in-flight reconcile."""
+
+    def __init__(self) -> None:
+        self._lock = threading.Lock()
+        self._cond = threading.Condition(self._lock)
+        self._order: List[Tuple[ReconcileFunc, str, str]] = []
+        self._pending: set = set()
+
+    def put(self, item: Tuple[ReconcileFunc, str, str]) -> None:
+        with self._cond:
+            if item in self._pending:
+                return
+            self._pending.add(item)
+            self._order.append(item)
+            self._cond.notify()
+
+    def get(self, timeout: float) -> Optional[Tuple[ReconcileFunc, str, str]]:
+        with self._cond:
+            if not self._order:
+                self._cond.wait(timeout)
+            if not self._order:
+                return None
+            item = self._order.pop(0)
+            self._pending.discard(item)
+            return item
+
+    def empty(self) -> bool:
+        with self._lock:
+            return not self._order
+
+
 class Runnable(Protocol):
     def start(self) -> None: ...
     def stop(self) -> None: ...
@@ -139,7 +174,7 @@ class Manager:
         self._registrations: List[_Registration] = []
         self._runnables: List[Runnable] = []
         self._va_reconcile: Optional[ReconcileFunc] = None
-        self._work: "queue.Queue[Tuple[ReconcileFunc, str, str]]" = queue.Queue()
+        self._work = DedupWorkQueue()
         self._stop = threading.Event()
         self._threads: List[threading.Thread] = []
         self._watch_q: Optional["queue.Queue[WatchEvent]"] = None
@@ -286,10 +321,10 @@ class Manager:
 
     def _worker_loop(self) -> None:
         while not self._stop.is_set():
-            try:
-                reconcile, ns, name = self._work.get(timeout=0.2)
-            except queue.Empty:
+            item = self._work.get(timeout=0.2)
+            if item is None:
                 continue
+            reconcile, ns, name = item
             try:
                 reconcile(ns, name)
             except Exception as e:  # noqa: BLE001
