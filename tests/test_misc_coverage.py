@@ -527,3 +527,81 @@ class TestDebugThreadsEndpoint:
             srv.stop()
         assert "--- thread" in text
         assert "MainThread" in text
+
+
+class TestLocalDiscoveryParsing:
+    """discovery/local.py parser units with canned amd-smi / rocm-smi
+    shapes (probe-based on hardware; here the parsing + label mapping
+    are pinned)."""
+
+    def test_product_label_mapping(self):
+        from wva_amd.discovery.local import LocalGPU, _product_label
+
+        assert _product_label(LocalGPU(0, "AMD Instinct MI355X", 294912)) \
+            == "AMD-Instinct-MI355X-288GB"
+        assert _product_label(LocalGPU(0, "AMD Instinct MI300X", 196608)) \
+            == "AMD-Instinct-MI300X-192GB"
+        # no MI-token: gfx950 / 288GB-class heuristics
+        assert _product_label(
+            LocalGPU(0, "AMD Radeon Graphics", 294912, gfx_arch="gfx950")
+        ) == "AMD-Instinct-MI355X-288GB"
+        assert _product_label(LocalGPU(0, "Some GPU", 16384)) \
+            == "AMD-SOME-GPU-16GB"
+
+    def test_amd_smi_json_parsing(self, monkeypatch):
+        import json as _json
+        import wva_amd.discovery.local as local
+
+        payload = _json.dumps([{
+            "asic": {"market_name": "AMD Instinct MI355X",
+                     "target_graphics_version": "gfx950"},
+            "vram": {"size": {"value": 294912, "unit": "MB"}},
+        }] * 8)
+
+        class R:
+            stdout = payload
+
+        monkeypatch.setattr(local.shutil, "which", lambda b: "/usr/bin/" + b)
+        monkeypatch.setattr(
+            local.subprocess, "run", lambda *a, **k: R()
+        )
+        gpus = local.discover_via_amd_smi()
+        assert len(gpus) == 8
+        assert gpus[0].memory_mib == 294912
+        assert gpus[0].gfx_arch == "gfx950"
+        labels = local.node_labels_for_local_gpus(gpus)
+        assert labels == {
+            "amd.com/gpu.product": "AMD-Instinct-MI355X-288GB",
+            "amd.com/gpu.memory": "294912",
+            "amd.com/gpu.count": "8",
+        }
+
+    def test_rocm_smi_json_parsing(self, monkeypatch):
+        import json as _json
+        import wva_amd.discovery.local as local
+
+        payload = _json.dumps({
+            "card0": {"Card Series": "AMD Instinct MI355X",
+                      "VRAM Total Memory (B)": 294912 * 1024 * 1024},
+            "card1": {"Card Series": "AMD Instinct MI355X",
+                      "VRAM Total Memory (B)": 294912 * 1024 * 1024},
+        })
+
+        class R:
+            stdout = payload
+
+        monkeypatch.setattr(local.shutil, "which", lambda b: "/usr/bin/" + b)
+        monkeypatch.setattr(local.subprocess, "run", lambda *a, **k: R())
+        gpus = local.discover_via_rocm_smi()
+        assert len(gpus) == 2 and gpus[1].index == 1
+        assert gpus[0].memory_mib == 294912
+
+    def test_degrades_to_empty(self, monkeypatch):
+        import wva_amd.discovery.local as local
+
+        monkeypatch.setattr(local.shutil, "which", lambda b: None)
+        monkeypatch.setattr(
+            local, "discover_via_torch", lambda: None
+        )
+        assert local.discover_local_gpus() == []
+        assert local.node_labels_for_local_gpus([]) == {}
