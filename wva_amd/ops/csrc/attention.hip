@@ -803,7 +803,7 @@ __global__ __launch_bounds__(256, 2) void gqa_decode_attn_v5_kernel(
 // through LDS. A single-wave merge looping over ~224 splits serially
 // was HALF the long-context step time (profiles/longctx_kernel_stats
 // .txt pre-fix: 101 µs vs the 115 µs attention sweep).
-#define MERGE_WAVES 4
+#define MERGE_WAVES 8
 __global__ __launch_bounds__(64 * MERGE_WAVES) void gqa_decode_attn_merge_kernel(
     bf16* __restrict__ out,          // [B, Hq, 128]
     const float* __restrict__ workspace,  // [B, Hk, G, splits, 2+128]
