@@ -59,6 +59,13 @@ def main():
     lc.reset(4, LCTX)
     engines["llama8b_longctx64k"] = (lc, lc.decode_step, 4, LCTX, LCTX + 2048)
 
+    # 128k single-sequence engine: the deepest split-KV + 8-wave-merge
+    # path under sustained load
+    XCTX = 131072
+    xl = LlamaDecodeModel(LLAMA_3_8B, max_batch=1, max_seq=XCTX + 2048)
+    xl.reset(1, XCTX)
+    engines["llama8b_ctx128k"] = (xl, xl.decode_step, 1, XCTX, XCTX + 2048)
+
     # CPU-side control plane on the emulated cluster, ticking between bursts
     from prometheus_client import CollectorRegistry
 
