@@ -236,3 +236,108 @@ class TestMoEDensePath:
         torch.testing.assert_close(
             dense.float(), sparse.float(), atol=3e-2, rtol=3e-2
         )
+
+
+class TestEmulatorPhysics:
+    """The bench substrate must obey queueing physics — these pin the
+    emulator against first principles, independent of any analyzer."""
+
+    def _stack(self, replicas=2, **prof_kwargs):
+        from wva_amd.emulator.cluster_sim import ClusterSim
+        from wva_amd.emulator.vllm_sim import ServiceProfile
+        from wva_amd.kube.fake import FakeCluster
+        from wva_amd.kube.objects import (
+            Container, Deployment, PodTemplateSpec,
+        )
+        from wva_amd.api.types import ObjectMeta
+
+        cluster = FakeCluster()
+        cluster.create(Deployment(
+            metadata=ObjectMeta(name="v", namespace="ns"),
+            replicas=replicas,
+            selector={"app": "v"},
+            template=PodTemplateSpec(
+                labels={"app": "v"},
+                containers=[Container(requests={"amd.com/gpu": "1"})],
+            ),
+        ))
+        prof = ServiceProfile(**{
+            "alpha_ms": 10.0, "beta_ms": 1.0, "max_num_seqs": 16,
+            "num_gpu_blocks": 50_000, **prof_kwargs,
+        })
+        sim = ClusterSim(cluster, warm_start=True, seed=3)
+        sim.register_variant("m", "ns", "v", prof)
+        sim.reconcile_deployments()
+        return sim, sim.model("m", "ns"), prof
+
+    def _run(self, sim, model, qps, seconds, dt=0.25):
+        from wva_amd.emulator.workload import constant_qps
+
+        prof = constant_qps(qps)
+        for _ in range(int(seconds / dt)):
+            sim.generate_arrivals(model, prof, dt, 100, 50)
+            sim.advance(dt)
+
+    def test_steady_state_throughput_matches_offered(self):
+        """Under-loaded system: completions/s ≈ offered qps (flow
+        conservation)."""
+        sim, model, prof = self._stack(replicas=2)
+        self._run(sim, model, qps=5.0, seconds=60)
+        n0, t0 = len(model.completed), sim.now
+        self._run(sim, model, qps=5.0, seconds=120)
+        rate = (len(model.completed) - n0) / (sim.now - t0)
+        assert rate == pytest.approx(5.0, rel=0.15)
+
+    def test_overload_throughput_capped_at_service_rate(self):
+        """Overloaded replicas complete at their saturated service rate
+        (B/ITL(B)/O per replica), regardless of offered load."""
+        sim, model, prof = self._stack(replicas=1)
+        cap = prof.max_num_seqs / (
+            prof.itl_ms(prof.max_num_seqs) / 1000.0
+        ) / 50.0
+        self._run(sim, model, qps=cap * 4, seconds=30)
+        n0, t0 = len(model.completed), sim.now
+        self._run(sim, model, qps=cap * 4, seconds=60)
+        rate = (len(model.completed) - n0) / (sim.now - t0)
+        assert rate == pytest.approx(cap, rel=0.2)
+        assert rate < cap * 1.2
+
+    def test_ttft_grows_under_backlog(self):
+        sim, model, prof = self._stack(replicas=1)
+        self._run(sim, model, qps=1.0, seconds=30)
+        light = [c.ttft for c in model.completed[-10:]]
+        cap = prof.max_num_seqs / (
+            prof.itl_ms(prof.max_num_seqs) / 1000.0
+        ) / 50.0
+        self._run(sim, model, qps=cap * 3, seconds=60)
+        heavy = [c.ttft for c in model.completed[-10:]]
+        assert sum(heavy) / len(heavy) > 5 * (sum(light) / len(light))
+
+    def test_itl_reflects_batch_occupancy(self):
+        """Measured per-request ITL ≈ the profile's ITL at the running
+        occupancy (the service curve actually drives the sim)."""
+        sim, model, prof = self._stack(replicas=1, max_num_seqs=8)
+        self._run(sim, model, qps=1.0, seconds=90)
+        recents = model.completed[-15:]
+        itls = [c.itl * 1000 for c in recents if c.spec.output_tokens > 1]
+        mean_itl = sum(itls) / len(itls)
+        # light load: occupancy between 1 and ~4 → ITL within the curve
+        assert prof.itl_ms(1) * 0.9 <= mean_itl <= prof.itl_ms(8) * 1.1
+
+    def test_kv_capacity_bounds_admission(self):
+        """Tiny KV pool: concurrent tokens never exceed capacity (the
+        k1 constraint the V2 analyzer scales on)."""
+        sim, model, prof = self._stack(
+            replicas=1, num_gpu_blocks=100, max_num_seqs=64
+        )  # 1600-token pool vs ~150-token requests
+        cap = prof.kv_capacity_tokens
+        peak = 0
+        from wva_amd.emulator.workload import constant_qps
+
+        w = constant_qps(50.0)
+        for _ in range(200):
+            sim.generate_arrivals(model, w, 0.25, 100, 50)
+            sim.advance(0.25)
+            for _pod, (rep, _r, _d, _n) in list(sim.replicas.items()):
+                peak = max(peak, sum(r.kv_tokens() for r in rep.running))
+        assert 0 < peak <= cap

@@ -222,9 +222,24 @@ class ReplicaSim:
             self._start_times[id(candidate)] = now
             self.running.append(candidate)
 
+    def _preempt_if_over_capacity(self) -> None:
+        """vLLM recompute-style preemption: when decode growth pushes
+        the running set past the KV pool, the NEWEST request is evicted
+        back to the head of the waiting queue and must re-prefill its
+        prompt + generated-so-far on re-admission (its KV blocks free
+        immediately). Keeps kv_tokens_in_use ≤ capacity like the real
+        engine — the admission watermark alone only bounds the INPUT
+        footprint."""
+        cap = self.profile.kv_capacity_tokens
+        while len(self.running) > 1 and self.kv_tokens_in_use() > cap:
+            victim = self.running.pop()
+            victim.prefill_remaining = float(victim.kv_tokens())
+            self.waiting.appendleft(victim)
+
     def step(self, now: float, dt: float) -> List[CompletedRequest]:
         """Advance the replica by dt seconds of simulated time."""
         self._try_admit(now)
+        self._preempt_if_over_capacity()
         finished: List[CompletedRequest] = []
         if self.running:
             # prefill first (chunked-prefill approximation: prefill shares
@@ -262,6 +277,10 @@ class ReplicaSim:
             r for r in self.running if r.generated < r.spec.output_tokens
         ]
         self._try_admit(now + dt)
+        # decode growth during this step may have crossed the pool —
+        # preempt before the state is observed (scrapes see ≤ capacity,
+        # exactly like the real engine's per-iteration scheduler)
+        self._preempt_if_over_capacity()
         self.observe_peaks()
         return finished
 
