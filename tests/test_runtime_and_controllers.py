@@ -428,6 +428,40 @@ class TestConfigLoader:
                 require_prometheus=False,
             )
 
+    def test_invalid_bind_address_rejected(self):
+        with pytest.raises(ConfigLoadError):
+            load_config(env={"METRICS_BIND_ADDRESS": "not a bind addr"},
+                        require_prometheus=False)
+
+    def test_prometheus_url_scheme_enforced(self):
+        with pytest.raises(ConfigLoadError):
+            load_config(env={"PROMETHEUS_BASE_URL": "ftp://prom:9090"})
+        with pytest.raises(ConfigLoadError):
+            load_config(env={})  # required when prometheus is required
+        cfg = load_config(env={"PROMETHEUS_BASE_URL": "https://prom:9090"})
+        assert cfg.prometheus.base_url == "https://prom:9090"
+
+    def test_nonpositive_intervals_rejected(self):
+        with pytest.raises(ConfigLoadError):
+            load_config(env={"GLOBAL_OPT_INTERVAL": "0s"},
+                        require_prometheus=False)
+        with pytest.raises(ConfigLoadError):
+            load_config(env={"PROMETHEUS_METRICS_CACHE_TTL": "-5s",
+                             "PROMETHEUS_BASE_URL": "http://p:9090"})
+
+    def test_all_errors_aggregated_in_one_raise(self):
+        """Fail-fast startup reports EVERY problem at once
+        (validation.go aggregation), not just the first."""
+        with pytest.raises(ConfigLoadError) as ei:
+            load_config(env={
+                "METRICS_BIND_ADDRESS": "bogus addr",
+                "GLOBAL_OPT_INTERVAL": "0s",
+            }, require_prometheus=True)
+        msg = str(ei.value)
+        assert "METRICS_BIND_ADDRESS" in msg
+        assert "GLOBAL_OPT_INTERVAL" in msg
+        assert "PROMETHEUS_BASE_URL" in msg
+
 
 class TestInferencePoolReconciler:
     """Reference inferencepool_reconciler.go:41-118 + pool.go:34-148:
