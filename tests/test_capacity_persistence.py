@@ -182,3 +182,48 @@ class TestEngineIntegration:
         )
         assert cm is not None
         assert f"{ENS}|{EMODEL}|{VARIANT}" in cm.data["records"]
+
+
+class TestK2HistoryPersistence:
+    def test_history_rides_with_capacity_records(self):
+        from wva_amd.analyzers.saturation_v2 import SaturationAnalyzerV2
+        from wva_amd.kube.fake import FakeCluster
+
+        cluster = FakeCluster()
+        store = CapacityKnowledgeStore()
+        an = SaturationAnalyzerV2(store)
+        # seed a learned k2 history entry
+        an._compute_capacity_history = {}
+        from wva_amd.analyzers.history import RollingAverage
+
+        ra = RollingAverage(10)
+        for v in (30_000.0, 31_000.0, 32_000.0):
+            ra.add(v)
+        an._compute_capacity_history["m|MI355X|le500"] = ra
+        store.update(NS, MODEL, "v0", live_record())
+
+        p = CapacityStorePersistence(cluster, store, NS,
+                                     write_interval_seconds=0.0,
+                                     analyzer=an)
+        assert p.maybe_persist() is True
+
+        # restart: a fresh analyzer restores the history with its ages
+        store2 = CapacityKnowledgeStore()
+        an2 = SaturationAnalyzerV2(store2)
+        p2 = CapacityStorePersistence(cluster, store2, NS, analyzer=an2)
+        restored = p2.restore()
+        assert restored >= 2  # 1 record + 1 history key
+        ra2 = an2._compute_capacity_history["m|MI355X|le500"]
+        assert ra2.average() == pytest.approx(31_000.0)
+
+    def test_local_history_not_overwritten(self):
+        from wva_amd.analyzers.saturation_v2 import SaturationAnalyzerV2
+
+        an = SaturationAnalyzerV2(CapacityKnowledgeStore())
+        from wva_amd.analyzers.history import RollingAverage
+
+        ra = RollingAverage(10)
+        ra.add(99.0)
+        an._compute_capacity_history["k"] = ra
+        assert an.history_restore({"k": {"values": [1.0], "age_seconds": 0}}) == 0
+        assert an._compute_capacity_history["k"].average() == 99.0

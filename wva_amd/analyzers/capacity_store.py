@@ -359,11 +359,13 @@ class CapacityStorePersistence:
     """Periodic ConfigMap writer + bootstrap reader for the store."""
 
     def __init__(self, cluster, store: CapacityKnowledgeStore,
-                 namespace: str, write_interval_seconds: float = 60.0):
+                 namespace: str, write_interval_seconds: float = 60.0,
+                 analyzer=None):
         self.cluster = cluster
         self.store = store
         self.namespace = namespace
         self.write_interval_seconds = write_interval_seconds
+        self.analyzer = analyzer  # optional: k2 history rides along
         self._last_write = 0.0
         self._last_payload = ""
 
@@ -380,7 +382,14 @@ class CapacityStorePersistence:
             snap = json.loads(cm.data.get("records", "{}"))
         except ValueError:
             return 0
-        return restore_store(self.store, {"records": snap})
+        n = restore_store(self.store, {"records": snap})
+        if self.analyzer is not None and "history" in cm.data:
+            try:
+                hist = json.loads(cm.data["history"])
+            except ValueError:
+                hist = {}
+            n += self.analyzer.history_restore(hist)
+        return n
 
     def maybe_persist(self) -> bool:
         """Write the snapshot if the interval elapsed and something
@@ -403,6 +412,11 @@ class CapacityStorePersistence:
         from ..api.types import ObjectMeta
         from ..kube.objects import ConfigMap
 
+        data = {"records": payload}
+        if self.analyzer is not None:
+            data["history"] = json.dumps(
+                self.analyzer.history_snapshot(), sort_keys=True
+            )
         cm = self.cluster.try_get(
             "ConfigMap", self.namespace, CAPACITY_STORE_CONFIG_MAP_NAME
         )
@@ -413,10 +427,10 @@ class CapacityStorePersistence:
                         name=CAPACITY_STORE_CONFIG_MAP_NAME,
                         namespace=self.namespace,
                     ),
-                    data={"records": payload},
+                    data=data,
                 ))
             else:
-                cm.data = {"records": payload}
+                cm.data = data
                 self.cluster.update(cm)
         except Exception:  # noqa: BLE001 — best-effort (incl. 409s)
             return False

@@ -141,6 +141,47 @@ class SaturationAnalyzerV2:
     def name(self) -> str:
         return "saturation-token-based"
 
+    def history_snapshot(self) -> dict:
+        """Serializable k2 rolling-history state (checkpoint/resume
+        companion to the capacity store: SURVEY §5 lists both as
+        restart-lost)."""
+        import time as _time
+
+        now = _time.monotonic()
+        with self._lock:
+            return {
+                key: {
+                    "values": list(ra._values),
+                    "age_seconds": max(now - ra.last_updated, 0.0),
+                }
+                for key, ra in self._compute_capacity_history.items()
+            }
+
+    def history_restore(self, snap: dict) -> int:
+        """Restore k2 history; never overwrites locally observed keys;
+        ages preserved for the 24 h eviction."""
+        import time as _time
+
+        from .history import RollingAverage
+        from .capacity_store import ROLLING_AVERAGE_WINDOW_SIZE
+
+        now = _time.monotonic()
+        n = 0
+        with self._lock:
+            for key, d in (snap or {}).items():
+                if key in self._compute_capacity_history:
+                    continue
+                try:
+                    ra = RollingAverage(ROLLING_AVERAGE_WINDOW_SIZE)
+                    for v in d.get("values", []):
+                        ra._values.append(float(v))
+                    ra.last_updated = now - float(d.get("age_seconds", 0.0))
+                except (TypeError, ValueError):
+                    continue
+                self._compute_capacity_history[key] = ra
+                n += 1
+        return n
+
     def evict_stale_history(self, timeout_seconds: float) -> int:
         with self._lock:
             now = time.monotonic()

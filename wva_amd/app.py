@@ -235,7 +235,8 @@ def build_app(
     from .controllers.configmap import controller_namespace
 
     persistence = CapacityStorePersistence(
-        cluster, capacity_store, controller_namespace()
+        cluster, capacity_store, controller_namespace(),
+        analyzer=saturation_engine.v2_analyzer,
     )
     restored = persistence.restore()
     if restored:
