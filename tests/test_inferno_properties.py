@@ -319,3 +319,77 @@ class TestSolverProperties:
             cand = srv.all_allocations.get(srv.allocation.accelerator)
             assert cand is not None
             assert srv.allocation.num_replicas <= cand.num_replicas
+
+
+class TestV1AnalyzerProperties:
+    """Property sweep over the V1 percentage analyzer (reference
+    internal/saturation/analyzer.go:31-280): threshold monotonicity and
+    scale-down safety invariants across generated replica states."""
+
+    def _analyze(self, kvs_qs):
+        from wva_amd.analyzers.interfaces import ReplicaMetrics
+        from wva_amd.analyzers.saturation_v1 import SaturationAnalyzerV1
+        from wva_amd.config.saturation import SaturationScalingConfig
+
+        metrics = [
+            ReplicaMetrics(pod_name=f"p{i}", kv_cache_usage=kv,
+                           queue_length=q, variant_name="v",
+                           accelerator_name="MI355X", cost=10.0)
+            for i, (kv, q) in enumerate(kvs_qs)
+        ]
+        return SaturationAnalyzerV1().analyze_model_saturation(
+            "m", "ns", metrics, SaturationScalingConfig()
+        )
+
+    @given(st.lists(
+        st.tuples(st.floats(min_value=0, max_value=1),
+                  st.integers(min_value=0, max_value=20)),
+        min_size=1, max_size=8,
+    ))
+    @settings(max_examples=200, deadline=None)
+    def test_saturation_classification_consistent(self, kvs_qs):
+        """A replica is saturated iff kv >= 0.80 OR q >= 5 (defaults);
+        the analysis counts must agree with the definition."""
+        a = self._analyze(kvs_qs)
+        expected_sat = sum(
+            1 for kv, q in kvs_qs if kv >= 0.80 or q >= 5
+        )
+        assert a.total_replicas == len(kvs_qs)
+        assert a.total_replicas - a.non_saturated_count == expected_sat
+
+    @given(st.lists(
+        st.tuples(st.floats(min_value=0, max_value=0.79),
+                  st.integers(min_value=0, max_value=4)),
+        min_size=2, max_size=8,
+    ))
+    @settings(max_examples=200, deadline=None)
+    def test_scale_down_safety_monotone_in_load(self, kvs_qs):
+        """If scale-down is UNSAFE at some load, it stays unsafe when
+        every replica's load increases (monotonicity of the N/(N−1)
+        redistribution simulation, analyzer.go:233-280)."""
+        a_low = self._analyze(kvs_qs)
+        heavier = [
+            (min(kv + 0.1, 0.79), min(q + 1, 4)) for kv, q in kvs_qs
+        ]
+        a_high = self._analyze(heavier)
+        if not a_low.scale_down_safe:
+            assert not a_high.scale_down_safe
+
+    @given(st.lists(
+        st.tuples(st.floats(min_value=0, max_value=1),
+                  st.integers(min_value=0, max_value=20)),
+        min_size=1, max_size=8,
+    ))
+    @settings(max_examples=200, deadline=None)
+    def test_never_up_and_down_together(self, kvs_qs):
+        a = self._analyze(kvs_qs)
+        assert not (a.should_scale_up and a.scale_down_safe)
+
+    @given(st.integers(min_value=1, max_value=8))
+    @settings(max_examples=50, deadline=None)
+    def test_idle_fleet_scale_down_safe_when_big_enough(self, n):
+        """All-idle replicas: scale-down safe iff >= 2 non-saturated
+        (MinNonSaturatedReplicasForScaleDown, constants.go:8)."""
+        a = self._analyze([(0.0, 0)] * n)
+        assert a.scale_down_safe == (n >= 2)
+        assert not a.should_scale_up
