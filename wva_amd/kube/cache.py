@@ -44,6 +44,7 @@ DEFAULT_CACHED_KINDS = [
     "Deployment",
     "Pod",
     "Node",
+    "Namespace",  # read per VA event by the exclude-annotation predicate
     "Service",
     "ConfigMap",
     "InferencePool",
