@@ -341,3 +341,94 @@ class TestEmulatorPhysics:
             for _pod, (rep, _r, _d, _n) in list(sim.replicas.items()):
                 peak = max(peak, sum(r.kv_tokens() for r in rep.running))
         assert 0 < peak <= cap
+
+
+class TestRecomputePreemption:
+    """vLLM recompute-style preemption: decode growth past the KV pool
+    evicts the newest request, which re-prefills prompt + generated on
+    re-admission (ReplicaSim._preempt_if_over_capacity)."""
+
+    def _overcommitted(self):
+        # 320 token slots; two requests admitted at input=140 each (280
+        # < 0.98·320) then decode growth pushes past capacity
+        r = replica(seqs=8, blocks=20)
+        # 320 slots: each request peaks at 140+60=200 (fits alone) but the
+        # pair's decode growth crosses 320, forcing preemption
+        for i in range(2):
+            r.submit(RequestSpec(
+                input_tokens=140, output_tokens=60, arrival_time=0.0
+            ))
+        return r
+
+    def test_kv_never_exceeds_capacity(self):
+        r = self._overcommitted()
+        t = 0.0
+        for _ in range(200):
+            r.step(t, 0.05)
+            assert r.kv_tokens_in_use() <= r.profile.kv_capacity_tokens
+            t += 0.05
+
+    def test_victim_requeued_and_re_prefills_generated(self):
+        r = self._overcommitted()
+        t = 0.0
+        preempted = None
+        for _ in range(200):
+            r.step(t, 0.05)
+            t += 0.05
+            if r.waiting and r.waiting[0].generated > 0:
+                preempted = r.waiting[0]
+                break
+        assert preempted is not None, "over-committed pair never preempted"
+        # eviction charges re-prefill for prompt + generated-so-far
+        assert preempted.prefill_remaining == pytest.approx(
+            float(preempted.spec.input_tokens + preempted.generated)
+        )
+
+    def test_readmission_keeps_recompute_prefill(self):
+        """_try_admit must not overwrite the eviction-assigned
+        prefill_remaining with prompt-only (or prefix-cache-skipped)
+        prefill, and must project admission on kv_tokens()."""
+        r = replica(seqs=8, blocks=100)  # 1600 slots, plenty of room
+        r.profile.prefix_cache_hit_rate = 1.0  # would zero a fresh prefill
+        from wva_amd.emulator.vllm_sim import _RunningRequest
+
+        victim = _RunningRequest(
+            spec=RequestSpec(input_tokens=100, output_tokens=50), generated=30
+        )
+        victim.prefill_remaining = 130.0  # as _preempt_if_over_capacity sets
+        r.waiting.append(victim)
+        queries_before = r.prefix_cache_queries
+        r._try_admit(now=1.0)
+        assert victim in r.running
+        assert victim.prefill_remaining == 130.0
+        # recomputed tokens never hit the prefix cache
+        assert r.prefix_cache_queries == queries_before
+
+    def test_victim_completes_and_ttft_counted_once(self):
+        r = self._overcommitted()
+        t = 0.0
+        done = []
+        for _ in range(4000):
+            done += r.step(t, 0.05)
+            t += 0.05
+            if len(done) == 2:
+                break
+        assert len(done) == 2
+        assert r.ttft_count == 2  # no double TTFT on re-admission
+
+    def test_sole_runner_never_preempted(self):
+        # len(running) > 1 guard: a single request over capacity keeps
+        # running (evicting it would livelock)
+        r = replica(seqs=8, blocks=8)  # 128 slots < 100+100 peak
+        r.submit(RequestSpec(input_tokens=100, output_tokens=100,
+                             arrival_time=0.0))
+        t = 0.0
+        done = []
+        for _ in range(400):
+            done = r.step(t, 0.05)
+            if done:
+                break
+            assert r.num_requests_running() == 1
+            assert r.num_requests_waiting() == 0
+            t += 0.05
+        assert len(done) == 1  # ran to completion despite exceeding pool

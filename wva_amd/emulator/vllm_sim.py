@@ -207,19 +207,25 @@ class ReplicaSim:
     def _try_admit(self, now: float) -> None:
         while self.waiting and len(self.running) < self.profile.max_num_seqs:
             candidate = self.waiting[0]
-            projected = self.kv_tokens_in_use() + candidate.spec.input_tokens
+            # a preemption victim re-acquires prompt + generated-so-far
+            projected = self.kv_tokens_in_use() + candidate.kv_tokens()
             # vLLM-style watermark: keep a little headroom for decode growth
             if projected > 0.98 * self.profile.kv_capacity_tokens:
                 break
             self.waiting.popleft()
-            # prefix cache: a hit fraction of prompt tokens skips prefill
-            self.prefix_cache_queries += 1
-            hit = random.random() < self.profile.prefix_cache_hit_rate
-            if hit:
-                self.prefix_cache_hits += 1
-            effective_prefill = candidate.spec.input_tokens * (0.0 if hit else 1.0)
-            candidate.prefill_remaining = effective_prefill
-            self._start_times[id(candidate)] = now
+            if candidate.generated == 0:
+                # prefix cache: a hit fraction of prompt tokens skips prefill
+                self.prefix_cache_queries += 1
+                hit = random.random() < self.profile.prefix_cache_hit_rate
+                if hit:
+                    self.prefix_cache_hits += 1
+                candidate.prefill_remaining = (
+                    candidate.spec.input_tokens * (0.0 if hit else 1.0)
+                )
+            # else: re-admission after recompute preemption — keep the
+            # prefill_remaining the eviction assigned (prompt + generated;
+            # recomputed tokens never hit the prefix cache)
+            self._start_times.setdefault(id(candidate), now)
             self.running.append(candidate)
 
     def _preempt_if_over_capacity(self) -> None:
