@@ -67,8 +67,6 @@ def main():
     engines["llama8b_ctx128k"] = (xl, xl.decode_step, 1, XCTX, XCTX + 2048)
 
     # CPU-side control plane on the emulated cluster, ticking between bursts
-    from prometheus_client import CollectorRegistry
-
     sys.path.insert(0, os.path.join(
         os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tests"
     ))

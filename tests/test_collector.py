@@ -5,7 +5,6 @@ Mirrors reference prometheus_source_test.go (450 LoC) /
 pod_scraping_source_test.go (956) / replica_metrics coverage.
 """
 import json
-import math
 import threading
 import time
 from http.server import BaseHTTPRequestHandler, HTTPServer

@@ -26,14 +26,12 @@ from wva_amd.config.config import Config
 from wva_amd.config.saturation import SaturationScalingConfig
 from wva_amd.emulator.cluster_sim import ClusterSim
 from wva_amd.emulator.sim_source import SimMetricsSource
-from wva_amd.emulator.vllm_sim import RequestSpec, ServiceProfile
+from wva_amd.emulator.vllm_sim import ServiceProfile
 from wva_amd.emulator.workload import constant_qps
 from wva_amd.kube.fake import FakeCluster
 from wva_amd.kube.objects import (
     Container,
     Deployment,
-    EndpointPicker,
-    EndpointPool,
     InferencePool,
     Node,
     PodTemplateSpec,

@@ -23,7 +23,7 @@ from wva_amd.api.types import (
 from wva_amd.emulator.vllm_sim import ServiceProfile
 from wva_amd.kube.objects import Container, Deployment, PodTemplateSpec
 
-from test_e2e_emulated import MODEL, NS, VARIANT, make_stack, mi355x_node, run_sim
+from test_e2e_emulated import MODEL, NS, VARIANT, make_stack, run_sim
 
 OVERLOAD_PROFILE = dict(
     alpha_ms=50.0, beta_ms=2.0, max_num_seqs=8, num_gpu_blocks=500

@@ -13,7 +13,6 @@ Parity: reference pkg/solver/{solver,greedy,optimizer}.go —
 from __future__ import annotations
 
 import bisect
-import math
 import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
