@@ -410,3 +410,109 @@ class TestServiceClassConfigMap:
         assert t.model == "default/default"
         assert t.slo_itl == 24.0
         assert t.slo_ttft == 500.0
+
+
+class TestCapacitySpec:
+    def test_wire_shape(self):
+        from wva_amd.inferno.types import CapacitySpec
+
+        spec = CapacitySpec.from_dict({"count": {"MI355X": 16, "MI300X": "8"}})
+        assert spec.counts == {"MI355X": 16, "MI300X": 8}
+
+    def test_bare_map_and_garbage(self):
+        from wva_amd.inferno.types import CapacitySpec
+
+        assert CapacitySpec.from_dict({"MI355X": 4}).counts == {"MI355X": 4}
+        assert CapacitySpec.from_dict(None).counts == {}
+        assert CapacitySpec.from_dict({"count": "x"}).counts == {}
+        assert CapacitySpec.from_dict(
+            {"count": {"MI355X": "not-a-number"}}
+        ).counts == {}
+
+    def test_system_data_uses_it(self):
+        data = SystemData.from_dict({
+            "spec": {"capacity": {"count": {"MI355X": 12}}},
+        })
+        assert data.capacity == {"MI355X": 12}
+
+
+class TestModelAnalyzer:
+    """Reference pkg/analyzer ModelAnalyzer parity: per-model feasible
+    allocations across every accelerator, independent of the solver."""
+
+    def _va(self):
+        from wva_amd.api.types import (
+            ObjectMeta, VariantAutoscaling, VariantAutoscalingSpec,
+        )
+
+        return VariantAutoscaling(
+            metadata=ObjectMeta(name="srv-a", namespace="default"),
+            spec=VariantAutoscalingSpec(model_id="llama-8b"),
+        )
+
+    def test_feasible_on_both_accelerators(self):
+        from wva_amd.analyzers.modelanalyzer import ModelAnalyzer
+
+        sys_ = mi355x_system()
+        out = ModelAnalyzer(sys_).analyze_model(
+            self._va(), arrival_rate_per_min=40.0,
+            avg_in_tokens=100, avg_out_tokens=50,
+            service_class="premium",
+        )
+        assert set(out) == {"MI355X", "MI300X"}
+        for alloc in out.values():
+            assert alloc.num_replicas >= 1
+            assert alloc.itl <= 24.0 * 1.05  # meets the premium SLO
+
+    def test_infeasible_accelerator_omitted(self):
+        """An accelerator whose floor ITL (α+β at batch 1) already breaks
+        the SLO yields no allocation for that accelerator."""
+        from wva_amd.analyzers.modelanalyzer import ModelAnalyzer
+
+        data = SystemData(
+            accelerators=[
+                AcceleratorSpec(name="MI355X", type="MI355X", cost=50.0),
+                AcceleratorSpec(name="SLOW", type="SLOW", cost=1.0),
+            ],
+            models=[
+                ModelAcceleratorPerfData(
+                    name="llama-8b", acc="MI355X", acc_count=1,
+                    max_batch_size=256, at_tokens=50,
+                    service_parms=ServiceParmsSpec(alpha=11.28, beta=0.0152),
+                ),
+                ModelAcceleratorPerfData(
+                    name="llama-8b", acc="SLOW", acc_count=1,
+                    max_batch_size=256, at_tokens=50,
+                    service_parms=ServiceParmsSpec(alpha=100.0, beta=5.0),
+                ),
+            ],
+            service_classes=[
+                ServiceClassSpec(
+                    name="premium", priority=1,
+                    model_targets=[ModelTarget(
+                        model="llama-8b", slo_itl=24.0, slo_ttft=500.0,
+                    )],
+                )
+            ],
+            capacity={"MI355X": 16, "SLOW": 16},
+        )
+        from wva_amd.analyzers.modelanalyzer import ModelAnalyzer as MA
+
+        out = MA(System(data)).analyze_model(
+            self._va(), arrival_rate_per_min=40.0,
+            avg_in_tokens=100, avg_out_tokens=50,
+            service_class="premium",
+        )
+        assert "MI355X" in out
+        assert "SLOW" not in out
+
+    def test_higher_load_needs_more_replicas(self):
+        from wva_amd.analyzers.modelanalyzer import ModelAnalyzer
+
+        lo = ModelAnalyzer(mi355x_system()).analyze_model(
+            self._va(), 40.0, 100, 50, service_class="premium",
+        )["MI355X"].num_replicas
+        hi = ModelAnalyzer(mi355x_system()).analyze_model(
+            self._va(), 100_000.0, 100, 50, service_class="premium",
+        )["MI355X"].num_replicas
+        assert hi > lo

@@ -234,9 +234,25 @@ class ServerSpec:
 
 @dataclass
 class CapacitySpec:
-    """Available accelerator units per type."""
+    """Available accelerator units per type (wire shape
+    ``{"count": {"MI355X": 16}}``; a bare type→count map is accepted)."""
 
     counts: Dict[str, int] = field(default_factory=dict)
+
+    @classmethod
+    def from_dict(cls, d: Any) -> "CapacitySpec":
+        if not isinstance(d, dict):
+            return cls()
+        src = d.get("count", d)
+        if not isinstance(src, dict):
+            return cls()
+        counts: Dict[str, int] = {}
+        for k, v in src.items():
+            try:
+                counts[str(k)] = int(v)
+            except (TypeError, ValueError):
+                continue
+        return cls(counts=counts)
 
 
 @dataclass
@@ -292,11 +308,7 @@ class SystemData:
             ] if isinstance(spec.get("servers"), dict) else [
                 ServerSpec.from_dict(s) for s in spec.get("servers", [])
             ],
-            capacity=dict(
-                (spec.get("capacity") or {}).get("count", {})
-                if isinstance(spec.get("capacity"), dict)
-                else {}
-            ),
+            capacity=CapacitySpec.from_dict(spec.get("capacity") or {}).counts,
             optimizer=OptimizerSpec.from_dict(spec.get("optimizer") or {}),
         )
 
