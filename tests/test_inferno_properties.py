@@ -393,3 +393,125 @@ class TestV1AnalyzerProperties:
         a = self._analyze([(0.0, 0)] * n)
         assert a.scale_down_safe == (n >= 2)
         assert not a.should_scale_up
+
+
+class TestTunerRobustness:
+    """EKF tuner structural invariants under arbitrary observation
+    streams (the reference tuner is dormant and untested against hostile
+    inputs; here it is live on the engine path, so it must never emit
+    NaN/negative parameters no matter what the collector feeds it)."""
+
+    def _tuner(self, alpha, beta, gamma):
+        from wva_amd.inferno.tuner import ServiceParmsTuner, TunerConfig
+
+        return ServiceParmsTuner(
+            ServiceParms(alpha=alpha, beta=beta, gamma=gamma),
+            TunerConfig(max_batch_size=32, max_queue_size=128),
+        )
+
+    observations = st.lists(
+        st.tuples(
+            st.floats(min_value=0.0, max_value=1e4),    # request_rate
+            st.floats(min_value=1.0, max_value=4096.0),  # avg_input
+            st.floats(min_value=1.0, max_value=1024.0),  # avg_output
+            st.floats(min_value=0.0, max_value=1e6),     # ttft_ms
+            st.floats(min_value=0.0, max_value=1e5),     # itl_ms
+            st.booleans(),                               # itl_only
+        ),
+        min_size=1,
+        max_size=8,
+    )
+
+    @settings(max_examples=40, deadline=None)
+    @given(
+        seed=st.tuples(
+            st.floats(min_value=0.5, max_value=100.0),   # alpha
+            st.floats(min_value=1e-4, max_value=1.0),    # beta
+            st.floats(min_value=0.0, max_value=1e-3),    # gamma
+        ),
+        stream=observations,
+    )
+    def test_theta_always_finite_and_physical(self, seed, stream):
+        import numpy as np
+        from wva_amd.inferno.tuner import Observation
+
+        tuner = self._tuner(*seed)
+        for rate, avg_in, avg_out, ttft, itl, itl_only in stream:
+            tuner.update(
+                Observation(
+                    request_rate=rate,
+                    avg_input_tokens=avg_in,
+                    avg_output_tokens=avg_out,
+                    ttft_ms=ttft,
+                    itl_ms=itl,
+                ),
+                itl_only=itl_only,
+            )
+            assert np.all(np.isfinite(tuner.theta))
+            assert np.all(np.isfinite(tuner.P))
+            p = tuner.parms()
+            assert p.alpha >= 1e-6
+            assert p.beta >= 0.0
+            assert p.gamma >= 0.0
+            # covariance stays symmetric with non-negative diagonal
+            assert np.allclose(tuner.P, tuner.P.T, atol=1e-6)
+            assert np.all(np.diag(tuner.P) >= -1e-9)
+
+    @settings(max_examples=30, deadline=None)
+    @given(
+        seed=st.tuples(
+            st.floats(min_value=1.0, max_value=50.0),
+            st.floats(min_value=1e-3, max_value=0.5),
+            st.floats(min_value=0.0, max_value=1e-4),
+        ),
+        stream=observations,
+    )
+    def test_accepted_step_respects_trust_region(self, seed, stream):
+        import numpy as np
+        from wva_amd.inferno.tuner import Observation
+
+        tuner = self._tuner(*seed)
+        frac = tuner.config.max_step_frac
+        for rate, avg_in, avg_out, ttft, itl, itl_only in stream:
+            before = tuner.theta.copy()
+            accepted = tuner.update(
+                Observation(
+                    request_rate=rate,
+                    avg_input_tokens=avg_in,
+                    avg_output_tokens=avg_out,
+                    ttft_ms=ttft,
+                    itl_ms=itl,
+                ),
+                itl_only=itl_only,
+            )
+            if accepted:
+                limit = np.maximum(np.abs(before) * frac, [0.5, 1e-3, 1e-3])
+                # +tiny: the non-negativity projection can only shrink
+                assert np.all(np.abs(tuner.theta - before) <= limit + 1e-12)
+            else:
+                assert np.array_equal(tuner.theta, before)
+
+    @settings(max_examples=20, deadline=None)
+    @given(
+        seed=st.tuples(
+            st.floats(min_value=1.0, max_value=50.0),
+            st.floats(min_value=1e-3, max_value=0.5),
+            st.floats(min_value=0.0, max_value=1e-4),
+        ),
+    )
+    def test_model_consistent_observation_accepted_and_stable(self, seed):
+        """Feeding the tuner its OWN prediction is a zero-innovation
+        observation: always accepted, θ unchanged."""
+        import numpy as np
+        from wva_amd.inferno.tuner import Observation
+
+        tuner = self._tuner(*seed)
+        obs = Observation(
+            request_rate=1.0, avg_input_tokens=100, avg_output_tokens=50,
+            ttft_ms=0.0, itl_ms=0.0,
+        )
+        pred = tuner._h(tuner.theta, obs)
+        obs.ttft_ms, obs.itl_ms = float(pred[0]), float(pred[1])
+        before = tuner.theta.copy()
+        assert tuner.update(obs) is True
+        assert np.allclose(tuner.theta, before, rtol=1e-6, atol=1e-9)

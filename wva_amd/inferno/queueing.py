@@ -106,15 +106,21 @@ class MM1KModel:
         self._compute_statistics()
 
     def _compute_probabilities(self) -> None:
+        """Geometric p_i ∝ rho^i, anchored at the distribution's dominant
+        end so no intermediate overflows: for rho ≤ 1 work up from p_0,
+        for rho > 1 down from p_K with (1/rho)^(K-i) (rho^(K+1) overflows
+        float64 once rho^(K+1) > 1e308, e.g. rho≈40 at K≈190 — found by
+        the hypothesis suite). Explicit normalization also removes the
+        cancellation of (1-rho)/(1-rho^(K+1)) near rho = 1."""
         K, rho = self.K, self.rho
-        if rho == 1:
-            p0 = 1.0 / (K + 1)
+        if rho <= 1:
+            q = [rho**i for i in range(K + 1)]
         else:
-            p0 = (1 - rho) / (1 - rho ** (K + 1))
-        total = 0.0
+            r_inv = 1.0 / rho
+            q = [r_inv ** (K - i) for i in range(K + 1)]
+        total = sum(q)
         for i in range(K + 1):
-            self.p[i] = p0 * rho**i
-            total += self.p[i]
+            self.p[i] = q[i] / total
 
     def _compute_statistics(self) -> None:
         if not self.is_valid:
