@@ -1,0 +1,196 @@
+"""Property tests for the remaining numerical/structural primitives:
+RFC 7386 merge patch (the #731 status-patch path), the measured-ITL
+interpolation profile, and the 3-parameter service-surface fit.
+"""
+import math
+
+from hypothesis import assume, given, settings, strategies as st
+
+from wva_amd.kube.openapi import merge_patch
+
+
+# --- RFC 7386 merge patch ---
+
+JSON = st.recursive(
+    st.none() | st.booleans() | st.integers(-1000, 1000)
+    | st.text(max_size=8),
+    lambda children: st.lists(children, max_size=3)
+    | st.dictionaries(st.text(max_size=6), children, max_size=3),
+    max_leaves=12,
+)
+
+
+def spec_merge(target, patch):
+    """Literal transcription of RFC 7386 §2 pseudocode."""
+    if isinstance(patch, dict):
+        if not isinstance(target, dict):
+            target = {}
+        else:
+            target = dict(target)
+        for name, value in patch.items():
+            if value is None:
+                target.pop(name, None)
+            else:
+                target[name] = spec_merge(target.get(name), value)
+        return target
+    return patch
+
+
+class TestMergePatchRFC7386:
+    @settings(max_examples=300, deadline=None)
+    @given(target=JSON, patch=JSON)
+    def test_matches_spec_pseudocode(self, target, patch):
+        assert merge_patch(target, patch) == spec_merge(target, patch)
+
+    @settings(max_examples=200, deadline=None)
+    @given(target=JSON, patch=JSON)
+    def test_idempotent(self, target, patch):
+        once = merge_patch(target, patch)
+        assert merge_patch(once, patch) == once
+
+    @settings(max_examples=200, deadline=None)
+    @given(target=JSON, patch=JSON)
+    def test_target_not_mutated(self, target, patch):
+        import copy
+
+        snapshot = copy.deepcopy(target)
+        merge_patch(target, patch)
+        assert target == snapshot
+
+    @settings(max_examples=200, deadline=None)
+    @given(
+        target=st.dictionaries(st.text(max_size=6), JSON, max_size=4),
+        patch=st.dictionaries(
+            st.text(max_size=6),
+            st.none() | st.integers(-10, 10),
+            max_size=4,
+        ),
+    )
+    def test_null_deletes_everything_it_names(self, target, patch):
+        out = merge_patch(target, patch)
+        for k, v in patch.items():
+            if v is None:
+                assert k not in out
+            else:
+                assert out[k] == v
+
+
+# --- measured ITL-table interpolation ---
+
+ITL_TABLE = st.lists(
+    st.tuples(
+        st.integers(min_value=1, max_value=512),
+        st.floats(min_value=0.1, max_value=500.0),
+    ),
+    min_size=2,
+    max_size=8,
+    unique_by=lambda p: p[0],
+)
+
+
+class TestITLTableInterpolation:
+    @settings(max_examples=200, deadline=None)
+    @given(table=ITL_TABLE)
+    def test_passes_through_measured_points(self, table):
+        from wva_amd.emulator.vllm_sim import ServiceProfile
+
+        prof = ServiceProfile.from_itl_table(
+            [b for b, _ in table], [t for _, t in table]
+        )
+        for b, t in table:
+            assert math.isclose(prof.itl_ms(b), t, rel_tol=1e-12)
+
+    @settings(max_examples=200, deadline=None)
+    @given(table=ITL_TABLE)
+    def test_permutation_invariant(self, table):
+        from wva_amd.emulator.vllm_sim import ServiceProfile
+
+        a = ServiceProfile.from_itl_table(
+            [b for b, _ in table], [t for _, t in table]
+        )
+        rev = list(reversed(table))
+        b_ = ServiceProfile.from_itl_table(
+            [b for b, _ in rev], [t for _, t in rev]
+        )
+        assert a.itl_table == b_.itl_table
+        for q in (1, 3, 17, 100, 600):
+            assert a.itl_ms(q) == b_.itl_ms(q)
+
+    @settings(max_examples=200, deadline=None)
+    @given(table=ITL_TABLE, q=st.integers(min_value=1, max_value=512))
+    def test_within_range_bounded_by_neighbors(self, table, q):
+        from wva_amd.emulator.vllm_sim import ServiceProfile
+
+        table = sorted(table)
+        assume(table[0][0] <= q <= table[-1][0])
+        prof = ServiceProfile.from_itl_table(
+            [b for b, _ in table], [t for _, t in table]
+        )
+        v = prof.itl_ms(q)
+        for (b0, t0), (b1, t1) in zip(table, table[1:]):
+            if b0 <= q <= b1:
+                assert min(t0, t1) - 1e-9 <= v <= max(t0, t1) + 1e-9
+                break
+
+    @settings(max_examples=100, deadline=None)
+    @given(table=ITL_TABLE)
+    def test_derived_linear_parms_nonnegative(self, table):
+        from wva_amd.emulator.vllm_sim import ServiceProfile
+
+        prof = ServiceProfile.from_itl_table(
+            [b for b, _ in table], [t for _, t in table]
+        )
+        assert prof.alpha_ms >= 0.0
+        assert prof.beta_ms >= 0.0
+
+
+# --- 3-parameter service-surface fit ---
+
+class TestSurfaceFitProperties:
+    grids = st.lists(
+        st.tuples(
+            st.integers(min_value=1, max_value=256),      # batch
+            st.integers(min_value=128, max_value=65536),  # ctx
+        ),
+        min_size=4,
+        max_size=12,
+        unique=True,
+    )
+
+    @settings(max_examples=100, deadline=None)
+    @given(
+        grid=grids,
+        alpha=st.floats(min_value=0.5, max_value=50.0),
+        beta=st.floats(min_value=1e-3, max_value=1.0),
+        gamma=st.floats(min_value=0.0, max_value=1e-3),
+    )
+    def test_recovers_exact_planar_data(self, grid, alpha, beta, gamma):
+        from wva_amd.calibration.itl_benchmark import fit_itl_surface
+
+        # need rank-3 design: at least two distinct batches and two
+        # distinct n·ctx products
+        assume(len({b for b, _ in grid}) >= 2)
+        assume(len({b * c for b, c in grid}) >= 3)
+        pts = [(b, c, alpha + b * (beta + gamma * c)) for b, c in grid]
+        a, be, ga, r2 = fit_itl_surface(pts)
+        if r2 > 0.999999:  # well-conditioned draw
+            assert math.isclose(a, alpha, rel_tol=1e-3, abs_tol=1e-2)
+            assert math.isclose(be, beta, rel_tol=1e-3, abs_tol=1e-3)
+
+    @settings(max_examples=100, deadline=None)
+    @given(
+        grid=grids,
+        noise=st.lists(
+            st.floats(min_value=-5.0, max_value=5.0), min_size=12, max_size=12
+        ),
+    )
+    def test_outputs_always_physical(self, grid, noise):
+        from wva_amd.calibration.itl_benchmark import fit_itl_surface
+
+        pts = [
+            (b, c, max(0.1, 10.0 + noise[i % len(noise)]))
+            for i, (b, c) in enumerate(grid)
+        ]
+        a, be, ga, r2 = fit_itl_surface(pts)
+        assert be >= 0.0 and ga >= 0.0
+        assert math.isfinite(a) and math.isfinite(r2)
