@@ -78,12 +78,6 @@ class CostAwareOptimizer:
         sorted_vcs = sorted(
             result.variant_capacities, key=lambda vc: vc.cost, reverse=True
         )
-        cheapest = ""
-        min_cost = _INF
-        for vc in result.variant_capacities:
-            if vc.cost < min_cost:
-                min_cost = vc.cost
-                cheapest = vc.variant_name
         remaining = result.spare_capacity
         for vc in sorted_vcs:
             if remaining <= 0:
@@ -91,13 +85,18 @@ class CostAwareOptimizer:
             if vc.per_replica_capacity <= 0:
                 continue
             current = targets.get(vc.variant_name, 0)
-            min_replicas = 0
-            if vc.variant_name == cheapest:
-                other_has = any(
-                    t > 0 for name, t in targets.items() if name != cheapest
-                )
-                if not other_has:
-                    min_replicas = 1
+            # spare-driven scale-down never drains the model to zero —
+            # that transition belongs to the enforcer's scale-to-zero
+            # policy. Removal order is most-expensive-first, so the
+            # floor lands on the cheapest variant still holding
+            # replicas. (Protecting a precomputed "cheapest" NAME is
+            # wrong when the cheapest variant has no replicas — e.g.
+            # cost ties — and would let the last real holder drain.)
+            other_has = any(
+                t > 0 for name, t in targets.items()
+                if name != vc.variant_name
+            )
+            min_replicas = 0 if other_has else 1
             removable = current - min_replicas
             if removable <= 0:
                 continue
