@@ -336,3 +336,108 @@ class TestCostAwareOptimizerProperties:
                 assert d.target_replicas > d.current_replicas
             else:
                 assert d.target_replicas == d.current_replicas
+
+
+# --- GPU allocation (TypeAllocator + GreedyBySaturation) invariants ---
+
+from wva_amd.analyzers.interfaces import VariantDecision
+from wva_amd.pipeline.greedy_saturation import GreedyBySaturation
+from wva_amd.pipeline.inventory import TypeAllocator
+
+alloc_decisions = st.lists(
+    st.tuples(
+        st.integers(min_value=0, max_value=8),    # current replicas
+        st.integers(min_value=0, max_value=12),   # extra replicas wanted
+        st.integers(min_value=1, max_value=8),    # gpus per replica
+        st.floats(min_value=-1e5, max_value=1e5),  # spare capacity
+        st.floats(min_value=1.0, max_value=100.0),  # cost
+    ),
+    min_size=1,
+    max_size=6,
+)
+
+
+def _decisions(rows, acc="MI355X"):
+    out = []
+    for i, (cur, extra, gpr, spare, cost) in enumerate(rows):
+        out.append(VariantDecision(
+            variant_name=f"v{i}", namespace="ns", model_id="m",
+            accelerator_name=acc, cost=cost,
+            current_replicas=cur, target_replicas=cur + extra,
+            gpus_per_replica=gpr, spare_capacity=spare,
+        ))
+    return out
+
+
+class TestAllocationProperties:
+    @settings(max_examples=200, deadline=None)
+    @given(rows=alloc_decisions, pool=st.integers(min_value=0, max_value=64))
+    def test_pool_conserved_and_whole_replicas(self, rows, pool):
+        decisions = _decisions(rows)
+        allocator = TypeAllocator({"MI355X": pool})
+        GreedyBySaturation().allocate(decisions, allocator)
+        total_allocated = 0
+        for i, d in enumerate(decisions):
+            cur, extra, gpr, _, _ = rows[i]
+            # whole-replica granularity, no partial-replica leak
+            assert d.gpus_allocated % gpr == 0
+            assert d.gpus_allocated == (d.target_replicas - cur) * gpr
+            # never scales beyond the ask, never below current
+            assert cur <= d.target_replicas <= cur + extra
+            if d.target_replicas < cur + extra and extra > 0:
+                assert d.was_limited
+            total_allocated += d.gpus_allocated
+        assert total_allocated <= pool
+        assert allocator.remaining() == pool - total_allocated
+        assert allocator.remaining_for_type("MI355X") >= 0
+
+    @settings(max_examples=200, deadline=None)
+    @given(rows=alloc_decisions, pool=st.integers(min_value=1, max_value=64))
+    def test_most_saturated_candidate_served_first(self, rows, pool):
+        decisions = _decisions(rows)
+        allocator = TypeAllocator({"MI355X": pool})
+        GreedyBySaturation().allocate(decisions, allocator)
+        # order candidates as the algorithm does; once one is limited,
+        # every later one gets nothing more than pool leftovers allow —
+        # in particular a FULLY limited candidate (0 allocated despite
+        # wanting some) means everyone later with >= its gpus/replica
+        # got 0 too
+        cands = sorted(
+            (d for i, d in enumerate(decisions)
+             if d.target_replicas + (0) >= 0 and rows[i][1] > 0),
+            key=lambda d: (d.spare_capacity, d.cost),
+        )
+        seen_starved_gpr = None
+        for d in cands:
+            if seen_starved_gpr is not None \
+                    and d.gpus_per_replica >= seen_starved_gpr:
+                assert d.gpus_allocated == 0
+            if d.gpus_allocated == 0:
+                if seen_starved_gpr is None \
+                        or d.gpus_per_replica < seen_starved_gpr:
+                    seen_starved_gpr = d.gpus_per_replica
+
+    @settings(max_examples=100, deadline=None)
+    @given(rows=alloc_decisions,
+           hive=st.integers(min_value=1, max_value=8),
+           pool=st.integers(min_value=8, max_value=64))
+    def test_hive_constraint_rejects_oversized_replicas(self, rows, pool,
+                                                        hive):
+        decisions = _decisions(rows)
+        allocator = TypeAllocator({"MI355X": pool}, {"MI355X": hive})
+        GreedyBySaturation().allocate(decisions, allocator)
+        for i, d in enumerate(decisions):
+            cur, extra, gpr, _, _ = rows[i]
+            if gpr > hive:
+                assert d.gpus_allocated == 0
+                assert d.target_replicas == cur
+
+    @settings(max_examples=100, deadline=None)
+    @given(rows=alloc_decisions, pool=st.integers(min_value=0, max_value=64))
+    def test_wrong_type_pool_untouched(self, rows, pool):
+        decisions = _decisions(rows, acc="MI300X")
+        allocator = TypeAllocator({"MI355X": pool})
+        GreedyBySaturation().allocate(decisions, allocator)
+        assert allocator.remaining_for_type("MI355X") == pool
+        for d in decisions:
+            assert d.gpus_allocated == 0

@@ -52,6 +52,12 @@ class TypeAllocator:
         if available <= 0:
             return 0
         allocated = min(gpus_requested, available)
+        # whole-replica granularity: a partial replica's GPUs would be
+        # deducted from the pool but unusable (the algorithm truncates to
+        # whole replicas), silently starving later candidates in the
+        # same batch
+        gpr = max(decision.gpus_per_replica, 1)
+        allocated -= allocated % gpr
         self._remaining[acc_type] = available - allocated
         self._total -= allocated
         return allocated
