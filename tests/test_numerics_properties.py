@@ -441,3 +441,123 @@ class TestAllocationProperties:
         assert allocator.remaining_for_type("MI355X") == pool
         for d in decisions:
             assert d.gpus_allocated == 0
+
+
+# --- V2 token analyzer invariants ---
+
+from wva_amd.analyzers.capacity_store import CapacityKnowledgeStore
+from wva_amd.analyzers.interfaces import AnalyzerInput, ReplicaMetrics
+from wva_amd.analyzers.saturation_v2 import SaturationAnalyzerV2
+from wva_amd.config.saturation import SaturationScalingConfig
+
+replica_metrics = st.lists(
+    st.tuples(
+        st.integers(min_value=0, max_value=2),           # variant index
+        st.integers(min_value=0, max_value=3_000_000),   # kv capacity tokens
+        st.floats(min_value=0.0, max_value=1.0),         # kv usage frac
+        st.integers(min_value=0, max_value=50),          # queue length
+        st.floats(min_value=0.0, max_value=4096.0),      # avg input tokens
+        st.floats(min_value=0.0, max_value=1024.0),      # avg output tokens
+    ),
+    min_size=0,
+    max_size=8,
+)
+
+v2_config = st.tuples(
+    st.floats(min_value=0.05, max_value=1.0),  # scale_up_threshold
+    st.floats(min_value=0.05, max_value=1.0),  # scale_down_boundary
+    st.floats(min_value=0.1, max_value=1.0),   # kv_cache_threshold
+)
+
+
+def _v2_input(metrics_rows, cfg_row):
+    up, down, kv = cfg_row
+    cfg = SaturationScalingConfig(analyzer_name="saturation-v2")
+    cfg.apply_defaults()
+    cfg.scale_up_threshold = max(up, down)  # validated invariant: up > down
+    cfg.scale_down_boundary = min(up, down) * 0.99
+    cfg.kv_cache_threshold = kv
+    rms, states = [], {}
+    from wva_amd.analyzers.interfaces import VariantReplicaState
+
+    for i, (vi, cap, usage, qlen, avg_in, avg_out) in enumerate(metrics_rows):
+        name = f"var{vi}"
+        rms.append(ReplicaMetrics(
+            pod_name=f"p{i}", variant_name=name, namespace="ns",
+            model_id="m", accelerator_name="MI355X",
+            kv_cache_usage=usage, queue_length=qlen,
+            total_kv_capacity_tokens=cap,
+            tokens_in_use=int(usage * cap),
+            avg_input_tokens=avg_in, avg_output_tokens=avg_out,
+        ))
+        s = states.setdefault(name, VariantReplicaState(variant_name=name))
+        s.current_replicas += 1
+    return AnalyzerInput(
+        model_id="m", namespace="ns", replica_metrics=rms,
+        variant_states=list(states.values()), config=cfg,
+    )
+
+
+class TestV2AnalyzerProperties:
+    @settings(max_examples=200, deadline=None)
+    @given(rows=replica_metrics, cfg=v2_config)
+    def test_result_always_finite_and_consistent(self, rows, cfg):
+        inp = _v2_input(rows, cfg)
+        res = SaturationAnalyzerV2(CapacityKnowledgeStore()).analyze(inp)
+        assert math.isfinite(res.total_supply) and res.total_supply >= 0
+        assert math.isfinite(res.total_demand) and res.total_demand >= 0
+        assert math.isfinite(res.utilization) and res.utilization >= 0
+        assert res.required_capacity >= 0.0
+        assert res.spare_capacity >= 0.0
+        # supply equals the sum over variants
+        assert math.isclose(
+            res.total_supply,
+            sum(vc.total_capacity for vc in res.variant_capacities),
+            rel_tol=1e-9, abs_tol=1e-6,
+        )
+        # utilization is exactly demand/supply (or 0 on empty supply)
+        if res.total_supply > 0:
+            assert math.isclose(
+                res.utilization, res.total_demand / res.total_supply,
+                rel_tol=1e-9,
+            )
+        else:
+            assert res.utilization == 0.0
+
+    @settings(max_examples=200, deadline=None)
+    @given(rows=replica_metrics, cfg=v2_config)
+    def test_never_up_and_down_at_once(self, rows, cfg):
+        """required > 0 and spare > 0 simultaneously would make the
+        optimizer direction ambiguous; the up>down config invariant
+        forbids it whenever there are no pending replicas (anticipated
+        supply == current supply)."""
+        inp = _v2_input(rows, cfg)
+        res = SaturationAnalyzerV2(CapacityKnowledgeStore()).analyze(inp)
+        if res.required_capacity > 1e-9:
+            assert res.spare_capacity <= 1e-6 * max(res.total_supply, 1.0)
+
+    @settings(max_examples=100, deadline=None)
+    @given(rows=replica_metrics, cfg=v2_config)
+    def test_per_replica_capacity_bounded_by_k1(self, rows, cfg):
+        """Effective capacity = min(k1, k2) ≤ k1 = kv_threshold × total."""
+        inp = _v2_input(rows, cfg)
+        res = SaturationAnalyzerV2(CapacityKnowledgeStore()).analyze(inp)
+        caps_by_variant = {}
+        for vi, cap, *_ in rows:
+            caps_by_variant.setdefault(f"var{vi}", []).append(cap)
+        kv = inp.config.kv_cache_threshold
+        for vc in res.variant_capacities:
+            caps = caps_by_variant.get(vc.variant_name)
+            if caps and vc.per_replica_capacity > 0:
+                assert vc.per_replica_capacity <= max(caps) * kv + 1.0
+
+    @settings(max_examples=100, deadline=None)
+    @given(cfg=v2_config)
+    def test_empty_input_is_calm(self, cfg):
+        """No replicas → no demand, no supply, no scaling pressure."""
+        inp = _v2_input([], cfg)
+        res = SaturationAnalyzerV2(CapacityKnowledgeStore()).analyze(inp)
+        assert res.total_supply == 0.0
+        assert res.total_demand == 0.0
+        assert res.required_capacity == 0.0
+        assert res.spare_capacity == 0.0
