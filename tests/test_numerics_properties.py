@@ -561,3 +561,123 @@ class TestV2AnalyzerProperties:
         assert res.total_demand == 0.0
         assert res.required_capacity == 0.0
         assert res.spare_capacity == 0.0
+
+
+# --- Enforcer fail-safe invariants ---
+
+from wva_amd.analyzers.interfaces import VariantSaturationAnalysis
+from wva_amd.config.scale_to_zero import ModelScaleToZeroConfig
+from wva_amd.pipeline.enforcer import Enforcer
+
+targets_maps = st.dictionaries(
+    st.sampled_from(["v-a", "v-b", "v-c"]),
+    st.integers(min_value=0, max_value=10),
+    min_size=1,
+    max_size=3,
+)
+
+
+def _s2z(enabled):
+    return {"default": ModelScaleToZeroConfig(
+        enable_scale_to_zero=enabled, retention_period="10m",
+    )}
+
+
+class TestEnforcerProperties:
+    @settings(max_examples=150, deadline=None)
+    @given(targets=targets_maps,
+           count=st.floats(min_value=0.0, max_value=1e6))
+    def test_scale_to_zero_only_on_exact_zero_traffic(self, targets, count):
+        enf = Enforcer(lambda m, ns, r: count)
+        out, applied = enf.enforce_policy(
+            "m", "ns", dict(targets), [], _s2z(True),
+        )
+        if count > 0:
+            assert out == targets
+            assert applied is False
+        else:
+            assert all(v == 0 for v in out.values())
+            assert applied is True
+
+    @settings(max_examples=150, deadline=None)
+    @given(targets=targets_maps)
+    def test_query_error_never_zeroes(self, targets):
+        def boom(m, ns, r):
+            raise RuntimeError("prometheus down")
+
+        out, applied = Enforcer(boom).enforce_policy(
+            "m", "ns", dict(targets), [], _s2z(True),
+        )
+        assert out == targets  # fail safe: keep whatever the optimizer said
+        assert applied is False
+
+    @settings(max_examples=150, deadline=None)
+    @given(targets=targets_maps,
+           costs=st.lists(st.floats(min_value=1.0, max_value=100.0),
+                          min_size=3, max_size=3))
+    def test_minimum_replica_floor_when_disabled(self, targets, costs):
+        analyses = [
+            VariantSaturationAnalysis(variant_name=n, cost=c)
+            for n, c in zip(["v-a", "v-b", "v-c"], costs)
+        ]
+        out, applied = Enforcer(lambda m, ns, r: 0.0).enforce_policy(
+            "m", "ns", dict(targets), analyses, _s2z(False),
+        )
+        assert sum(out.values()) >= 1  # never zero with s2z disabled
+        if sum(targets.values()) > 0:
+            assert out == targets and applied is False
+        else:
+            assert applied is True
+            # the floor landed on the cheapest variant in the target map
+            kept = [n for n, v in out.items() if v > 0]
+            assert len(kept) == 1
+            cost_of = {a.variant_name: a.cost for a in analyses}
+            best = min(
+                targets, key=lambda n: (cost_of.get(n, 10.0), n)
+            )
+            assert kept[0] == best
+
+
+# --- DedupWorkQueue semantics ---
+
+class TestDedupWorkQueueProperties:
+    @settings(max_examples=150, deadline=None)
+    @given(events=st.lists(
+        st.tuples(st.sampled_from(["a", "b", "c", "d"]),
+                  st.sampled_from(["ns1", "ns2"])),
+        min_size=1, max_size=30,
+    ))
+    def test_dedups_while_pending_and_preserves_first_seen_order(
+        self, events
+    ):
+        from wva_amd.runtime.manager import DedupWorkQueue
+
+        q = DedupWorkQueue()
+        fn = lambda ns, name: None  # noqa: E731
+        for name, ns in events:
+            q.put((fn, ns, name))
+        drained = []
+        while True:
+            item = q.get(timeout=0.0)
+            if item is None:
+                break
+            drained.append((item[2], item[1]))
+        # exactly the distinct items, in first-appearance order
+        seen, expected = set(), []
+        for e in events:
+            if e not in seen:
+                seen.add(e)
+                expected.append(e)
+        assert drained == expected
+        assert q.empty()
+
+    def test_readd_after_pickup_requeues(self):
+        from wva_amd.runtime.manager import DedupWorkQueue
+
+        q = DedupWorkQueue()
+        fn = lambda ns, name: None  # noqa: E731
+        q.put((fn, "ns", "x"))
+        item = q.get(timeout=0.0)
+        assert item is not None
+        q.put((fn, "ns", "x"))  # in-flight item may be queued again
+        assert q.get(timeout=0.0) == item
