@@ -681,3 +681,102 @@ class TestDedupWorkQueueProperties:
         assert item is not None
         q.put((fn, "ns", "x"))  # in-flight item may be queued again
         assert q.get(timeout=0.0) == item
+
+
+# --- vLLM args parsing + PromQL templating robustness ---
+
+from wva_amd.analyzers.deployment_parser import (
+    parse_vllm_args,
+    split_shell_string,
+)
+from wva_amd.collector.query_template import QueryTemplate, escape_promql_value
+from wva_amd.kube.objects import Container, Deployment, PodTemplateSpec
+
+
+class TestVLLMArgsParsing:
+    def test_multiline_block_scalar_command(self):
+        """`sh -c` with a YAML `|` block scalar: newlines and `\\`
+        continuations separate args like a real shell."""
+        cmd = (
+            "vllm serve meta-llama/Llama-3.1-8B \\\n"
+            "  --max-num-seqs 512 \\\n"
+            "\t--block-size 16\n"
+            "  --kv-cache-dtype fp8\r\n"
+        )
+        d = Deployment(
+            metadata=__import__(
+                "wva_amd.api.types", fromlist=["ObjectMeta"]
+            ).ObjectMeta(name="d", namespace="ns"),
+            template=PodTemplateSpec(containers=[
+                Container(name="vllm", command=["sh", "-c", cmd]),
+            ]),
+        )
+        p = parse_vllm_args(d)
+        assert p.max_num_seqs == 512
+        assert p.block_size == 16
+        assert p.kv_cache_dtype == "fp8"
+
+    @settings(max_examples=200, deadline=None)
+    @given(s=st.text(max_size=120))
+    def test_splitter_never_crashes_and_never_emits_empty(self, s):
+        toks = split_shell_string(s)
+        assert all(isinstance(t, str) and t for t in toks)
+        assert all(t != "\\" for t in toks)
+
+    @settings(max_examples=150, deadline=None)
+    @given(args=st.lists(st.text(max_size=40), max_size=12))
+    def test_arbitrary_container_args_never_crash(self, args):
+        from wva_amd.api.types import ObjectMeta
+
+        d = Deployment(
+            metadata=ObjectMeta(name="d", namespace="ns"),
+            template=PodTemplateSpec(containers=[
+                Container(name="vllm", args=list(args)),
+            ]),
+        )
+        p = parse_vllm_args(d)
+        assert p.block_size >= 1
+        assert p.max_num_seqs >= 1
+        assert p.tensor_parallel_size >= 1
+
+
+class TestPromQLTemplateSafety:
+    @settings(max_examples=300, deadline=None)
+    @given(value=st.text(max_size=60))
+    def test_escaped_value_cannot_break_out_of_label_matcher(self, value):
+        """Whatever a model/namespace name contains, the rendered query
+        keeps it inside the quoted label value: the escaped form has no
+        unescaped double quote."""
+        esc = escape_promql_value(value)
+        i, n = 0, len(esc)
+        while i < n:
+            if esc[i] == "\\":
+                i += 2  # escape consumes the next char
+                continue
+            assert esc[i] != '"'
+            i += 1
+
+    @settings(max_examples=200, deadline=None)
+    @given(value=st.text(max_size=60))
+    def test_render_roundtrip_preserves_value(self, value):
+        """Unescaping the rendered label value gives back the input —
+        escaping is injective (no two models collide onto one query)."""
+        t = QueryTemplate(
+            name="q", template='metric{model_name="{{.model}}"}',
+            params=["model"],
+        )
+        out = t.render({"model": value})
+        prefix, suffix = 'metric{model_name="', '"}'
+        assert out.startswith(prefix) and out.endswith(suffix)
+        inner = out[len(prefix):-len(suffix)]
+        unescaped = inner.replace('\\"', '"').replace("\\\\", "\\")
+        # precise unescape (left-to-right)
+        res, i = [], 0
+        while i < len(inner):
+            if inner[i] == "\\" and i + 1 < len(inner):
+                res.append(inner[i + 1])
+                i += 2
+            else:
+                res.append(inner[i])
+                i += 1
+        assert "".join(res) == value

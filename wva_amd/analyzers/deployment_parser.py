@@ -60,7 +60,10 @@ class VLLMEngineParams:
 
 def split_shell_string(s: str) -> List[str]:
     """Basic shell-like splitting with single/double quote support
-    (no escapes / expansion — same scope as the reference)."""
+    (no expansion — same scope as the reference). All unquoted
+    whitespace separates tokens — `sh -c` commands in YAML manifests
+    are typically `|` block scalars with newlines — and a bare `\\`
+    line continuation is dropped."""
     tokens: List[str] = []
     current: List[str] = []
     in_single = False
@@ -70,7 +73,7 @@ def split_shell_string(s: str) -> List[str]:
             in_single = not in_single
         elif ch == '"' and not in_single:
             in_double = not in_double
-        elif ch == " " and not in_single and not in_double:
+        elif ch in " \t\n\r" and not in_single and not in_double:
             if current:
                 tokens.append("".join(current))
                 current = []
@@ -78,7 +81,7 @@ def split_shell_string(s: str) -> List[str]:
             current.append(ch)
     if current:
         tokens.append("".join(current))
-    return tokens
+    return [t for t in tokens if t != "\\"]
 
 
 def _collect_args(command: List[str], args: List[str]) -> List[str]:
