@@ -164,3 +164,60 @@ class TestSerdeRoundtrips:
         for kind, (_, _, (prefix, plural, namespaced)) in serde.SERDE.items():
             path = serde.resource_path(kind, ns if namespaced else None, name)
             assert plural in path and path.startswith("/" + prefix.split("/")[0])
+
+
+class TestRemainingKindRoundtrips:
+    """InferencePool / Namespace / ServiceMonitor / Lease — the kinds the
+    controller watches but the original fuzz suite didn't cover."""
+
+    @settings(max_examples=100, deadline=None)
+    @given(name=NAME, ns=NAME, selector=LABELS,
+           port=st.integers(min_value=1, max_value=65535),
+           epp=NAME)
+    def test_inferencepool(self, name, ns, selector, port, epp):
+        from wva_amd.kube.objects import InferencePool
+
+        roundtrip(InferencePool(
+            metadata=ObjectMeta(name=name, namespace=ns),
+            selector=dict(selector), target_port=port,
+            epp_service_name=epp,
+        ))
+
+    @settings(max_examples=100, deadline=None)
+    @given(name=NAME, labels=LABELS,
+           annotations=st.dictionaries(LABEL_KEY, FREE_TEXT, max_size=3))
+    def test_namespace_with_exclusion_annotation(self, name, labels,
+                                                 annotations):
+        from wva_amd.kube.objects import Namespace
+
+        meta = ObjectMeta(name=name, labels=dict(labels))
+        meta.annotations = dict(annotations)
+        dec = roundtrip(Namespace(metadata=meta))
+        assert dec.metadata.annotations == dict(annotations)
+
+    @settings(max_examples=100, deadline=None)
+    @given(name=NAME, ns=NAME)
+    def test_servicemonitor_if_registered(self, name, ns):
+        try:
+            from wva_amd.kube.objects import ServiceMonitor
+        except ImportError:
+            pytest.skip("no ServiceMonitor kind")
+        roundtrip(ServiceMonitor(
+            metadata=ObjectMeta(name=name, namespace=ns),
+        ))
+
+    @settings(max_examples=100, deadline=None)
+    @given(name=NAME, ns=NAME, holder=NAME,
+           duration=st.integers(min_value=1, max_value=3600))
+    def test_lease(self, name, ns, holder, duration):
+        try:
+            from wva_amd.kube.objects import Lease
+        except ImportError:
+            pytest.skip("no Lease kind")
+        lease = Lease(
+            metadata=ObjectMeta(name=name, namespace=ns),
+            holder_identity=holder,
+            lease_duration_seconds=duration,
+        )
+        dec = roundtrip(lease)
+        assert dec.holder_identity == holder

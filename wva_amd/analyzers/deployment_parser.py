@@ -90,7 +90,6 @@ def _collect_args(command: List[str], args: List[str]) -> List[str]:
         base = all_args[i]
         if (
             base in ("/bin/sh", "/bin/bash", "sh", "bash")
-            and i + 2 < len(all_args) + 0
             and all_args[i + 1] == "-c"
             and i + 2 < len(all_args)
         ):
