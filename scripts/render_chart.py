@@ -267,9 +267,17 @@ def render_chart(chart_dir: str, overrides: Optional[dict] = None) -> List[dict]
             continue
         with open(os.path.join(tdir, name)) as f:
             rendered = render_template(f.read(), values)
-        for doc in yaml.safe_load_all(rendered):
-            if doc:
-                docs.append(doc)
+        try:
+            for doc in yaml.safe_load_all(rendered):
+                if doc:
+                    docs.append(doc)
+        except yaml.YAMLError as e:
+            # a value that breaks the document structure (stray quote,
+            # newline) must fail with the renderer's own error type, so
+            # callers distinguish "bad values" from a renderer crash
+            raise TemplateError(
+                f"{name}: rendered output is not valid YAML: {e}"
+            ) from e
     return docs
 
 
