@@ -780,3 +780,41 @@ class TestPromQLTemplateSafety:
                 res.append(inner[i])
                 i += 1
         assert "".join(res) == value
+
+
+# --- Prometheus exposition label escaping roundtrip ---
+
+from wva_amd.collector.pod_scraping_source import parse_prometheus_text
+
+
+def _escape_label_value(v: str) -> str:
+    """Exposition-format escape, per the Prometheus text format spec."""
+    return v.replace("\\", "\\\\").replace('"', '\\"').replace("\n", "\\n")
+
+
+class TestExpositionLabelRoundtrip:
+    @settings(max_examples=300, deadline=None)
+    @given(value=st.text(
+        alphabet=st.characters(blacklist_categories=("Cs",)), max_size=40,
+    ))
+    def test_escaped_label_value_roundtrips(self, value):
+        line = f'vllm:metric{{model="{_escape_label_value(value)}"}} 1.0\n'
+        vals = parse_prometheus_text(line)
+        assert len(vals) == 1
+        assert vals[0].labels["model"] == value
+
+    def test_escaped_backslash_before_n_is_not_newline(self):
+        # raw bytes: model="a\\n" → escaped backslash + literal n
+        vals = parse_prometheus_text('m{model="a\\\\n"} 1\n')
+        assert vals[0].labels["model"] == "a\\n"
+        # while a real \n escape IS a newline
+        vals = parse_prometheus_text('m{model="a\\nb"} 1\n')
+        assert vals[0].labels["model"] == "a\nb"
+
+    @settings(max_examples=150, deadline=None)
+    @given(text=st.text(max_size=200))
+    def test_garbage_never_crashes(self, text):
+        for mv in parse_prometheus_text(text):
+            assert math.isfinite(mv.value) or math.isnan(mv.value) or \
+                math.isinf(mv.value)
+            assert "__name__" in mv.labels

@@ -30,11 +30,38 @@ DEFAULT_MAX_CONCURRENT_SCRAPES = 10
 
 _SAMPLE_RE = re.compile(
     r'^(?P<name>[a-zA-Z_:][a-zA-Z0-9_:]*)'
-    r'(?:\{(?P<labels>[^}]*)\})?'
+    # label block: quoted strings are consumed atomically so a } (or
+    # anything else) INSIDE a label value cannot terminate the block
+    r'(?:\{(?P<labels>(?:[^"}]|"(?:[^"\\]|\\.)*")*)\})?'
     r'\s+(?P<value>[^\s]+)'
     r'(?:\s+(?P<ts>\d+))?$'
 )
 _LABEL_RE = re.compile(r'(\w+)="((?:[^"\\]|\\.)*)"')
+
+
+def _unescape_label_value(raw: str) -> str:
+    """Exposition-format label unescape (\\\\, \\", \\n) in ONE
+    left-to-right pass — sequential str.replace corrupts sequences like
+    ``\\\\n`` (escaped backslash + literal n), turning them into a
+    newline."""
+    out = []
+    i, n = 0, len(raw)
+    while i < n:
+        ch = raw[i]
+        if ch == "\\" and i + 1 < n:
+            nxt = raw[i + 1]
+            if nxt == "n":
+                out.append("\n")
+            elif nxt in ('"', "\\"):
+                out.append(nxt)
+            else:
+                out.append(ch)
+                out.append(nxt)
+            i += 2
+        else:
+            out.append(ch)
+            i += 1
+    return "".join(out)
 
 
 def parse_prometheus_text(text: str) -> List[MetricValue]:
@@ -42,7 +69,10 @@ def parse_prometheus_text(text: str) -> List[MetricValue]:
     __name__ labels."""
     now = time.time()
     values: List[MetricValue] = []
-    for line in text.splitlines():
+    # split on \n ONLY: the exposition format is newline-delimited, and
+    # str.splitlines() would also break lines at \x0b/\x1c-\x1e/ …,
+    # which are legal raw bytes inside a label value
+    for line in text.split("\n"):
         line = line.strip()
         if not line or line.startswith("#"):
             continue
@@ -53,10 +83,7 @@ def parse_prometheus_text(text: str) -> List[MetricValue]:
         raw_labels = m.group("labels")
         if raw_labels:
             for lm in _LABEL_RE.finditer(raw_labels):
-                labels[lm.group(1)] = (
-                    lm.group(2).replace('\\"', '"').replace("\\\\", "\\")
-                    .replace("\\n", "\n")
-                )
+                labels[lm.group(1)] = _unescape_label_value(lm.group(2))
         try:
             value = float(m.group("value"))
         except ValueError:
