@@ -35,7 +35,9 @@ from urllib.parse import parse_qs, urlparse
 
 from wva_amd.api.crd import variantautoscaling_crd
 from wva_amd.kube import serde
-from wva_amd.kube.fake import ConflictError, FakeCluster, NotFoundError
+from wva_amd.kube.fake import (
+    ConflictError, ExpiredError, FakeCluster, NotFoundError,
+)
 from wva_amd.kube.openapi import merge_patch, validate
 
 # path prefix → kind (built from the serde table)
@@ -176,7 +178,12 @@ class K8sTestServer:
                 # subscription — the race a real API server closes with
                 # its watch cache); rv < 0: live-only
                 if rv >= 0:
-                    q = outer.cluster.watch_since([kind], rv)
+                    try:
+                        q = outer.cluster.watch_since([kind], rv)
+                    except ExpiredError as e:
+                        return self._send(410, _status_body(
+                            410, "Expired", str(e)
+                        ))
                 else:
                     q = outer.cluster.watch([kind])
                 drop = threading.Event()

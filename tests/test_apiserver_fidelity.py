@@ -787,3 +787,76 @@ class TestCacheConsistencyUnderChurn:
         }
         assert cached == upstream
         cache.stop()
+
+
+class TestWatchHistoryExpiry:
+    """410 Gone handling (client-go Reflector semantics): a reconnect
+    whose resourceVersion predates the retained watch history must
+    RELIST, and a read cache must prune objects deleted during the gap
+    (their DELETED events are unrecoverable)."""
+
+    def test_watch_since_expired_raises(self):
+        from wva_amd.kube.fake import ExpiredError, FakeCluster
+
+        c = FakeCluster()
+        c.create(make_deployment("a"))
+        c.create(make_deployment("b"))
+        c.expire_watch_history()
+        with pytest.raises(ExpiredError):
+            c.watch_since(["Deployment"], 1)
+        # resuming at the current head is still fine
+        q = c.watch_since(["Deployment"], 10**9)
+        assert q.empty()
+
+    def test_server_returns_410_for_expired_rv(self, server, client):
+        client.create(make_deployment("a"))
+        client.create(make_deployment("b"))  # floor advances past rv=1
+        server.cluster.expire_watch_history()
+        url = (server.url + "/apis/apps/v1/deployments"
+               "?watch=true&resourceVersion=1")
+        req = urllib.request.Request(url)
+        with pytest.raises(urllib.error.HTTPError) as exc:
+            urllib.request.urlopen(req, timeout=5)
+        assert exc.value.code == 410
+
+    def test_pump_relists_and_cache_prunes_after_gap(self, server, client):
+        from wva_amd.kube.cache import CachedCluster
+
+        client.create(make_deployment("keep"))
+        client.create(make_deployment("doomed"))
+        cache = CachedCluster(client, kinds=["Deployment"]).start()
+        try:
+            assert cache.wait_for_sync(10)
+            assert cache.wait_caught_up(10)
+            sub = cache.watch(["Deployment"])  # downstream subscriber
+
+            # wait for the live stream, then sever it and erase history
+            deadline = time.time() + 10
+            while time.time() < deadline and server.active_watch_count() == 0:
+                time.sleep(0.02)
+            assert server.drop_watches() >= 1
+            # mutations the client will never see as events:
+            backing = server.cluster
+            backing.delete("Deployment", NS, "doomed")
+            backing.create(make_deployment("late"))
+            backing.expire_watch_history()
+
+            # reconnect gets 410 → relist → cache converges
+            assert cache.wait_caught_up(20)
+            assert cache.try_get("Deployment", NS, "doomed") is None
+            assert cache.try_get("Deployment", NS, "late") is not None
+            assert cache.try_get("Deployment", NS, "keep") is not None
+
+            # the subscriber got a synthetic DELETED for the ghost
+            deadline = time.time() + 5
+            deleted = set()
+            while time.time() < deadline and "doomed" not in deleted:
+                try:
+                    evt = sub.get(timeout=0.2)
+                except Exception:
+                    continue
+                if evt.type == "DELETED" and evt.obj is not None:
+                    deleted.add(evt.obj.name)
+            assert "doomed" in deleted
+        finally:
+            cache.stop()

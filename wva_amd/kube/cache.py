@@ -33,6 +33,7 @@ from typing import Any, Dict, List, Optional, Tuple
 
 from ..utils.logging import get_logger
 from .fake import ADDED, DELETED, MODIFIED, WatchEvent, _clone
+from .rest import RESYNC
 
 log = get_logger("kube.cache")
 
@@ -148,6 +149,9 @@ class CachedCluster:
                     self._synced[evt.kind] = True
                     self._sync_cv.notify_all()
                 continue
+            if evt.type == RESYNC:
+                self._resync(evt.kind, evt.obj or set())
+                continue
             if evt.type not in (ADDED, MODIFIED, DELETED):
                 continue
             self._apply(evt)
@@ -172,6 +176,22 @@ class CachedCluster:
                 # store an isolated copy — the same event object fans out
                 # to subscribers, which must not alias the store
                 self._store[key] = _clone(evt.obj)
+
+    def _resync(self, kind: str, present_keys) -> None:
+        """After a 410 relist: drop store entries of this kind that the
+        fresh LIST no longer contains (their DELETED events were lost
+        with the expired history) and deliver synthetic DELETEDs so
+        level-triggered subscribers reconcile them away."""
+        with self._lock:
+            stale = [
+                k for k in self._store
+                if k[0] == kind and (k[1], k[2]) not in present_keys
+            ]
+            removed = [self._store.pop(k) for k in stale]
+        for obj in removed:
+            log.info("resync %s: pruning %s/%s (deleted during watch gap)",
+                     kind, obj.metadata.namespace, obj.metadata.name)
+            self._fan_out(WatchEvent(DELETED, kind, obj))
 
     def _fan_out(self, evt: WatchEvent) -> None:
         with self._lock:

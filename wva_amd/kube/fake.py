@@ -44,6 +44,12 @@ class ConflictError(RuntimeError):
     pass
 
 
+class ExpiredError(RuntimeError):
+    """Requested resourceVersion predates the retained watch history —
+    the API server's 410 Gone; the watcher must relist."""
+    pass
+
+
 def _kind_of(obj: Any) -> str:
     return getattr(obj, "kind", obj.__class__.__name__)
 
@@ -90,6 +96,9 @@ class FakeCluster:
         # monotonic resourceVersion at mutation time
         self._event_log: List[Tuple[int, WatchEvent]] = []
         self._event_log_cap = 4096
+        # highest rv ever discarded from the log: resuming below it
+        # raises ExpiredError (410) instead of silently missing events
+        self._event_log_floor = 0
 
     # --- internals ---
 
@@ -102,7 +111,9 @@ class FakeCluster:
         evt = WatchEvent(type=event_type, kind=kind, obj=_clone(obj))
         self._event_log.append((self._rv, evt))
         if len(self._event_log) > self._event_log_cap:
-            del self._event_log[: len(self._event_log) // 2]
+            cut = len(self._event_log) // 2
+            self._event_log_floor = self._event_log[cut - 1][0]
+            del self._event_log[:cut]
         for kinds, q in list(self._watchers):
             if kinds is None or kind in kinds:
                 q.put(evt)
@@ -266,11 +277,27 @@ class FakeCluster:
         q: "queue.Queue[WatchEvent]" = queue.Queue()
         ks = set(kinds) if kinds else None
         with self._lock:
+            if resource_version < self._event_log_floor:
+                raise ExpiredError(
+                    f"resourceVersion {resource_version} is too old "
+                    f"(history starts after {self._event_log_floor})"
+                )
             for rv, evt in self._event_log:
                 if rv > resource_version and (ks is None or evt.kind in ks):
                     q.put(evt)
             self._watchers.append((ks, q))
         return q
+
+    def expire_watch_history(self) -> None:
+        """Discard ALL retained watch history (test helper): any
+        subsequent resume from an earlier resourceVersion gets
+        ExpiredError, like a compacted API-server watch cache."""
+        with self._lock:
+            if self._event_log:
+                self._event_log_floor = self._event_log[-1][0]
+                self._event_log.clear()
+            else:
+                self._event_log_floor = self._rv
 
     def snapshot(
         self,

@@ -36,6 +36,11 @@ from ..utils.logging import get_logger
 from . import serde
 from .fake import ADDED, DELETED, MODIFIED, ConflictError, NotFoundError, WatchEvent
 
+# relist marker after a 410 Gone: obj carries the set of (ns, name) keys
+# present in the fresh LIST so a read cache can prune deleted objects
+# whose DELETED events fell outside the retained watch history
+RESYNC = "RESYNC"
+
 log = get_logger("kube.rest")
 
 SERVICEACCOUNT_TOKEN = "/var/run/secrets/kubernetes.io/serviceaccount/token"
@@ -314,11 +319,42 @@ class _WatchPump(threading.Thread):
         while not self._stop.is_set():
             try:
                 self._stream_once()
+            except urllib.error.HTTPError as e:
+                if self._stop.is_set():
+                    return
+                if e.code == 410:
+                    # resourceVersion expired (compacted watch cache):
+                    # client-go Reflector semantics — relist and emit a
+                    # RESYNC so downstream caches drop objects whose
+                    # DELETED events were lost in the gap
+                    log.info("watch %s expired (410 Gone); relisting",
+                             self.kind)
+                    self._relist()
+                else:
+                    log.debug("watch %s reconnect after: %s", self.kind, e)
+                self._stop.wait(1.0)
             except Exception as e:  # noqa: BLE001
                 if self._stop.is_set():
                     return
                 log.debug("watch %s reconnect after: %s", self.kind, e)
                 self._stop.wait(1.0)
+
+    def _relist(self) -> None:
+        try:
+            path = serde.resource_path(self.kind, None)
+            out = self.cluster._request("GET", path)
+        except Exception as e:  # noqa: BLE001 — retry on next loop
+            log.warning("relist %s failed: %s", self.kind, e)
+            return
+        self._resource_version = (out.get("metadata") or {}).get(
+            "resourceVersion"
+        )
+        keys = set()
+        for item in out.get("items", []):
+            obj = serde.decode(self.kind, item)
+            keys.add((obj.metadata.namespace, obj.metadata.name))
+            self.out.put(WatchEvent(ADDED, self.kind, obj))
+        self.out.put(WatchEvent(RESYNC, self.kind, keys))
 
     def _stream_once(self) -> None:
         path = serde.resource_path(self.kind, None) + "?watch=true"
