@@ -584,3 +584,35 @@ class TestDedupWorkQueue:
         for name in ("x", "y", "z"):
             q.put((fn, "ns", name))
         assert [q.get(0.1)[2] for _ in range(3)] == ["x", "y", "z"]
+
+
+class TestLeaderReleaseRobustness:
+    def test_release_swallows_conflict(self):
+        """ReleaseOnCancel is best-effort: a competitor's concurrent
+        lease write (409) must not abort Manager.stop()."""
+        from wva_amd.kube.fake import ConflictError, FakeCluster
+        from wva_amd.runtime.manager import LeaderElector
+
+        cluster = FakeCluster()
+        el = LeaderElector(cluster, "wva-lease", identity="me")
+        assert el.try_acquire_or_renew()
+
+        class Conflicting:
+            def try_get(self, *a):
+                return cluster.try_get(*a)
+
+            def update(self, obj, **kw):
+                raise ConflictError("409")
+
+        el.cluster = Conflicting()
+        el.release()  # must not raise
+
+    def test_release_swallows_connection_error(self):
+        from wva_amd.runtime.manager import LeaderElector
+
+        class Down:
+            def try_get(self, *a):
+                raise OSError("connection refused")
+
+        el = LeaderElector(Down(), "wva-lease", identity="me")
+        el.release()  # must not raise

@@ -151,14 +151,20 @@ class LeaderElector:
         return False
 
     def release(self) -> None:
-        lease = self.cluster.try_get("Lease", self.LEASE_NAMESPACE, self.lease_name)
-        if lease is not None and lease.holder_identity == self.identity:
-            lease.holder_identity = ""
-            lease.renew_time = None
-            try:
+        # best-effort (ReleaseOnCancel): failure here must never abort
+        # shutdown — a competitor writing the lease concurrently
+        # (ConflictError) or an unreachable API server just means the
+        # lease expires on its own
+        try:
+            lease = self.cluster.try_get(
+                "Lease", self.LEASE_NAMESPACE, self.lease_name
+            )
+            if lease is not None and lease.holder_identity == self.identity:
+                lease.holder_identity = ""
+                lease.renew_time = None
                 self.cluster.update(lease)
-            except NotFoundError:
-                pass
+        except Exception as e:  # noqa: BLE001
+            log.debug("lease release failed (ignored): %s", e)
 
 
 class Manager:
