@@ -416,6 +416,54 @@ class TestEndToEndOverRest(TestEndToEnd):
         va = cluster.get("VariantAutoscaling", NS, VARIANT)
         assert va.status.desired_optimized_alloc.num_replicas >= 2
 
+    def test_convergence_across_watch_history_expiry(self):
+        """410 Gone chaos: the server's watch history is compacted while
+        streams are down, forcing the pump through the relist+RESYNC
+        path mid-run; the stack must still converge and the cache must
+        not retain ghosts of objects deleted during the gap."""
+        prof = ServiceProfile(
+            alpha_ms=30.0, beta_ms=1.0, max_num_seqs=16, num_gpu_blocks=2_000
+        )
+        cluster, sim, app = self.stack(replicas=1, profile=prof)
+        server = make_stack.last_server
+        model = sim.model(MODEL, NS)
+        # a doomed ConfigMap whose DELETED event will fall into the gap
+        from wva_amd.kube.objects import ConfigMap
+
+        server.cluster.create(ConfigMap(
+            metadata=ObjectMeta(name="ghost-cm", namespace=NS),
+            data={"k": "v"},
+        ))
+
+        for tick in range(10):
+            run_sim(sim, model, qps=30, seconds=5)
+            if tick == 3:
+                server.drop_watches()
+                server.cluster.delete("ConfigMap", NS, "ghost-cm")
+                server.cluster.expire_watch_history()
+            try:
+                app.saturation_engine.optimize()
+            except Exception:
+                pass
+            try:
+                app.va_reconciler.reconcile(NS, VARIANT)
+            except Exception:
+                pass
+            d = app.decision_cache.get(NS, VARIANT)
+            if d is not None and d.target_replicas > 0:
+                deploy = cluster.get("Deployment", NS, VARIANT)
+                if deploy.replicas != d.target_replicas:
+                    cluster.scale(
+                        "Deployment", NS, VARIANT, d.target_replicas
+                    )
+            sim.reconcile_deployments()
+        deploy = cluster.get("Deployment", NS, VARIANT)
+        assert deploy.replicas >= 2
+        # the cache pruned the ghost after the 410 relist
+        cache = app.cluster
+        assert cache.wait_caught_up(15)
+        assert cache.try_get("ConfigMap", NS, "ghost-cm") is None
+
     def test_stale_status_write_conflict_retried_next_tick(self):
         """A competing writer bumps the VA between the engine's read and
         a main-resource write; the next tick recovers (engine re-reads)."""
