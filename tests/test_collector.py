@@ -660,3 +660,58 @@ class TestPodVAMapperOwnerChain:
         # repeated lookups consistent (test.go:235)
         for _ in range(3):
             assert mapper.find_va_for_pod("a-11-p", "ns1", deploys) == "a"
+
+
+class TestNaNSampleRobustness:
+    """NaN samples (legal in the exposition format; only
+    PrometheusSource maps them to 0) must be skipped per-sample, never
+    crash the whole collection with int(NaN)."""
+
+    def test_nan_queue_and_kv_samples_skipped(self):
+        pod = "vllm-llama-abc12"
+        nan = float("nan")
+        results = {
+            reg.QUERY_KV_CACHE_USAGE: MetricResult(
+                query=reg.QUERY_KV_CACHE_USAGE,
+                values=[
+                    MetricValue(value=nan, labels={"pod": pod}),
+                ],
+            ),
+            reg.QUERY_QUEUE_LENGTH: MetricResult(
+                query=reg.QUERY_QUEUE_LENGTH,
+                values=[
+                    MetricValue(value=nan, labels={"pod": pod}),
+                ],
+            ),
+        }
+        c, deploy = make_cluster_with_deployment()
+        collector = ReplicaMetricsCollector(
+            _FakeSource(results), PodVAMapper(c)
+        )
+        out = collector.collect_replica_metrics(
+            "m", "default", {"default/vllm-llama": deploy}, {}, {}
+        )
+        # the only pod's samples were all NaN → it simply has no metrics
+        assert out == []
+
+    def test_nan_sample_does_not_poison_healthy_pod(self):
+        healthy, sick = "vllm-llama-abc12", "vllm-llama-def34"
+        nan = float("nan")
+        results = {
+            reg.QUERY_QUEUE_LENGTH: MetricResult(
+                query=reg.QUERY_QUEUE_LENGTH,
+                values=[
+                    MetricValue(value=2, labels={"pod": healthy}),
+                    MetricValue(value=nan, labels={"pod": sick}),
+                ],
+            ),
+        }
+        c, deploy = make_cluster_with_deployment()
+        collector = ReplicaMetricsCollector(
+            _FakeSource(results), PodVAMapper(c)
+        )
+        out = collector.collect_replica_metrics(
+            "m", "default", {"default/vllm-llama": deploy}, {}, {}
+        )
+        assert [m.pod_name for m in out] == [healthy]
+        assert out[0].queue_length == 2

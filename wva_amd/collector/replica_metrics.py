@@ -104,7 +104,10 @@ class ReplicaMetricsCollector:
                 raise RuntimeError(f"KV cache query failed: {kv.error}")
             for v in kv.values:
                 d = data_for(v.labels)
-                if d is not None:
+                # NaN guard: the exposition format allows NaN samples and
+                # only PrometheusSource maps them to 0 — int()/round() on
+                # NaN would crash the whole collection
+                if d is not None and _finite(v.value):
                     d.kv_usage = v.value
                     d.has_kv = True
                     name = _pod_name(v.labels)
@@ -117,7 +120,7 @@ class ReplicaMetricsCollector:
                 raise RuntimeError(f"queue length query failed: {q.error}")
             for v in q.values:
                 d = data_for(v.labels)
-                if d is not None:
+                if d is not None and _finite(v.value):
                     d.queue_len = int(v.value)
                     d.has_queue = True
 
