@@ -818,3 +818,62 @@ class TestExpositionLabelRoundtrip:
             assert math.isfinite(mv.value) or math.isnan(mv.value) or \
                 math.isinf(mv.value)
             assert "__name__" in mv.labels
+
+
+# --- OpenAPI validator spec conformance ---
+
+from wva_amd.kube.openapi import validate
+
+
+class TestOpenAPIValidator:
+    def test_pattern_is_partial_match_per_json_schema(self):
+        schema = {"type": "string", "pattern": "abc"}
+        assert validate(schema, "xxabcxx") == []     # partial match OK
+        assert validate(schema, "xyz") != []
+        anchored = {"type": "string", "pattern": "^abc$"}
+        assert validate(anchored, "abc") == []
+        assert validate(anchored, "xxabcxx") != []
+
+    @settings(max_examples=200, deadline=None)
+    @given(obj=JSON)
+    def test_untyped_schema_accepts_anything(self, obj):
+        assert validate({}, obj) == []
+
+    @settings(max_examples=200, deadline=None)
+    @given(
+        n=st.integers(min_value=-100, max_value=100),
+        lo=st.integers(min_value=-50, max_value=50),
+        hi=st.integers(min_value=-50, max_value=50),
+    )
+    def test_integer_bounds_exact(self, n, lo, hi):
+        schema = {"type": "integer", "minimum": lo, "maximum": hi}
+        ok = validate(schema, n) == []
+        assert ok == (lo <= n <= hi)
+
+    @settings(max_examples=150, deadline=None)
+    @given(obj=JSON)
+    def test_bool_is_not_an_integer(self, obj):
+        # kube treats booleans and integers as distinct
+        errs = validate({"type": "integer"}, True)
+        assert errs
+        errs = validate({"type": "number"}, False)
+        assert errs
+
+    @settings(max_examples=150, deadline=None)
+    @given(
+        required=st.lists(st.sampled_from(["a", "b", "c"]),
+                          max_size=3, unique=True),
+        present=st.dictionaries(
+            st.sampled_from(["a", "b", "c"]),
+            st.integers(-5, 5) | st.none(),
+            max_size=3,
+        ),
+    )
+    def test_required_means_present_and_non_null(self, required, present):
+        schema = {"type": "object", "required": list(required)}
+        errs = validate(schema, present)
+        expect_missing = [
+            r for r in required
+            if r not in present or present[r] is None
+        ]
+        assert len(errs) == len(expect_missing)

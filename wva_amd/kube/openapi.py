@@ -80,7 +80,9 @@ def _validate(schema: Dict[str, Any], obj: Any, path: str, errors: List[str]) ->
             )
         if "maxLength" in schema and len(obj) > schema["maxLength"]:
             errors.append(f"{path}: too long (max {schema['maxLength']})")
-        if "pattern" in schema and not re.fullmatch(schema["pattern"], obj):
+        # JSON Schema `pattern` is a PARTIAL match (anchor explicitly
+        # with ^...$ for a full match — both CRD patterns here do)
+        if "pattern" in schema and not re.search(schema["pattern"], obj):
             errors.append(
                 f"{path}: Invalid value: \"{obj}\": must match pattern "
                 f"{schema['pattern']}"
