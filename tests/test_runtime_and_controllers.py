@@ -616,3 +616,67 @@ class TestLeaderReleaseRobustness:
 
         el = LeaderElector(Down(), "wva-lease", identity="me")
         el.release()  # must not raise
+
+
+class TestReconcilerConflictRetry:
+    def test_status_conflict_retried_against_fresh_read(self):
+        """A competing writer bumping the VA between the reconciler's
+        read and its status write (409) must not drop the decision —
+        one retry against a fresh read persists it."""
+        from wva_amd.analyzers.interfaces import VariantDecision
+        from wva_amd.api.types import (
+            CrossVersionObjectReference, ObjectMeta, VariantAutoscaling,
+            VariantAutoscalingSpec, utcnow,
+        )
+        from wva_amd.controllers.variantautoscaling import (
+            VariantAutoscalingReconciler,
+        )
+        from wva_amd.datastore.datastore import Datastore
+        from wva_amd.engines.common import DecisionCache
+        from wva_amd.kube.fake import FakeCluster
+        from wva_amd.kube.objects import Deployment
+
+        cluster = FakeCluster()
+        cluster.create(Deployment(
+            metadata=ObjectMeta(name="d", namespace="ns"),
+        ))
+        cluster.create(VariantAutoscaling(
+            metadata=ObjectMeta(name="va", namespace="ns"),
+            spec=VariantAutoscalingSpec(
+                scale_target_ref=CrossVersionObjectReference(name="d"),
+                model_id="m",
+            ),
+        ))
+        cache = DecisionCache()
+        cache.set("ns", "va", VariantDecision(
+            variant_name="va", namespace="ns",
+            accelerator_name="MI355X", target_replicas=3,
+            last_run_time=utcnow(), metrics_available=True,
+        ))
+        rec = VariantAutoscalingReconciler(
+            cluster, Datastore(cluster), cache,
+        )
+
+        # simulate a precondition-enforcing status write: the first
+        # attempt 409s (competitor won the race and also relabeled the
+        # VA); the retry against a fresh read must persist the decision
+        from wva_amd.kube.fake import ConflictError
+
+        real_update_status = cluster.update_status
+        calls = {"n": 0}
+
+        def conflicting(obj):
+            if calls["n"] == 0:
+                calls["n"] += 1
+                competitor = cluster.get("VariantAutoscaling", "ns", "va")
+                competitor.metadata.labels["competitor"] = "1"
+                cluster.update(competitor)
+                raise ConflictError("simulated 409")
+            return real_update_status(obj)
+
+        cluster.update_status = conflicting
+        rec.reconcile("ns", "va")
+        assert calls["n"] == 1  # the conflict branch actually fired
+        va = cluster.get("VariantAutoscaling", "ns", "va")
+        assert va.status.desired_optimized_alloc.num_replicas == 3
+        assert va.metadata.labels.get("competitor") == "1"  # both writes held

@@ -33,7 +33,7 @@ from ..api.types import (
 )
 from ..datastore.datastore import Datastore
 from ..engines.common import DecisionCache
-from ..kube.fake import FakeCluster, NotFoundError
+from ..kube.fake import ConflictError, FakeCluster, NotFoundError
 from ..utils.backoff import retry_with_backoff
 from ..utils.logging import get_logger
 from ..utils.variant import matches_controller_instance, namespace_excluded
@@ -147,3 +147,20 @@ class VariantAutoscalingReconciler:
             self.cluster.update_status(va)
         except NotFoundError:
             log.debug("VA %s deleted during reconcile", va.full_name())
+        except ConflictError:
+            # a competing writer bumped the VA between our read and this
+            # write — controller-runtime would requeue the whole
+            # reconcile; one retry against a fresh read is the same
+            # convergence without re-running target resolution
+            fresh = self.cluster.try_get(
+                "VariantAutoscaling", va.namespace, va.name
+            )
+            if fresh is None:
+                return
+            fresh.status = va.status
+            try:
+                self.cluster.update_status(fresh)
+            except (NotFoundError, ConflictError) as e:
+                # the next watch event for the competing write retriggers
+                log.debug("status retry for %s dropped: %s",
+                          va.full_name(), e)
