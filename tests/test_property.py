@@ -137,3 +137,26 @@ def test_service_profile_itl_monotone(alpha, beta, batch):
     p = ServiceProfile(alpha_ms=alpha, beta_ms=beta)
     assert p.itl_ms(batch) >= p.itl_ms(max(batch - 1, 1)) - 1e-9
     assert p.itl_ms(batch) >= alpha - 1e-9
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    st.integers(min_value=50, max_value=500),
+    st.floats(min_value=1e-6, max_value=1e-2, allow_nan=False),
+    st.floats(min_value=1e4, max_value=1e9, allow_nan=False),
+)
+def test_state_dependent_overflow_rescaling(K, rate, lam):
+    """Extreme λ/μ ratios at large K drive the birth-death recursion
+    through the MaxFloat64 rescaling loop
+    (mm1modelstatedependent.go:70-113 analog) — probabilities must
+    still normalize and stay finite."""
+    import math as _math
+
+    m = MM1StateDependentModel(K, [rate])
+    m.solve(lam)
+    assert m.is_valid
+    assert all(_math.isfinite(p) and p >= 0 for p in m.p)
+    assert abs(sum(m.p) - 1.0) < 1e-6
+    # at λ ≫ μ the system is pinned at capacity
+    assert m.avg_num_in_system > 0.9 * K
+    assert _math.isfinite(m.throughput)
