@@ -221,3 +221,30 @@ class TestRemainingKindRoundtrips:
         )
         dec = roundtrip(lease)
         assert dec.holder_identity == holder
+
+
+class TestMalformedWirePayloads:
+    """A degraded API server / proxy can hand the REST client anything;
+    decode must degrade to defaults, never crash inside a watch pump."""
+
+    @pytest.mark.parametrize("payload", [
+        {}, {"metadata": None}, {"spec": "nope"}, {"status": []},
+        {"metadata": 7}, {"spec": ["x"]}, None,
+    ])
+    def test_every_kind_survives_garbage(self, payload):
+        for kind in serde.SERDE:
+            obj = serde.decode(kind, payload)
+            assert obj.kind == kind
+            # still round-trippable after the degradation
+            serde.encode(obj)
+
+    @settings(max_examples=150, deadline=None)
+    @given(garbage=st.dictionaries(
+        st.sampled_from(["metadata", "spec", "status", "data", "x"]),
+        st.none() | st.integers() | st.text(max_size=5)
+        | st.lists(st.integers(), max_size=2),
+        max_size=4,
+    ))
+    def test_va_decode_fuzz(self, garbage):
+        obj = serde.decode("VariantAutoscaling", garbage)
+        assert obj.kind == "VariantAutoscaling"

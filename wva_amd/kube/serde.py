@@ -496,6 +496,21 @@ def encode(obj: Any) -> Dict[str, Any]:
 
 
 def decode(kind: str, d: Dict[str, Any]) -> Any:
+    # harden against malformed wire payloads (a proxy error body, a
+    # truncated chunk): the per-kind decoders expect mapping sections,
+    # and a non-dict there must degrade to "absent", not AttributeError
+    # deep inside a watch pump
+    if not isinstance(d, dict):
+        d = {}
+    else:
+        cleaned = None
+        for section in ("metadata", "spec", "status"):
+            if section in d and not isinstance(d[section], dict):
+                if cleaned is None:
+                    cleaned = dict(d)
+                cleaned.pop(section)
+        if cleaned is not None:
+            d = cleaned
     return SERDE[kind][1](d)
 
 
