@@ -304,8 +304,16 @@ class K8sTestServer:
                             f"VariantAutoscaling is invalid: {'; '.join(errs)}",
                         ))
                 obj = serde.decode(kind, body)
+                # API-server semantics: metadata.generation bumps when
+                # the spec (not status/metadata) changes
+                prev = outer.cluster.try_get(kind, ns or "", name)
+                bump = False
+                if prev is not None:
+                    prev_spec = serde.encode(prev).get("spec")
+                    if prev_spec != body.get("spec"):
+                        bump = True
                 try:
-                    updated = outer.cluster.update(obj)
+                    updated = outer.cluster.update(obj, bump_generation=bump)
                 except NotFoundError:
                     return self._send(404, _status_body(404, "NotFound", "not found"))
                 except ConflictError as e:

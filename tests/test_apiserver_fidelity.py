@@ -860,3 +860,30 @@ class TestWatchHistoryExpiry:
             assert "doomed" in deleted
         finally:
             cache.stop()
+
+
+class TestGenerationSemantics:
+    def test_spec_change_bumps_generation(self, server, client):
+        client.create(make_deployment("g"))
+        d = client.get("Deployment", NS, "g")
+        gen0 = d.metadata.generation
+        d.replicas = 7  # spec change
+        d2 = client.update(d)
+        assert d2.metadata.generation == gen0 + 1
+
+    def test_metadata_only_change_keeps_generation(self, server, client):
+        client.create(make_deployment("g2"))
+        d = client.get("Deployment", NS, "g2")
+        gen0 = d.metadata.generation
+        d.metadata.labels["touched"] = "1"
+        d2 = client.update(d)
+        assert d2.metadata.generation == gen0
+
+    def test_status_write_keeps_generation(self, server, client):
+        client.create(make_va("g3"))
+        va = client.get("VariantAutoscaling", NS, "g3")
+        gen0 = va.metadata.generation
+        va.status.desired_optimized_alloc.accelerator = "MI355X"
+        va.status.desired_optimized_alloc.num_replicas = 2
+        out = client.update_status(va)
+        assert out.metadata.generation == gen0
