@@ -887,3 +887,40 @@ class TestGenerationSemantics:
         va.status.desired_optimized_alloc.num_replicas = 2
         out = client.update_status(va)
         assert out.metadata.generation == gen0
+
+    def test_condition_observed_generation_tracks_spec_edits(self, server):
+        """Full-stack: a spec edit bumps generation; the next reconcile
+        stamps conditions with the NEW observedGeneration (the signal
+        kubectl uses to show a condition is up to date)."""
+        from wva_amd.controllers.variantautoscaling import (
+            VariantAutoscalingReconciler,
+        )
+        from wva_amd.datastore.datastore import Datastore
+        from wva_amd.engines.common import DecisionCache
+
+        client = RestCluster(server.url)
+        try:
+            client.create(make_deployment("g4"))
+            client.create(make_va("g4"))
+            rec = VariantAutoscalingReconciler(
+                client, Datastore(client), DecisionCache(),
+            )
+            rec.reconcile(NS, "g4")
+            va = client.get("VariantAutoscaling", NS, "g4")
+            cond0 = next(
+                c for c in va.status.conditions if c.type == "TargetResolved"
+            )
+            gen0 = va.metadata.generation
+            assert cond0.observed_generation == gen0
+
+            va.spec.variant_cost = "42.0"  # spec edit → generation bump
+            client.update(va)
+            rec.reconcile(NS, "g4")
+            va2 = client.get("VariantAutoscaling", NS, "g4")
+            cond1 = next(
+                c for c in va2.status.conditions if c.type == "TargetResolved"
+            )
+            assert va2.metadata.generation == gen0 + 1
+            assert cond1.observed_generation == gen0 + 1
+        finally:
+            client.close()
