@@ -116,6 +116,7 @@ def main(argv=None) -> int:
     try:
         while not stop:
             time.sleep(0.5)
+        log.info("signal received, shutting down")
     finally:
         app.stop()
         cache_stop = getattr(cluster, "stop", None)
