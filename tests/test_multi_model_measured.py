@@ -151,3 +151,130 @@ class TestMeasuredMultiModel:
         )
         conds_l = {c.type: c.status for c in va_l.status.conditions}
         assert conds_l.get("OptimizationReady") == "True"
+
+
+class TestMeasuredMultiModelServiceClasses:
+    def test_priorities_and_keda_emission(self):
+        """BASELINE config #4 capstone: both measured model families
+        under the Inferno SLO analyzer with DISTINCT service classes
+        (premium llama prio 1, freemium mixtral prio 10), each sized
+        against its own class SLO, and the KEDA/HPA-shaped
+        `wva_desired_replicas` gauges emitted for both variants."""
+        from prometheus_client import CollectorRegistry
+        from wva_amd.analyzers.modelanalyzer import InfernoAnalyzer
+        from wva_amd.app import build_app
+        from wva_amd.inferno.system import System
+        from wva_amd.inferno.types import (
+            AcceleratorSpec,
+            ModelAcceleratorPerfData,
+            ModelTarget,
+            ServiceClassSpec,
+            ServiceParmsSpec,
+            SystemData,
+        )
+
+        llama = (load_profile("calibration_8b.json")
+                 if os.path.exists(os.path.join(
+                     PROFILES_DIR, "calibration_8b.json"))
+                 else ServiceProfile(alpha_ms=4.77, beta_ms=0.0266))
+        mixtral = load_profile("calibration_mixtral.json")
+
+        cluster = FakeCluster()
+        make_variant(cluster, "vllm-llama", "meta-llama/Llama-3.1-8B", "10.0")
+        make_variant(cluster, "vllm-mixtral", "mistralai/Mixtral-8x7B", "20.0")
+        sim = ClusterSim(cluster, warm_start=True)
+        sim.register_variant(
+            "meta-llama/Llama-3.1-8B", "default", "vllm-llama", llama
+        )
+        sim.register_variant(
+            "mistralai/Mixtral-8x7B", "default", "vllm-mixtral", mixtral
+        )
+        sim.reconcile_deployments()
+
+        config = Config()
+        config.update_saturation_config(
+            SaturationScalingConfig.from_dict({"analyzerName": "inferno"})
+        )
+        config.mark_bootstrap_complete()
+        registry = CollectorRegistry()
+        app = build_app(
+            cluster, config, source=SimMetricsSource(sim),
+            metrics_registry=registry, start_engines=False,
+        )
+        # Inferno system: per-model measured parms + two service classes
+        system = System(SystemData(
+            accelerators=[AcceleratorSpec(name="MI355X", type="MI355X",
+                                          cost=50.0)],
+            models=[
+                ModelAcceleratorPerfData(
+                    name="meta-llama/Llama-3.1-8B", acc="MI355X",
+                    max_batch_size=256, at_tokens=50,
+                    service_parms=ServiceParmsSpec(
+                        alpha=llama.alpha_ms, beta=llama.beta_ms,
+                    ),
+                ),
+                ModelAcceleratorPerfData(
+                    name="mistralai/Mixtral-8x7B", acc="MI355X",
+                    max_batch_size=64, at_tokens=50,
+                    service_parms=ServiceParmsSpec(
+                        alpha=mixtral.alpha_ms, beta=max(mixtral.beta_ms, 1e-4),
+                    ),
+                ),
+            ],
+            service_classes=[
+                ServiceClassSpec(
+                    name="premium", priority=1,
+                    model_targets=[ModelTarget(
+                        model="meta-llama/Llama-3.1-8B",
+                        slo_itl=20.0, slo_ttft=1000.0,
+                    )],
+                ),
+                ServiceClassSpec(
+                    name="freemium", priority=10,
+                    model_targets=[ModelTarget(
+                        model="mistralai/Mixtral-8x7B",
+                        slo_itl=60.0, slo_ttft=5000.0,
+                    )],
+                ),
+            ],
+        ))
+        app.saturation_engine.inferno_analyzer = InfernoAnalyzer(
+            system, enable_tuner=False,
+        )
+
+        m_llama = sim.model("meta-llama/Llama-3.1-8B", "default")
+        m_mix = sim.model("mistralai/Mixtral-8x7B", "default")
+        for tick in range(6):
+            for _ in range(60):
+                sim.generate_arrivals(m_llama, lambda t: 400.0, 0.25, 100, 50)
+                sim.generate_arrivals(m_mix, lambda t: 60.0, 0.25, 100, 50)
+                sim.advance(0.25)
+            app.saturation_engine.optimize()
+            app.va_reconciler.reconcile("default", "vllm-llama")
+            app.va_reconciler.reconcile("default", "vllm-mixtral")
+            for name in ("vllm-llama", "vllm-mixtral"):
+                d = app.decision_cache.get("default", name)
+                if d and d.target_replicas > 0:
+                    cluster.scale("Deployment", "default", name,
+                                  d.target_replicas)
+            sim.reconcile_deployments()
+
+        d_l = app.decision_cache.get("default", "vllm-llama")
+        d_m = app.decision_cache.get("default", "vllm-mixtral")
+        assert d_l is not None and d_l.target_replicas >= 1
+        assert d_m is not None and d_m.target_replicas >= 1
+        # 400 req/s exceeds one 8B replica's premium-SLO rate
+        assert d_l.target_replicas >= 2
+
+        # KEDA/HPA contract: wva_desired_replicas emitted per variant
+        # with the accelerator label
+        samples = {}
+        for fam in registry.collect():
+            if fam.name == "wva_desired_replicas":
+                for s in fam.samples:
+                    samples[s.labels["variant_name"]] = (
+                        s.value, s.labels["accelerator_type"],
+                    )
+        assert samples["vllm-llama"][0] == d_l.target_replicas
+        assert samples["vllm-mixtral"][0] == d_m.target_replicas
+        assert samples["vllm-llama"][1] == "MI355X"
