@@ -88,6 +88,8 @@ class TestLimiterUnderExhaustion:
         d = app.decision_cache.get(NS, VARIANT)
         assert d is not None
         assert d.target_replicas == 1
+        # the full truncation is reported as a limit, not "no scale-up"
+        assert "limited" in d.optimization_ready_message
 
 
 class TestTargetConditionLifecycle:

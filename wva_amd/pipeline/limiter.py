@@ -111,6 +111,14 @@ class DefaultLimiter:
     def _build_step_reason(d: VariantDecision) -> str:
         change = d.target_replicas - d.current_replicas
         if change <= 0:
+            if d.was_limited:
+                # the ask was truncated all the way back to current —
+                # say so instead of a misleading "no scale-up"
+                return (
+                    f"limited: pool exhausted, 0 GPUs allocated (wanted "
+                    f"{d.original_target_replicas}, kept "
+                    f"{d.target_replicas})"
+                )
             return f"no scale-up (target={d.target_replicas}, current={d.current_replicas})"
         if d.was_limited:
             return f"limited: allocated {d.gpus_allocated} GPUs for +{change} replicas"
